@@ -1,0 +1,12 @@
+from .datasets import (
+    ObjectAttributeDataset, SynthDataset, SyntheticImageDataset,
+    collate_fn, get_classnames, insert_rand_word,
+)
+from .tokenizer import HashTokenizer, load_tokenizer
+from .transforms import TrainTransform, EvalTransform
+
+__all__ = [
+    "ObjectAttributeDataset", "SynthDataset", "SyntheticImageDataset",
+    "collate_fn", "get_classnames", "insert_rand_word",
+    "HashTokenizer", "load_tokenizer", "TrainTransform", "EvalTransform",
+]

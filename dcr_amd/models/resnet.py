@@ -1,0 +1,81 @@
+"""UNet/VAE ResNet blocks and up/down-samplers (SD-2.1 shapes).
+
+Reference behavior: diffusers ResnetBlock2D / Downsample2D / Upsample2D as
+exercised by the finetune loop (/root/reference/diff_train.py:644) —
+rebuilt on dcr_amd ops: GroupNorm+SiLU is one fused HIP kernel
+(SURVEY.md §2.4.A), convs go through MIOpen via torch.
+"""
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from .layers import GroupNormOp
+
+
+class ResnetBlock2D(nn.Module):
+    def __init__(
+        self,
+        in_channels: int,
+        out_channels: Optional[int] = None,
+        temb_channels: Optional[int] = 1280,
+        groups: int = 32,
+        eps: float = 1e-6,
+        dropout: float = 0.0,
+        output_scale_factor: float = 1.0,
+    ):
+        super().__init__()
+        out_channels = out_channels or in_channels
+        self.in_channels = in_channels
+        self.out_channels = out_channels
+        self.output_scale_factor = output_scale_factor
+
+        self.norm1 = GroupNormOp(groups, in_channels, eps=eps, fused_silu=True)
+        self.conv1 = nn.Conv2d(in_channels, out_channels, 3, padding=1)
+        if temb_channels is not None:
+            self.time_emb_proj = nn.Linear(temb_channels, out_channels)
+        else:
+            self.time_emb_proj = None
+        self.norm2 = GroupNormOp(groups, out_channels, eps=eps, fused_silu=True)
+        self.dropout = nn.Dropout(dropout)
+        self.conv2 = nn.Conv2d(out_channels, out_channels, 3, padding=1)
+        if in_channels != out_channels:
+            self.conv_shortcut = nn.Conv2d(in_channels, out_channels, 1)
+        else:
+            self.conv_shortcut = None
+
+    def forward(self, x: torch.Tensor, temb: Optional[torch.Tensor] = None) -> torch.Tensor:
+        h = self.norm1(x)          # fused GN+SiLU
+        h = self.conv1(h)
+        if self.time_emb_proj is not None and temb is not None:
+            h = h + self.time_emb_proj(F.silu(temb))[:, :, None, None]
+        h = self.norm2(h)          # fused GN+SiLU
+        h = self.dropout(h)
+        h = self.conv2(h)
+        if self.conv_shortcut is not None:
+            x = self.conv_shortcut(x)
+        return (x + h) / self.output_scale_factor
+
+
+class Downsample2D(nn.Module):
+    def __init__(self, channels: int, out_channels: Optional[int] = None):
+        super().__init__()
+        out_channels = out_channels or channels
+        self.conv = nn.Conv2d(channels, out_channels, 3, stride=2, padding=1)
+
+    def forward(self, x):
+        return self.conv(x)
+
+
+class Upsample2D(nn.Module):
+    def __init__(self, channels: int, out_channels: Optional[int] = None):
+        super().__init__()
+        out_channels = out_channels or channels
+        self.conv = nn.Conv2d(channels, out_channels, 3, padding=1)
+
+    def forward(self, x):
+        x = F.interpolate(x, scale_factor=2.0, mode="nearest")
+        return self.conv(x)

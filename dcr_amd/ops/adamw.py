@@ -1,0 +1,127 @@
+"""Flat-buffer fused AdamW for MI355X.
+
+Reference behavior: torch.optim.AdamW over ~865M UNet params
+(/root/reference/diff_train.py:435-446; betas 0.9/0.999, eps 1e-8,
+weight_decay 1e-2). Instead of ~700 per-tensor kernel launches per step,
+this optimizer re-parameterizes the model so that
+
+* every parameter tensor is a VIEW into one contiguous flat buffer,
+* every ``.grad`` is a view into one contiguous flat gradient buffer,
+* Adam state (m, v) are two more flat buffers,
+
+so one step is ONE elementwise HIP kernel over four flat arrays —
+HBM-bound by design (~8 TB/s on MI355X), and the flat grad buffer doubles
+as the bucket arena for RCCL all-reduce (dcr_amd/parallel/ddp.py).
+
+Flatten AFTER moving the model to its device; ``p.data`` is replaced by
+buffer views.
+"""
+from __future__ import annotations
+
+import math
+from typing import Iterable, List
+
+import torch
+
+from . import use_hip, require_hip
+
+
+class FusedAdamW:
+    def __init__(
+        self,
+        params: Iterable[torch.nn.Parameter],
+        lr: float = 5e-6,
+        betas=(0.9, 0.999),
+        eps: float = 1e-8,
+        weight_decay: float = 1e-2,
+    ):
+        self.lr = lr
+        self.beta1, self.beta2 = betas
+        self.eps = eps
+        self.weight_decay = weight_decay
+        self.step_count = 0
+
+        self.params: List[torch.nn.Parameter] = [p for p in params if p.requires_grad]
+        if not self.params:
+            raise ValueError("FusedAdamW: no trainable parameters")
+        device = self.params[0].device
+        dtype = self.params[0].dtype
+        for p in self.params:
+            if p.device != device or p.dtype != dtype:
+                raise ValueError("FusedAdamW requires uniform param dtype/device")
+
+        total = sum(p.numel() for p in self.params)
+        self.flat_param = torch.empty(total, device=device, dtype=dtype)
+        self.flat_grad = torch.zeros(total, device=device, dtype=dtype)
+        self.exp_avg = torch.zeros(total, device=device, dtype=torch.float32)
+        self.exp_avg_sq = torch.zeros(total, device=device, dtype=torch.float32)
+
+        # Re-parameterize: params become views of flat_param, grads views of flat_grad.
+        offset = 0
+        self.offsets: List[int] = []
+        for p in self.params:
+            n = p.numel()
+            self.flat_param[offset : offset + n].copy_(p.data.reshape(-1))
+            p.data = self.flat_param[offset : offset + n].view_as(p.data)
+            p.grad = self.flat_grad[offset : offset + n].view_as(p.data)
+            self.offsets.append(offset)
+            offset += n
+        self.numel = total
+
+    # -- torch.optim-ish surface ------------------------------------------
+    def zero_grad(self, set_to_none: bool = False):
+        # grads are persistent views; autograd accumulates into them (+=)
+        self.flat_grad.zero_()
+
+    @torch.no_grad()
+    def step(self, lr: float | None = None):
+        if lr is not None:
+            self.lr = lr
+        self.step_count += 1
+        t = self.step_count
+        if use_hip(self.flat_param):
+            m = require_hip("adamw")
+            if m is not None:
+                m.adamw_step(
+                    self.flat_param, self.flat_grad, self.exp_avg, self.exp_avg_sq,
+                    self.lr, self.beta1, self.beta2, self.eps, self.weight_decay, t,
+                )
+                return
+        # reference implementation (CPU tests / debug fallback)
+        bc1 = 1.0 - self.beta1 ** t
+        bc2 = 1.0 - self.beta2 ** t
+        g = self.flat_grad.float()
+        self.exp_avg.mul_(self.beta1).add_(g, alpha=1 - self.beta1)
+        self.exp_avg_sq.mul_(self.beta2).addcmul_(g, g, value=1 - self.beta2)
+        denom = (self.exp_avg_sq / bc2).sqrt_().add_(self.eps)
+        upd = (self.exp_avg / bc1) / denom
+        self.flat_param.add_(
+            (upd + self.weight_decay * self.flat_param.float()).to(self.flat_param.dtype),
+            alpha=-self.lr,
+        )
+
+    @torch.no_grad()
+    def clip_grad_norm_(self, max_norm: float) -> torch.Tensor:
+        """Global L2 grad clip (reference: diff_train.py:657-663, max 1.0)."""
+        norm = torch.linalg.vector_norm(self.flat_grad.float())
+        scale = max_norm / (norm + 1e-6)
+        if float(norm) > max_norm:
+            self.flat_grad.mul_(scale.to(self.flat_grad.dtype))
+        return norm
+
+    # -- checkpointing -----------------------------------------------------
+    def state_dict(self):
+        return {
+            "step": self.step_count,
+            "lr": self.lr,
+            "exp_avg": self.exp_avg,
+            "exp_avg_sq": self.exp_avg_sq,
+            "flat_param": self.flat_param,
+        }
+
+    def load_state_dict(self, sd):
+        self.step_count = sd["step"]
+        self.lr = sd["lr"]
+        self.exp_avg.copy_(sd["exp_avg"])
+        self.exp_avg_sq.copy_(sd["exp_avg_sq"])
+        self.flat_param.copy_(sd["flat_param"])

@@ -1,0 +1,215 @@
+"""Functional op wrappers: HIP kernels on GPU, PyTorch reference on CPU.
+
+Each op corresponds to a fused hot spot of the SD-2.1 finetune / sampling
+loop the reference runs through libraries:
+
+* ``group_norm_silu``  — every ResNet-block norm in UNet/VAE
+  (reference: diffusers ResnetBlock2D, /root/reference/diff_train.py:644).
+* ``layer_norm``       — transformer blocks in UNet + CLIP text encoder.
+* ``geglu``            — UNet transformer FeedForward gate.
+* ``attention``        — self/cross attention (flash-style HIP kernel).
+* ``add_noise`` / ``get_velocity`` — DDPM scheduler math
+  (reference: diff_train.py:632,650).
+* ``cfg_combine``      — classifier-free guidance in sampling.
+"""
+from __future__ import annotations
+
+import math
+from typing import Optional
+
+import torch
+import torch.nn.functional as F
+
+from . import use_hip, require_hip
+
+
+# --------------------------------------------------------------------------
+# GroupNorm (+ optional fused SiLU)
+# --------------------------------------------------------------------------
+class _GroupNormSiLU(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, num_groups, eps, apply_silu):
+        m = require_hip("group_norm_silu")
+        if m is None:  # debug fallback on GPU
+            return _gn_silu_ref(x, weight, bias, num_groups, eps, apply_silu)
+        y, mean, rstd = m.groupnorm_silu_fwd(x, weight, bias, num_groups, eps, apply_silu)
+        ctx.save_for_backward(x, weight, bias, mean, rstd)
+        ctx.num_groups = num_groups
+        ctx.apply_silu = apply_silu
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight, bias, mean, rstd = ctx.saved_tensors
+        m = require_hip("group_norm_silu")
+        dx, dw, db = m.groupnorm_silu_bwd(
+            dy.contiguous(), x, weight, bias, mean, rstd, ctx.num_groups, ctx.apply_silu
+        )
+        return dx, dw, db, None, None, None
+
+
+def _gn_silu_ref(x, weight, bias, num_groups, eps, apply_silu):
+    # fp32 accumulation regardless of input dtype (matches HIP kernel numerics)
+    y = F.group_norm(x.float(), num_groups, weight.float(), bias.float(), eps)
+    if apply_silu:
+        y = F.silu(y)
+    return y.to(x.dtype)
+
+
+def group_norm_silu(
+    x: torch.Tensor,
+    weight: torch.Tensor,
+    bias: torch.Tensor,
+    num_groups: int = 32,
+    eps: float = 1e-5,
+    apply_silu: bool = True,
+) -> torch.Tensor:
+    if use_hip(x):
+        return _GroupNormSiLU.apply(x.contiguous(), weight, bias, num_groups, eps, apply_silu)
+    return _gn_silu_ref(x, weight, bias, num_groups, eps, apply_silu)
+
+
+# --------------------------------------------------------------------------
+# LayerNorm
+# --------------------------------------------------------------------------
+class _LayerNorm(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, eps):
+        m = require_hip("layer_norm")
+        if m is None:
+            return F.layer_norm(x, (x.shape[-1],), weight, bias, eps)
+        y, mean, rstd = m.layernorm_fwd(x, weight, bias, eps)
+        ctx.save_for_backward(x, weight, mean, rstd)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight, mean, rstd = ctx.saved_tensors
+        m = require_hip("layer_norm")
+        dx, dw, db = m.layernorm_bwd(dy.contiguous(), x, weight, mean, rstd)
+        return dx, dw, db, None
+
+
+def layer_norm(
+    x: torch.Tensor,
+    weight: torch.Tensor,
+    bias: torch.Tensor,
+    eps: float = 1e-5,
+) -> torch.Tensor:
+    if use_hip(x):
+        return _LayerNorm.apply(x.contiguous(), weight, bias, eps)
+    return F.layer_norm(x, (x.shape[-1],), weight, bias, eps)
+
+
+# --------------------------------------------------------------------------
+# GEGLU: out = a * gelu(g) with [a, g] = split(x, 2, dim=-1)
+# --------------------------------------------------------------------------
+class _GEGLU(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        m = require_hip("geglu")
+        if m is None:
+            a, g = x.chunk(2, dim=-1)
+            return a * F.gelu(g)
+        y = m.geglu_fwd(x)
+        ctx.save_for_backward(x)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (x,) = ctx.saved_tensors
+        m = require_hip("geglu")
+        return m.geglu_bwd(dy.contiguous(), x)
+
+
+def geglu(x: torch.Tensor) -> torch.Tensor:
+    """x[..., :N] * gelu(x[..., N:]) — exact (erf) GELU."""
+    if use_hip(x):
+        return _GEGLU.apply(x.contiguous())
+    a, g = x.chunk(2, dim=-1)
+    return a * F.gelu(g)
+
+
+# --------------------------------------------------------------------------
+# Attention (flash-style). q,k,v: [B, H, Lq/Lk, D].
+# --------------------------------------------------------------------------
+def attention(
+    q: torch.Tensor,
+    k: torch.Tensor,
+    v: torch.Tensor,
+    causal: bool = False,
+    scale: Optional[float] = None,
+) -> torch.Tensor:
+    from . import ext
+
+    m = ext()
+    if use_hip(q) and m is not None and hasattr(m, "attn_fwd") and not q.requires_grad \
+            and q.dtype == torch.bfloat16 and q.shape[-1] <= 128 and not causal:
+        # inference path: hand-written CDNA4 flash forward
+        return _AttnFwdOnly.apply(q, k, v, scale)
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    return F.scaled_dot_product_attention(q, k, v, is_causal=causal, scale=scale)
+
+
+class _AttnFwdOnly(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, scale):
+        from . import ext
+
+        m = ext()
+        s = scale if scale is not None else 1.0 / math.sqrt(q.shape[-1])
+        return m.attn_fwd(q.contiguous(), k.contiguous(), v.contiguous(), s)
+
+    @staticmethod
+    def backward(ctx, dy):  # pragma: no cover
+        raise RuntimeError("attn_fwd is inference-only; training uses SDPA for now")
+
+
+# --------------------------------------------------------------------------
+# Diffusion scheduler math (no autograd needed: inputs are leaves w/o grad)
+# --------------------------------------------------------------------------
+def _gather_sqrt(alphas_cumprod: torch.Tensor, t: torch.Tensor, ndim: int):
+    ac = alphas_cumprod.to(torch.float32).gather(0, t)
+    shape = [t.shape[0]] + [1] * (ndim - 1)
+    sqrt_ac = ac.sqrt().view(shape)
+    sqrt_1mac = (1.0 - ac).sqrt().view(shape)
+    return sqrt_ac, sqrt_1mac
+
+
+def add_noise(
+    x0: torch.Tensor, noise: torch.Tensor, alphas_cumprod: torch.Tensor, t: torch.Tensor
+) -> torch.Tensor:
+    """x_t = sqrt(ac_t) x0 + sqrt(1-ac_t) noise, per-sample t."""
+    from . import ext
+
+    m = ext()
+    if use_hip(x0) and m is not None:
+        return m.add_noise(x0.contiguous(), noise.contiguous(),
+                           alphas_cumprod.to(x0.device, torch.float32), t.contiguous())
+    sa, sb = _gather_sqrt(alphas_cumprod.to(x0.device), t, x0.dim())
+    return (sa * x0.float() + sb * noise.float()).to(x0.dtype)
+
+
+def get_velocity(
+    x0: torch.Tensor, noise: torch.Tensor, alphas_cumprod: torch.Tensor, t: torch.Tensor
+) -> torch.Tensor:
+    """v = sqrt(ac_t) noise - sqrt(1-ac_t) x0 (v-prediction target)."""
+    from . import ext
+
+    m = ext()
+    if use_hip(x0) and m is not None:
+        return m.get_velocity(x0.contiguous(), noise.contiguous(),
+                              alphas_cumprod.to(x0.device, torch.float32), t.contiguous())
+    sa, sb = _gather_sqrt(alphas_cumprod.to(x0.device), t, x0.dim())
+    return (sa * noise.float() - sb * x0.float()).to(x0.dtype)
+
+
+def cfg_combine(eps_uncond: torch.Tensor, eps_text: torch.Tensor, scale: float) -> torch.Tensor:
+    """Classifier-free guidance: eps_u + s * (eps_t - eps_u)."""
+    from . import ext
+
+    m = ext()
+    if use_hip(eps_uncond) and m is not None:
+        return m.cfg_combine(eps_uncond.contiguous(), eps_text.contiguous(), float(scale))
+    return eps_uncond + scale * (eps_text - eps_uncond)

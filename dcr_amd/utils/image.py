@@ -1,0 +1,39 @@
+"""Image grid helpers (the reference imports a missing utils.draw_utils;
+SURVEY.md §2.6 defect 1 — we implement the intent)."""
+from __future__ import annotations
+
+from typing import Sequence
+
+import numpy as np
+import torch
+from PIL import Image
+
+
+def tensor_to_pil(t: torch.Tensor) -> Image.Image:
+    """[-1,1] or [0,1] CHW float tensor -> PIL RGB."""
+    t = t.detach().float().cpu()
+    if t.min() < -0.01:
+        t = (t + 1) / 2
+    arr = (t.clamp(0, 1) * 255).round().to(torch.uint8).permute(1, 2, 0).numpy()
+    return Image.fromarray(arr)
+
+
+def concat_h(images: Sequence[Image.Image]) -> Image.Image:
+    """Horizontal concat (reference: missing utils.draw_utils.concat_h,
+    used at diff_train.py:611,701)."""
+    h = max(im.height for im in images)
+    w = sum(im.width for im in images)
+    out = Image.new("RGB", (w, h))
+    x = 0
+    for im in images:
+        out.paste(im, (x, 0))
+        x += im.width
+    return out
+
+
+def image_grid(images: Sequence[Image.Image], rows: int, cols: int) -> Image.Image:
+    w, h = images[0].size
+    grid = Image.new("RGB", (cols * w, rows * h))
+    for i, im in enumerate(images[: rows * cols]):
+        grid.paste(im, ((i % cols) * w, (i // cols) * h))
+    return grid
