@@ -60,7 +60,7 @@ class VAEAttention(nn.Module):
     def __init__(self, channels: int, groups: int):
         super().__init__()
         self.group_norm = GroupNormOp(groups, channels, eps=1e-6, fused_silu=False)
-        self.attn = Attention(channels, heads=1, dim_head=channels)
+        self.attn = Attention(channels, heads=1, dim_head=channels, bias=True)
 
     def forward(self, x):
         B, C, H, W = x.shape

@@ -1,0 +1,186 @@
+// PyTorch bindings for the dcr_amd HIP kernel library (MI355X / gfx950).
+// Host-only TU: dispatches dtypes + shapes and calls launchers from the
+// .hip TUs. Fails loudly on unsupported layouts rather than silently
+// falling back (bench validity: the HIP path must be the one that runs).
+
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+
+#include "dcr_launchers.h"
+
+using namespace dcr;
+
+namespace {
+
+DType dtype_of(const at::Tensor& t) {
+  switch (t.scalar_type()) {
+    case at::kFloat: return DT_F32;
+    case at::kHalf: return DT_F16;
+    case at::kBFloat16: return DT_BF16;
+    default: TORCH_CHECK(false, "dcr_hip: unsupported dtype ", t.scalar_type());
+  }
+}
+
+hipStream_t cur_stream() {
+  return at::cuda::getCurrentCUDAStream().stream();
+}
+
+at::Tensor as_f32(const at::Tensor& t) {
+  return t.scalar_type() == at::kFloat ? t.contiguous()
+                                       : t.to(at::kFloat).contiguous();
+}
+
+}  // namespace
+
+// ---------------------------------------------------------------- GroupNorm
+std::vector<at::Tensor> groupnorm_silu_fwd(at::Tensor x, at::Tensor w, at::Tensor b,
+                                           int64_t groups, double eps, bool silu) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous(), "gn: x must be contiguous CUDA");
+  TORCH_CHECK(x.dim() >= 2, "gn: x must be [N, C, ...]");
+  const int64_t N = x.size(0), C = x.size(1);
+  const int64_t HW = x.numel() / (N * C);
+  TORCH_CHECK(C % groups == 0, "gn: C % groups != 0");
+  auto wf = as_f32(w), bf = as_f32(b);
+  auto y = at::empty_like(x);
+  auto mean = at::empty({N * groups}, x.options().dtype(at::kFloat));
+  auto rstd = at::empty_like(mean);
+  gn_fwd_launch(dtype_of(x), x.data_ptr(), wf.data_ptr<float>(), bf.data_ptr<float>(),
+                y.data_ptr(), mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                (int)(N * groups), (int)groups, (int)(C / groups), (int)HW,
+                (float)eps, silu, cur_stream());
+  return {y, mean, rstd};
+}
+
+std::vector<at::Tensor> groupnorm_silu_bwd(at::Tensor dy, at::Tensor x, at::Tensor w,
+                                           at::Tensor b, at::Tensor mean, at::Tensor rstd,
+                                           int64_t groups, bool silu) {
+  TORCH_CHECK(dy.is_contiguous() && x.is_contiguous());
+  const int64_t N = x.size(0), C = x.size(1);
+  const int64_t HW = x.numel() / (N * C);
+  auto wf = as_f32(w), bf = as_f32(b);
+  auto dx = at::empty_like(x);
+  auto dw = at::zeros({C}, x.options().dtype(at::kFloat));
+  auto db = at::zeros({C}, x.options().dtype(at::kFloat));
+  gn_bwd_launch(dtype_of(x), dy.data_ptr(), x.data_ptr(), wf.data_ptr<float>(),
+                bf.data_ptr<float>(), mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                dx.data_ptr(), dw.data_ptr<float>(), db.data_ptr<float>(),
+                (int)(N * groups), (int)groups, (int)(C / groups), (int)HW,
+                silu, cur_stream());
+  return {dx, dw.to(w.scalar_type()), db.to(b.scalar_type())};
+}
+
+// ---------------------------------------------------------------- LayerNorm
+std::vector<at::Tensor> layernorm_fwd(at::Tensor x, at::Tensor w, at::Tensor b,
+                                      double eps) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous());
+  const int64_t Nd = x.size(-1);
+  const int64_t M = x.numel() / Nd;
+  auto wf = as_f32(w), bf = as_f32(b);
+  auto y = at::empty_like(x);
+  auto mean = at::empty({M}, x.options().dtype(at::kFloat));
+  auto rstd = at::empty_like(mean);
+  ln_fwd_launch(dtype_of(x), x.data_ptr(), wf.data_ptr<float>(), bf.data_ptr<float>(),
+                y.data_ptr(), mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                M, (int)Nd, (float)eps, cur_stream());
+  return {y, mean, rstd};
+}
+
+std::vector<at::Tensor> layernorm_bwd(at::Tensor dy, at::Tensor x, at::Tensor w,
+                                      at::Tensor mean, at::Tensor rstd) {
+  TORCH_CHECK(dy.is_contiguous() && x.is_contiguous());
+  const int64_t Nd = x.size(-1);
+  const int64_t M = x.numel() / Nd;
+  TORCH_CHECK(Nd <= 16384, "ln_bwd: N too large for LDS accumulator");
+  auto wf = as_f32(w);
+  auto dx = at::empty_like(x);
+  auto dw = at::zeros({Nd}, x.options().dtype(at::kFloat));
+  auto db = at::zeros({Nd}, x.options().dtype(at::kFloat));
+  ln_bwd_launch(dtype_of(x), dy.data_ptr(), x.data_ptr(), wf.data_ptr<float>(),
+                mean.data_ptr<float>(), rstd.data_ptr<float>(), dx.data_ptr(),
+                dw.data_ptr<float>(), db.data_ptr<float>(), M, (int)Nd,
+                cur_stream());
+  return {dx, dw.to(w.scalar_type()), db.to(w.scalar_type())};
+}
+
+// ---------------------------------------------------------------- GEGLU
+at::Tensor geglu_fwd(at::Tensor x) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous());
+  const int64_t twoN = x.size(-1);
+  TORCH_CHECK(twoN % 2 == 0);
+  const int64_t N = twoN / 2;
+  TORCH_CHECK(N % 4 == 0, "geglu: inner dim must be divisible by 4");
+  const int64_t M = x.numel() / twoN;
+  auto sizes = x.sizes().vec();
+  sizes.back() = N;
+  auto y = at::empty(sizes, x.options());
+  geglu_fwd_launch(dtype_of(x), x.data_ptr(), y.data_ptr(), M, N, cur_stream());
+  return y;
+}
+
+at::Tensor geglu_bwd(at::Tensor dy, at::Tensor x) {
+  TORCH_CHECK(dy.is_contiguous() && x.is_contiguous());
+  const int64_t twoN = x.size(-1);
+  const int64_t N = twoN / 2;
+  const int64_t M = x.numel() / twoN;
+  auto dx = at::empty_like(x);
+  geglu_bwd_launch(dtype_of(x), dy.data_ptr(), x.data_ptr(), dx.data_ptr(), M, N,
+                   cur_stream());
+  return dx;
+}
+
+// ---------------------------------------------------------------- AdamW
+void adamw_step(at::Tensor p, at::Tensor g, at::Tensor m, at::Tensor v,
+                double lr, double beta1, double beta2, double eps, double wd,
+                int64_t step) {
+  TORCH_CHECK(p.is_cuda() && p.scalar_type() == at::kFloat,
+              "adamw: fp32 flat params expected");
+  TORCH_CHECK(p.is_contiguous() && g.is_contiguous() && m.is_contiguous() &&
+              v.is_contiguous());
+  adamw_launch(p.data_ptr<float>(), g.data_ptr<float>(), m.data_ptr<float>(),
+               v.data_ptr<float>(), p.numel(), (float)lr, (float)beta1,
+               (float)beta2, (float)eps, (float)wd, step, cur_stream());
+}
+
+// ------------------------------------------------------------- scheduler math
+static at::Tensor sched_common(int mode, at::Tensor a, at::Tensor b, at::Tensor ac,
+                               at::Tensor t) {
+  TORCH_CHECK(a.is_cuda() && a.is_contiguous() && b.is_contiguous());
+  TORCH_CHECK(ac.scalar_type() == at::kFloat && t.scalar_type() == at::kLong);
+  const int64_t B = t.size(0);
+  const int64_t per = a.numel() / B;
+  TORCH_CHECK(per % 4 == 0, "sched: per-sample numel must be divisible by 4");
+  auto out = at::empty_like(a);
+  sched_launch(dtype_of(a), mode, a.data_ptr(), b.data_ptr(), ac.data_ptr<float>(),
+               t.data_ptr<int64_t>(), out.data_ptr(), per, a.numel(), cur_stream());
+  return out;
+}
+
+at::Tensor add_noise(at::Tensor x0, at::Tensor noise, at::Tensor ac, at::Tensor t) {
+  return sched_common(0, x0, noise, ac, t);
+}
+
+at::Tensor get_velocity(at::Tensor x0, at::Tensor noise, at::Tensor ac, at::Tensor t) {
+  return sched_common(1, x0, noise, ac, t);
+}
+
+at::Tensor cfg_combine(at::Tensor eu, at::Tensor et, double scale) {
+  TORCH_CHECK(eu.is_cuda() && eu.is_contiguous() && et.is_contiguous());
+  TORCH_CHECK(eu.numel() % 4 == 0);
+  auto out = at::empty_like(eu);
+  cfg_launch(dtype_of(eu), eu.data_ptr(), et.data_ptr(), out.data_ptr(),
+             (float)scale, eu.numel(), cur_stream());
+  return out;
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("groupnorm_silu_fwd", &groupnorm_silu_fwd);
+  mod.def("groupnorm_silu_bwd", &groupnorm_silu_bwd);
+  mod.def("layernorm_fwd", &layernorm_fwd);
+  mod.def("layernorm_bwd", &layernorm_bwd);
+  mod.def("geglu_fwd", &geglu_fwd);
+  mod.def("geglu_bwd", &geglu_bwd);
+  mod.def("adamw_step", &adamw_step);
+  mod.def("add_noise", &add_noise);
+  mod.def("get_velocity", &get_velocity);
+  mod.def("cfg_combine", &cfg_combine);
+}

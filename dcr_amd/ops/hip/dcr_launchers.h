@@ -1,0 +1,41 @@
+// Host-side launcher API for the dcr_amd HIP kernels.
+// bindings.cpp (built by the host compiler) calls these; definitions live
+// in the .hip TUs compiled by hipcc for gfx950.
+#pragma once
+
+#include <hip/hip_runtime.h>
+
+namespace dcr {
+
+enum DType { DT_F32 = 0, DT_F16 = 1, DT_BF16 = 2 };
+
+// norms.hip
+void gn_fwd_launch(DType dt, const void* x, const float* w, const float* b,
+                   void* y, float* mean, float* rstd, int NG, int G, int Cg,
+                   int HW, float eps, bool silu, hipStream_t s);
+void gn_bwd_launch(DType dt, const void* dy, const void* x, const float* w,
+                   const float* b, const float* mean, const float* rstd,
+                   void* dx, float* dw, float* db, int NG, int G, int Cg,
+                   int HW, bool silu, hipStream_t s);
+void ln_fwd_launch(DType dt, const void* x, const float* w, const float* b,
+                   void* y, float* mean, float* rstd, long M, int N, float eps,
+                   hipStream_t s);
+void ln_bwd_launch(DType dt, const void* dy, const void* x, const float* w,
+                   const float* mean, const float* rstd, void* dx, float* dw,
+                   float* db, long M, int N, hipStream_t s);
+
+// elementwise.hip
+void geglu_fwd_launch(DType dt, const void* x, void* y, long M, long N,
+                      hipStream_t s);
+void geglu_bwd_launch(DType dt, const void* dy, const void* x, void* dx,
+                      long M, long N, hipStream_t s);
+void adamw_launch(float* p, const float* g, float* m, float* v, long n,
+                  float lr, float b1, float b2, float eps, float wd, long step,
+                  hipStream_t s);
+void sched_launch(DType dt, int mode, const void* x0, const void* noise,
+                  const float* ac, const long* t, void* out, long per_sample,
+                  long total, hipStream_t s);
+void cfg_launch(DType dt, const void* eu, const void* et, void* out, float s_,
+                long total, hipStream_t s);
+
+}  // namespace dcr

@@ -1,0 +1,423 @@
+// GroupNorm(+fused SiLU) and LayerNorm forward/backward for MI355X (gfx950).
+//
+// Design (HBM3E-bound ops):
+// * fp32 accumulation, bf16/f16/f32 IO, vec4 loads (dwordx4/dwordx2).
+// * GroupNorm: one workgroup per (n, group) — the group's data is one
+//   contiguous [C/G * H*W] segment of NCHW. Two-pass: block-reduce
+//   mean/var, then normalize (+SiLU fused into the same kernel, saving a
+//   full HBM round-trip per ResNet block).
+// * Backward recomputes z from (x, mean, rstd); per-channel dw/db go
+//   through an LDS accumulator, then one global atomicAdd per channel
+//   per block.
+// * LayerNorm: one wave per row, 4 waves per block.
+
+#include "dcr_common.h"
+
+using namespace dcr;
+
+// ===========================================================================
+// GroupNorm + SiLU forward
+// ===========================================================================
+template <typename T, bool SILU>
+__global__ void gn_fwd_kernel(const T* __restrict__ x, const float* __restrict__ w,
+                              const float* __restrict__ b, T* __restrict__ y,
+                              float* __restrict__ mean_out, float* __restrict__ rstd_out,
+                              int G, int Cg, int HW, float eps) {
+  const long base = (long)blockIdx.x * Cg * HW;
+  const int L = Cg * HW;
+  const int g = blockIdx.x % G;
+  const int tid = threadIdx.x;
+  const int nthr = blockDim.x;
+
+  float s1 = 0.f, s2 = 0.f;
+  if ((HW & 3) == 0) {
+    for (int i = tid * 4; i < L; i += nthr * 4) {
+      f32x4 v = load4<T>(x + base + i);
+      s1 += v.x + v.y + v.z + v.w;
+      s2 += v.x * v.x + v.y * v.y + v.z * v.z + v.w * v.w;
+    }
+  } else {
+    for (int i = tid; i < L; i += nthr) {
+      float v = to_f32<T>(x[base + i]);
+      s1 += v; s2 += v * v;
+    }
+  }
+  __shared__ float lds[2 * 16];
+  float2 s = block_reduce_sum2(s1, s2, lds);
+  const float m = s.x / L;
+  const float var = fmaxf(s.y / L - m * m, 0.f);
+  const float rs = rsqrtf(var + eps);
+  if (tid == 0) { mean_out[blockIdx.x] = m; rstd_out[blockIdx.x] = rs; }
+
+  const float* wg = w + (long)g * Cg;
+  const float* bg = b + (long)g * Cg;
+  if ((HW & 3) == 0) {
+    for (int i = tid * 4; i < L; i += nthr * 4) {
+      int c = i / HW;  // uniform across the 4 elements since HW % 4 == 0
+      float wc = wg[c], bc = bg[c];
+      f32x4 v = load4<T>(x + base + i);
+      f32x4 o;
+      o.x = (v.x - m) * rs * wc + bc;
+      o.y = (v.y - m) * rs * wc + bc;
+      o.z = (v.z - m) * rs * wc + bc;
+      o.w = (v.w - m) * rs * wc + bc;
+      if (SILU) { o.x = silu(o.x); o.y = silu(o.y); o.z = silu(o.z); o.w = silu(o.w); }
+      store4<T>(y + base + i, o);
+    }
+  } else {
+    for (int i = tid; i < L; i += nthr) {
+      int c = i / HW;
+      float z = (to_f32<T>(x[base + i]) - m) * rs * wg[c] + bg[c];
+      if (SILU) z = silu(z);
+      y[base + i] = from_f32<T>(z);
+    }
+  }
+}
+
+// ===========================================================================
+// GroupNorm + SiLU backward
+//   z  = yhat * w[c] + b[c],  yhat = (x - m) * rs
+//   dz = dy * (SILU ? dsilu(z) : 1)
+//   dx = rs * (w[c]*dz - mean(w*dz) - yhat * mean(w*dz*yhat))
+//   dw[c] = sum dz*yhat ; db[c] = sum dz   (fp32 atomics, LDS-staged)
+// ===========================================================================
+template <typename T, bool SILU>
+__global__ void gn_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x,
+                              const float* __restrict__ w, const float* __restrict__ b,
+                              const float* __restrict__ mean, const float* __restrict__ rstd,
+                              T* __restrict__ dx, float* __restrict__ dw,
+                              float* __restrict__ db, int G, int Cg, int HW) {
+  extern __shared__ float smem[];  // [2*Cg] channel partials + [32] reduce scratch
+  float* dw_l = smem;
+  float* db_l = smem + Cg;
+  float* red = smem + 2 * Cg;
+
+  const long base = (long)blockIdx.x * Cg * HW;
+  const int L = Cg * HW;
+  const int g = blockIdx.x % G;
+  const int tid = threadIdx.x;
+  const int nthr = blockDim.x;
+  const float m = mean[blockIdx.x];
+  const float rs = rstd[blockIdx.x];
+  const float* wg = w + (long)g * Cg;
+  const float* bg = b + (long)g * Cg;
+
+  for (int c = tid; c < Cg; c += nthr) { dw_l[c] = 0.f; db_l[c] = 0.f; }
+  __syncthreads();
+
+  const int lane = tid % DCR_WAVE;
+  const bool wave_uniform_c = (HW % (DCR_WAVE * 4) == 0);
+
+  float s1 = 0.f, s2 = 0.f;
+  if ((HW & 3) == 0) {
+    for (int i = tid * 4; i < L; i += nthr * 4) {
+      int c = i / HW;
+      float wc = wg[c], bc = bg[c];
+      f32x4 xv = load4<T>(x + base + i);
+      f32x4 gv = load4<T>(dy + base + i);
+      float dwp = 0.f, dbp = 0.f;
+#pragma unroll
+      for (int k = 0; k < 4; ++k) {
+        float xe = (&xv.x)[k], ge = (&gv.x)[k];
+        float yh = (xe - m) * rs;
+        float dz = ge;
+        if (SILU) dz *= dsilu(yh * wc + bc);
+        float gx = dz * wc;
+        s1 += gx;
+        s2 += gx * yh;
+        dwp += dz * yh;
+        dbp += dz;
+      }
+      if (wave_uniform_c) {
+        float2 ws = wave_reduce_sum2(dwp, dbp);
+        if (lane == 0) { atomicAdd(&dw_l[c], ws.x); atomicAdd(&db_l[c], ws.y); }
+      } else {
+        atomicAdd(&dw_l[c], dwp);
+        atomicAdd(&db_l[c], dbp);
+      }
+    }
+  } else {
+    for (int i = tid; i < L; i += nthr) {
+      int c = i / HW;
+      float wc = wg[c], bc = bg[c];
+      float xe = to_f32<T>(x[base + i]);
+      float ge = to_f32<T>(dy[base + i]);
+      float yh = (xe - m) * rs;
+      float dz = ge;
+      if (SILU) dz *= dsilu(yh * wc + bc);
+      float gx = dz * wc;
+      s1 += gx;
+      s2 += gx * yh;
+      atomicAdd(&dw_l[c], dz * yh);
+      atomicAdd(&db_l[c], dz);
+    }
+  }
+  __syncthreads();
+  float2 s = block_reduce_sum2(s1, s2, red);
+  const float m1 = s.x / L;
+  const float m2 = s.y / L;
+
+  if ((HW & 3) == 0) {
+    for (int i = tid * 4; i < L; i += nthr * 4) {
+      int c = i / HW;
+      float wc = wg[c], bc = bg[c];
+      f32x4 xv = load4<T>(x + base + i);
+      f32x4 gv = load4<T>(dy + base + i);
+      f32x4 o;
+#pragma unroll
+      for (int k = 0; k < 4; ++k) {
+        float xe = (&xv.x)[k], ge = (&gv.x)[k];
+        float yh = (xe - m) * rs;
+        float dz = ge;
+        if (SILU) dz *= dsilu(yh * wc + bc);
+        float gx = dz * wc;
+        (&o.x)[k] = rs * (gx - m1 - yh * m2);
+      }
+      store4<T>(dx + base + i, o);
+    }
+  } else {
+    for (int i = tid; i < L; i += nthr) {
+      int c = i / HW;
+      float wc = wg[c], bc = bg[c];
+      float xe = to_f32<T>(x[base + i]);
+      float ge = to_f32<T>(dy[base + i]);
+      float yh = (xe - m) * rs;
+      float dz = ge;
+      if (SILU) dz *= dsilu(yh * wc + bc);
+      float gx = dz * wc;
+      dx[base + i] = from_f32<T>(rs * (gx - m1 - yh * m2));
+    }
+  }
+  __syncthreads();
+  for (int c = tid; c < Cg; c += nthr) {
+    atomicAdd(&dw[(long)g * Cg + c], dw_l[c]);
+    atomicAdd(&db[(long)g * Cg + c], db_l[c]);
+  }
+}
+
+// ===========================================================================
+// LayerNorm forward: one wave per row
+// ===========================================================================
+template <typename T>
+__global__ void ln_fwd_kernel(const T* __restrict__ x, const float* __restrict__ w,
+                              const float* __restrict__ b, T* __restrict__ y,
+                              float* __restrict__ mean_out, float* __restrict__ rstd_out,
+                              long M, int N, float eps) {
+  const int wid = threadIdx.x / DCR_WAVE;
+  const int lane = threadIdx.x % DCR_WAVE;
+  const long row = (long)blockIdx.x * (blockDim.x / DCR_WAVE) + wid;
+  if (row >= M) return;
+  const T* xr = x + row * N;
+  T* yr = y + row * N;
+
+  float s1 = 0.f, s2 = 0.f;
+  if ((N & 3) == 0) {
+    for (int j = lane * 4; j < N; j += DCR_WAVE * 4) {
+      f32x4 v = load4<T>(xr + j);
+      s1 += v.x + v.y + v.z + v.w;
+      s2 += v.x * v.x + v.y * v.y + v.z * v.z + v.w * v.w;
+    }
+  } else {
+    for (int j = lane; j < N; j += DCR_WAVE) {
+      float v = to_f32<T>(xr[j]);
+      s1 += v; s2 += v * v;
+    }
+  }
+  float2 s = wave_reduce_sum2(s1, s2);
+  const float m = s.x / N;
+  const float var = fmaxf(s.y / N - m * m, 0.f);
+  const float rs = rsqrtf(var + eps);
+  if (lane == 0) { mean_out[row] = m; rstd_out[row] = rs; }
+
+  if ((N & 3) == 0) {
+    for (int j = lane * 4; j < N; j += DCR_WAVE * 4) {
+      f32x4 v = load4<T>(xr + j);
+      f32x4 wv = load4<float>(w + j);
+      f32x4 bv = load4<float>(b + j);
+      f32x4 o;
+      o.x = (v.x - m) * rs * wv.x + bv.x;
+      o.y = (v.y - m) * rs * wv.y + bv.y;
+      o.z = (v.z - m) * rs * wv.z + bv.z;
+      o.w = (v.w - m) * rs * wv.w + bv.w;
+      store4<T>(yr + j, o);
+    }
+  } else {
+    for (int j = lane; j < N; j += DCR_WAVE) {
+      yr[j] = from_f32<T>((to_f32<T>(xr[j]) - m) * rs * w[j] + b[j]);
+    }
+  }
+}
+
+// ===========================================================================
+// LayerNorm backward: one wave per row; dw/db via LDS[2N] then global atomics
+// ===========================================================================
+template <typename T>
+__global__ void ln_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x,
+                              const float* __restrict__ w,
+                              const float* __restrict__ mean, const float* __restrict__ rstd,
+                              T* __restrict__ dx, float* __restrict__ dw,
+                              float* __restrict__ db, long M, int N) {
+  extern __shared__ float smem[];  // [2N]
+  float* dw_l = smem;
+  float* db_l = smem + N;
+  const int wid = threadIdx.x / DCR_WAVE;
+  const int lane = threadIdx.x % DCR_WAVE;
+  const int waves = blockDim.x / DCR_WAVE;
+
+  for (int j = threadIdx.x; j < 2 * N; j += blockDim.x) smem[j] = 0.f;
+  __syncthreads();
+
+  const long row = (long)blockIdx.x * waves + wid;
+  if (row < M) {
+    const T* xr = x + row * N;
+    const T* gr = dy + row * N;
+    T* dr = dx + row * N;
+    const float m = mean[row];
+    const float rs = rstd[row];
+
+    float s1 = 0.f, s2 = 0.f;
+    for (int j = lane; j < N; j += DCR_WAVE) {
+      float g = to_f32<T>(gr[j]);
+      float yh = (to_f32<T>(xr[j]) - m) * rs;
+      float gw = g * w[j];
+      s1 += gw;
+      s2 += gw * yh;
+      atomicAdd(&dw_l[j], g * yh);
+      atomicAdd(&db_l[j], g);
+    }
+    float2 s = wave_reduce_sum2(s1, s2);
+    const float m1 = s.x / N;
+    const float m2 = s.y / N;
+    for (int j = lane; j < N; j += DCR_WAVE) {
+      float g = to_f32<T>(gr[j]);
+      float yh = (to_f32<T>(xr[j]) - m) * rs;
+      float gw = g * w[j];
+      dr[j] = from_f32<T>(rs * (gw - m1 - yh * m2));
+    }
+  }
+  __syncthreads();
+  for (int j = threadIdx.x; j < N; j += blockDim.x) {
+    atomicAdd(&dw[j], dw_l[j]);
+    atomicAdd(&db[j], db_l[j]);
+  }
+}
+
+// ===========================================================================
+// C++ launchers (instantiated per dtype; called from bindings.cpp)
+// ===========================================================================
+#define DCR_INST_T(T)                                                                   \
+  template __global__ void gn_fwd_kernel<T, true>(const T*, const float*, const float*, \
+      T*, float*, float*, int, int, int, float);                                        \
+  template __global__ void gn_fwd_kernel<T, false>(const T*, const float*, const float*,\
+      T*, float*, float*, int, int, int, float);                                        \
+  template __global__ void gn_bwd_kernel<T, true>(const T*, const T*, const float*,     \
+      const float*, const float*, const float*, T*, float*, float*, int, int, int);     \
+  template __global__ void gn_bwd_kernel<T, false>(const T*, const T*, const float*,    \
+      const float*, const float*, const float*, T*, float*, float*, int, int, int);     \
+  template __global__ void ln_fwd_kernel<T>(const T*, const float*, const float*, T*,   \
+      float*, float*, long, int, float);                                                \
+  template __global__ void ln_bwd_kernel<T>(const T*, const T*, const float*,           \
+      const float*, const float*, T*, float*, float*, long, int);
+
+DCR_INST_T(float)
+DCR_INST_T(__hip_bfloat16)
+DCR_INST_T(__half)
+
+// ===========================================================================
+// Host launchers
+// ===========================================================================
+#include "dcr_launchers.h"
+
+namespace dcr {
+
+template <typename T>
+static void gn_fwd_t(const void* x, const float* w, const float* b, void* y,
+                     float* mean, float* rstd, int NG, int G, int Cg, int HW,
+                     float eps, bool silu, hipStream_t s) {
+  dim3 grid(NG), block(256);
+  if (silu)
+    hipLaunchKernelGGL((gn_fwd_kernel<T, true>), grid, block, 0, s,
+                       (const T*)x, w, b, (T*)y, mean, rstd, G, Cg, HW, eps);
+  else
+    hipLaunchKernelGGL((gn_fwd_kernel<T, false>), grid, block, 0, s,
+                       (const T*)x, w, b, (T*)y, mean, rstd, G, Cg, HW, eps);
+}
+
+void gn_fwd_launch(DType dt, const void* x, const float* w, const float* b,
+                   void* y, float* mean, float* rstd, int NG, int G, int Cg,
+                   int HW, float eps, bool silu, hipStream_t s) {
+  switch (dt) {
+    case DT_F32: gn_fwd_t<float>(x, w, b, y, mean, rstd, NG, G, Cg, HW, eps, silu, s); break;
+    case DT_F16: gn_fwd_t<__half>(x, w, b, y, mean, rstd, NG, G, Cg, HW, eps, silu, s); break;
+    case DT_BF16: gn_fwd_t<__hip_bfloat16>(x, w, b, y, mean, rstd, NG, G, Cg, HW, eps, silu, s); break;
+  }
+}
+
+template <typename T>
+static void gn_bwd_t(const void* dy, const void* x, const float* w, const float* b,
+                     const float* mean, const float* rstd, void* dx, float* dw,
+                     float* db, int NG, int G, int Cg, int HW, bool silu,
+                     hipStream_t s) {
+  dim3 grid(NG), block(256);
+  size_t lds = (2 * Cg + 32) * sizeof(float);
+  if (silu)
+    hipLaunchKernelGGL((gn_bwd_kernel<T, true>), grid, block, lds, s,
+                       (const T*)dy, (const T*)x, w, b, mean, rstd, (T*)dx, dw, db, G, Cg, HW);
+  else
+    hipLaunchKernelGGL((gn_bwd_kernel<T, false>), grid, block, lds, s,
+                       (const T*)dy, (const T*)x, w, b, mean, rstd, (T*)dx, dw, db, G, Cg, HW);
+}
+
+void gn_bwd_launch(DType dt, const void* dy, const void* x, const float* w,
+                   const float* b, const float* mean, const float* rstd,
+                   void* dx, float* dw, float* db, int NG, int G, int Cg,
+                   int HW, bool silu, hipStream_t s) {
+  switch (dt) {
+    case DT_F32: gn_bwd_t<float>(dy, x, w, b, mean, rstd, dx, dw, db, NG, G, Cg, HW, silu, s); break;
+    case DT_F16: gn_bwd_t<__half>(dy, x, w, b, mean, rstd, dx, dw, db, NG, G, Cg, HW, silu, s); break;
+    case DT_BF16: gn_bwd_t<__hip_bfloat16>(dy, x, w, b, mean, rstd, dx, dw, db, NG, G, Cg, HW, silu, s); break;
+  }
+}
+
+template <typename T>
+static void ln_fwd_t(const void* x, const float* w, const float* b, void* y,
+                     float* mean, float* rstd, long M, int N, float eps,
+                     hipStream_t s) {
+  const int waves = 4;
+  dim3 grid((M + waves - 1) / waves), block(waves * DCR_WAVE);
+  hipLaunchKernelGGL((ln_fwd_kernel<T>), grid, block, 0, s, (const T*)x, w, b,
+                     (T*)y, mean, rstd, M, N, eps);
+}
+
+void ln_fwd_launch(DType dt, const void* x, const float* w, const float* b,
+                   void* y, float* mean, float* rstd, long M, int N, float eps,
+                   hipStream_t s) {
+  switch (dt) {
+    case DT_F32: ln_fwd_t<float>(x, w, b, y, mean, rstd, M, N, eps, s); break;
+    case DT_F16: ln_fwd_t<__half>(x, w, b, y, mean, rstd, M, N, eps, s); break;
+    case DT_BF16: ln_fwd_t<__hip_bfloat16>(x, w, b, y, mean, rstd, M, N, eps, s); break;
+  }
+}
+
+template <typename T>
+static void ln_bwd_t(const void* dy, const void* x, const float* w,
+                     const float* mean, const float* rstd, void* dx, float* dw,
+                     float* db, long M, int N, hipStream_t s) {
+  const int waves = 4;
+  dim3 grid((M + waves - 1) / waves), block(waves * DCR_WAVE);
+  size_t lds = 2 * (size_t)N * sizeof(float);
+  hipLaunchKernelGGL((ln_bwd_kernel<T>), grid, block, lds, s, (const T*)dy,
+                     (const T*)x, w, mean, rstd, (T*)dx, dw, db, M, N);
+}
+
+void ln_bwd_launch(DType dt, const void* dy, const void* x, const float* w,
+                   const float* mean, const float* rstd, void* dx, float* dw,
+                   float* db, long M, int N, hipStream_t s) {
+  switch (dt) {
+    case DT_F32: ln_bwd_t<float>(dy, x, w, mean, rstd, dx, dw, db, M, N, s); break;
+    case DT_F16: ln_bwd_t<__half>(dy, x, w, mean, rstd, dx, dw, db, M, N, s); break;
+    case DT_BF16: ln_bwd_t<__hip_bfloat16>(dy, x, w, mean, rstd, dx, dw, db, M, N, s); break;
+  }
+}
+
+}  // namespace dcr

@@ -1,0 +1,134 @@
+"""Data-layer tests: ObjectAttributeDataset caption/duplication semantics
+(/root/reference/datasets.py), collate, tokenizer."""
+import json
+import pickle
+
+import numpy as np
+import pytest
+import torch
+from PIL import Image
+
+from dcr_amd.data import (HashTokenizer, ObjectAttributeDataset, SynthDataset,
+                          SyntheticImageDataset, collate_fn, get_classnames,
+                          insert_rand_word)
+
+
+@pytest.fixture
+def image_folder(tmp_path):
+    root = tmp_path / "imagenette"
+    prompts = {}
+    for cls in ["church", "tench"]:
+        d = root / cls
+        d.mkdir(parents=True)
+        for i in range(4):
+            f = d / f"{i}.png"
+            Image.new("RGB", (80, 70), color=(i * 30, 100, 50)).save(f)
+            prompts[str(f)] = [f"a photo of a {cls} number {i}",
+                               f"another caption for {cls} {i}"]
+    pj = tmp_path / "blip.json"
+    pj.write_text(json.dumps(prompts))
+    return root, pj
+
+
+def test_classnames():
+    assert get_classnames("/data/imagenette_2class/train") == ["church", "garbage truck"]
+    assert len(get_classnames("/data/imagenette/train")) == 10
+
+
+def test_tokenizer_deterministic_and_shape():
+    tok = HashTokenizer()
+    out = tok(["hello world", "hello world"], return_tensors="pt").input_ids
+    assert out.shape == (2, 77)
+    assert torch.equal(out[0], out[1])
+    assert out[0, 0] == 49406  # BOS
+    assert (out[0] == 49407).any()  # EOS + padding
+
+
+def test_dataset_classlevel(image_folder):
+    root, pj = image_folder
+    tok = HashTokenizer()
+    ds = ObjectAttributeDataset(str(root), tok, class_prompt="classlevel", size=64)
+    ex = ds[0]
+    assert ex["instance_images"].shape == (3, 64, 64)
+    assert ex["instance_images"].min() >= -1.001 and ex["instance_images"].max() <= 1.001
+    assert ex["instance_prompt_ids"].shape == (1, 77)
+
+
+def test_dataset_instancelevel_blip(image_folder):
+    root, pj = image_folder
+    tok = HashTokenizer()
+    ds = ObjectAttributeDataset(str(root), tok, class_prompt="instancelevel_blip",
+                                size=64, prompt_json=str(pj))
+    ex = ds[0]
+    path0 = ds.samples[0][0]
+    expected = tok(ds.prompts[path0][0], truncation=True, padding="max_length",
+                   max_length=77, return_tensors="pt").input_ids
+    assert torch.equal(ex["instance_prompt_ids"], expected)
+
+
+def test_dup_weights_pickled_and_cached(image_folder):
+    root, pj = image_folder
+    tok = HashTokenizer()
+    np.random.seed(0)
+    ds = ObjectAttributeDataset(str(root), tok, class_prompt="classlevel", size=64,
+                                duplication="dup_both", weight_pc=0.25,
+                                dup_weight=5.0, seed=7)
+    w = ds.samplingweights
+    assert len(w) == len(ds.samples) == 8
+    assert sum(1 for x in w if x > 1) == 2  # 25% of 8
+    # cache file exists with reference naming (datasets.py:77)
+    cache = root / "weights_0.25_5.0_seed7.pickle"
+    assert cache.exists()
+    with open(cache, "rb") as fh:
+        assert pickle.load(fh) == w
+    # second construction loads the cache (even with different rng state)
+    np.random.seed(99)
+    ds2 = ObjectAttributeDataset(str(root), tok, class_prompt="classlevel", size=64,
+                                 duplication="dup_both", weight_pc=0.25,
+                                 dup_weight=5.0, seed=7)
+    assert ds2.samplingweights == w
+
+
+def test_trainspecial_randrepl(image_folder):
+    root, pj = image_folder
+    tok = HashTokenizer()
+    ds = ObjectAttributeDataset(str(root), tok, class_prompt="instancelevel_blip",
+                                size=64, prompt_json=str(pj),
+                                trainspecial="randrepl", trainspecial_prob=1.0)
+    ex = ds[0]  # always replaced with 4 random-token caption
+    assert ex["instance_prompt_ids"].shape == (1, 77)
+
+
+def test_insert_rand_word():
+    s = insert_rand_word("a b c", "X")
+    assert sorted(s.split(" ")) == ["X", "a", "b", "c"]
+
+
+def test_collate(image_folder):
+    root, pj = image_folder
+    tok = HashTokenizer()
+    ds = ObjectAttributeDataset(str(root), tok, class_prompt="nolevel", size=64)
+    batch = collate_fn([ds[0], ds[1]])
+    assert batch["pixel_values"].shape == (2, 3, 64, 64)
+    assert batch["pixel_values"].dtype == torch.float32
+    assert batch["input_ids"].shape == (2, 77)
+
+
+def test_synth_dataset_natural_order(tmp_path):
+    d = tmp_path / "gens"
+    d.mkdir()
+    for i in [0, 1, 2, 10, 11]:
+        Image.new("RGB", (32, 32)).save(d / f"{i}.png")
+    (d / "prompts.txt").write_text("\n".join(f"p{i}" for i in range(5)))
+    ds = SynthDataset(str(d), size=32, with_prompts=True)
+    names = [f.name for f in ds.files]
+    assert names == ["0.png", "1.png", "2.png", "10.png", "11.png"]
+    t, idx, p = ds[3]
+    assert t.shape == (3, 32, 32) and idx == 3 and p == "p3"
+
+
+def test_synthetic_dataset_deterministic():
+    tok = HashTokenizer()
+    ds = SyntheticImageDataset(4, size=32, tokenizer=tok, seed=5)
+    a, b = ds[2], ds[2]
+    assert torch.equal(a["instance_images"], b["instance_images"])

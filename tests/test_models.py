@@ -1,0 +1,99 @@
+"""Model architecture + checkpoint-layout tests (CPU).
+
+Parity targets: diffusers UNet2DConditionModel/AutoencoderKL naming and
+the checkpoint_{step}/ layout."""
+import torch
+
+from dcr_amd.models import (AutoencoderKL, CLIPTextModel, CLIPTextConfig,
+                            UNet2DConditionModel, UNetConfig, VAEConfig)
+
+
+def test_sd21_unet_param_count():
+    # diffusers stabilityai/stable-diffusion-2-1 unet == 865,910,724 params
+    unet = UNet2DConditionModel(UNetConfig.sd21())
+    assert sum(p.numel() for p in unet.parameters()) == 865_910_724
+
+
+def test_sd_vae_param_count():
+    vae = AutoencoderKL(VAEConfig.sd())
+    assert sum(p.numel() for p in vae.parameters()) == 83_653_863
+
+
+def test_unet_forward_shapes():
+    torch.manual_seed(0)
+    unet = UNet2DConditionModel(UNetConfig.tiny())
+    x = torch.randn(2, 4, 8, 8)
+    t = torch.randint(0, 1000, (2,))
+    ehs = torch.randn(2, 7, 32)
+    out = unet(x, t, ehs)
+    assert out.shape == (2, 4, 8, 8)
+    assert torch.isfinite(out).all()
+
+
+def test_unet_backward():
+    unet = UNet2DConditionModel(UNetConfig.tiny())
+    out = unet(torch.randn(1, 4, 8, 8), torch.tensor([3]), torch.randn(1, 7, 32))
+    out.mean().backward()
+    grads = [p.grad for p in unet.parameters() if p.requires_grad]
+    assert all(g is not None for g in grads)
+    assert all(torch.isfinite(g).all() for g in grads)
+
+
+def test_diffusers_state_dict_naming():
+    unet = UNet2DConditionModel(UNetConfig.sd21())
+    keys = set(unet.state_dict().keys())
+    expected = [
+        "conv_in.weight",
+        "time_embedding.linear_1.weight",
+        "down_blocks.0.resnets.0.norm1.weight",
+        "down_blocks.0.resnets.0.time_emb_proj.weight",
+        "down_blocks.0.attentions.0.transformer_blocks.0.attn1.to_q.weight",
+        "down_blocks.0.attentions.0.transformer_blocks.0.attn2.to_k.weight",
+        "down_blocks.0.attentions.0.transformer_blocks.0.ff.net.0.proj.weight",
+        "down_blocks.0.attentions.0.proj_in.weight",
+        "down_blocks.0.downsamplers.0.conv.weight",
+        "down_blocks.1.resnets.0.conv_shortcut.weight",
+        "mid_block.attentions.0.transformer_blocks.0.attn2.to_out.0.weight",
+        "up_blocks.1.upsamplers.0.conv.weight",
+        "conv_norm_out.weight",
+        "conv_out.bias",
+    ]
+    for k in expected:
+        assert k in keys, f"missing diffusers key {k}"
+
+
+def test_vae_roundtrip_shapes():
+    vae = AutoencoderKL(VAEConfig.tiny())
+    x = torch.randn(2, 3, 32, 32)
+    dist = vae.encode(x).latent_dist
+    z = dist.sample()
+    assert z.shape == (2, 4, 4, 4)
+    y = vae.decode(z).sample
+    assert y.shape == x.shape
+
+
+def test_clip_text_causal():
+    """Causal mask: token t must not see tokens > t."""
+    torch.manual_seed(0)
+    cfg = CLIPTextConfig.tiny()
+    te = CLIPTextModel(cfg).eval()
+    ids = torch.randint(1, 49406, (1, 10))
+    ids2 = ids.clone()
+    ids2[0, -1] = (ids2[0, -1] + 7) % 49406 + 1
+    with torch.no_grad():
+        h1 = te(ids).last_hidden_state
+        h2 = te(ids2).last_hidden_state
+    assert torch.allclose(h1[0, :9], h2[0, :9], atol=1e-5)
+    assert not torch.allclose(h1[0, 9], h2[0, 9], atol=1e-5)
+
+
+def test_checkpoint_roundtrip(tmp_path):
+    unet = UNet2DConditionModel(UNetConfig.tiny())
+    unet.save_pretrained(tmp_path / "unet")
+    assert (tmp_path / "unet" / "config.json").exists()
+    assert (tmp_path / "unet" / "diffusion_pytorch_model.safetensors").exists()
+    unet2 = UNet2DConditionModel.from_pretrained(tmp_path / "unet")
+    sd1, sd2 = unet.state_dict(), unet2.state_dict()
+    assert sd1.keys() == sd2.keys()
+    for k in sd1:
+        assert torch.equal(sd1[k], sd2[k])

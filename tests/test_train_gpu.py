@@ -1,0 +1,47 @@
+"""GPU end-to-end: tiny + flagship train step on MI355X, HIP path active."""
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+def test_tiny_train_step_gpu(tmp_path):
+    from dcr_amd.train import TrainConfig, Trainer
+    cfg = TrainConfig(model_size="tiny", synthetic_data=True, synthetic_size=8,
+                      resolution=64, train_batch_size=2, mixed_precision="bf16",
+                      dataloader_num_workers=0, max_train_steps=4, seed=0,
+                      learning_rate=1e-4, lr_warmup_steps=1,
+                      output_dir=str(tmp_path / "out"))
+    tr = Trainer(cfg, device=torch.device("cuda", 0))
+    batch = next(iter(tr.dataloader))
+    losses = [tr.train_step(batch).item() for _ in range(6)]
+    assert all(l == l for l in losses)
+    assert losses[-1] < losses[0], losses
+
+
+def test_sd21_train_step_gpu(tmp_path):
+    """One full-size flagship step must run and produce a finite loss."""
+    from dcr_amd.train import TrainConfig, Trainer
+    cfg = TrainConfig(model_size="sd21", synthetic_data=True, synthetic_size=4,
+                      resolution=256, train_batch_size=2, mixed_precision="bf16",
+                      dataloader_num_workers=0, max_train_steps=2, seed=0,
+                      output_dir=str(tmp_path / "out"))
+    tr = Trainer(cfg, device=torch.device("cuda", 0))
+    batch = next(iter(tr.dataloader))
+    loss = tr.train_step(batch)
+    torch.cuda.synchronize()
+    assert torch.isfinite(loss)
+
+
+def test_unet_fwd_hip_matches_cpu_reference():
+    """Tiny UNet forward on GPU (HIP kernels) vs CPU (torch fallback)."""
+    from dcr_amd.models import UNet2DConditionModel, UNetConfig
+    torch.manual_seed(0)
+    unet = UNet2DConditionModel(UNetConfig.tiny()).eval()
+    x = torch.randn(2, 4, 8, 8)
+    t = torch.tensor([17, 503])
+    ehs = torch.randn(2, 7, 32)
+    with torch.no_grad():
+        ref = unet(x, t, ehs)
+        gpu = unet.cuda()(x.cuda(), t.cuda(), ehs.cuda())
+    assert (gpu.cpu() - ref).abs().max().item() < 1e-3

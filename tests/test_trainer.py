@@ -1,0 +1,119 @@
+"""Trainer loop tests (CPU, tiny config): loss goes down, accumulation,
+checkpoint/resume, output-dir mangling (/root/reference/diff_train.py)."""
+import json
+
+import pytest
+import torch
+
+from dcr_amd.train import TrainConfig, Trainer, get_lr, mangle_output_dir, validate
+
+
+def tiny_cfg(tmp_path, **kw):
+    base = dict(model_size="tiny", synthetic_data=True, synthetic_size=8,
+                resolution=64, train_batch_size=2, mixed_precision="no",
+                dataloader_num_workers=0, max_train_steps=4, seed=0,
+                learning_rate=1e-4, lr_warmup_steps=1,
+                output_dir=str(tmp_path / "out"))
+    base.update(kw)
+    return TrainConfig(**base)
+
+
+def test_loss_decreases(tmp_path):
+    tr = Trainer(tiny_cfg(tmp_path))
+    batch = next(iter(tr.dataloader))
+    losses = [tr.train_step(batch).item() for _ in range(8)]
+    assert losses[-1] < losses[0], losses
+    assert all(l == l for l in losses)  # finite
+
+
+def test_gradient_accumulation_equivalence(tmp_path):
+    """2 micro-steps with accum == 1 step on concatenated batch (same data)."""
+    torch.manual_seed(0)
+    cfg = tiny_cfg(tmp_path)
+    tr1 = Trainer(cfg, device=torch.device("cpu"))
+    tr2 = Trainer(cfg, device=torch.device("cpu"))
+    tr2.optimizer.flat_param.copy_(tr1.optimizer.flat_param)
+
+    b = next(iter(tr1.dataloader))
+    # deterministic comparison needs fixed noise/timesteps: just check the
+    # mechanics — non-sync step must not change params, sync step must.
+    p0 = tr1.optimizer.flat_param.clone()
+    tr1.train_step(b, sync_gradients=False)
+    assert torch.equal(tr1.optimizer.flat_param, p0)
+    g_half = tr1.optimizer.flat_grad.clone()
+    tr1.train_step(b, sync_gradients=True)
+    assert not torch.equal(tr1.optimizer.flat_param, p0)
+    assert not torch.equal(g_half, torch.zeros_like(g_half))
+    del tr2
+
+
+def test_resume_roundtrip(tmp_path):
+    cfg = tiny_cfg(tmp_path)
+    tr = Trainer(cfg)
+    b = next(iter(tr.dataloader))
+    for _ in range(2):
+        tr.train_step(b)
+    tr.save_checkpoint(tmp_path / "ckpt")
+
+    tr2 = Trainer(cfg)
+    tr2.load_checkpoint(tmp_path / "ckpt")
+    assert tr2.global_step == 2
+    assert torch.allclose(tr2.optimizer.flat_param, tr.optimizer.flat_param)
+    assert torch.allclose(tr2.optimizer.exp_avg, tr.optimizer.exp_avg)
+    # training continues without blowup
+    l = tr2.train_step(b)
+    assert torch.isfinite(l)
+
+
+def test_mitigations_run(tmp_path):
+    tr = Trainer(tiny_cfg(tmp_path, rand_noise_lam=0.1, mixup_noise_lam=0.2))
+    b = next(iter(tr.dataloader))
+    assert torch.isfinite(tr.train_step(b))
+
+
+def test_v_prediction(tmp_path):
+    tr = Trainer(tiny_cfg(tmp_path, prediction_type="v_prediction"))
+    b = next(iter(tr.dataloader))
+    assert torch.isfinite(tr.train_step(b))
+
+
+def test_output_dir_mangling():
+    cfg = TrainConfig(output_dir="m", class_prompt="instancelevel_blip",
+                      duplication="dup_both", weight_pc=0.05, dup_weight=5.0)
+    assert mangle_output_dir(cfg) == "m_instancelevel_blip_dup_both_0.05_5.0"
+    cfg2 = TrainConfig(output_dir="m", class_prompt="nolevel", duplication="nodup",
+                       unet_from_scratch="yes", rand_noise_lam=0.1)
+    assert mangle_output_dir(cfg2) == "m_nolevel_nodup_unetfromscr_glam0.1"
+    cfg3 = TrainConfig(output_dir="m", trainspecial="allcaps",
+                       trainspecial_prob=0.3, trainsubset=500)
+    assert mangle_output_dir(cfg3) == "m_500subset_instancelevel_blip_nodup_special_allcaps_0.3"
+
+
+def test_validate_asserts():
+    with pytest.raises(AssertionError):
+        validate(TrainConfig(duplication="dup_image", class_prompt="instancelevel_ogcap"))
+    with pytest.raises(Exception):
+        validate(TrainConfig(trainspecial="allcaps", class_prompt="nolevel"))
+
+
+def test_lr_schedule():
+    cfg = TrainConfig(learning_rate=1e-3, lr_warmup_steps=10,
+                      lr_scheduler="constant_with_warmup", max_train_steps=100)
+    assert get_lr(cfg, 0) == pytest.approx(1e-4)
+    assert get_lr(cfg, 9) == pytest.approx(1e-3)
+    assert get_lr(cfg, 50) == pytest.approx(1e-3)
+
+
+def test_fit_writes_checkpoint_layout(tmp_path):
+    cfg = tiny_cfg(tmp_path, max_train_steps=2, modelsavesteps=2, save_steps=1000,
+                   log_every=1)
+    tr = Trainer(cfg)
+    tr.fit()
+    out = tmp_path / "out"
+    assert (out / "checkpoint_2" / "unet" / "diffusion_pytorch_model.safetensors").exists()
+    assert (out / "checkpoint" / "model_index.json").exists()
+    assert (out / "checkpoint" / "state.pt").exists()
+    # jsonl tracker wrote loss/lr events (wandb schema parity)
+    logf = out / "diffrep_ft_log.jsonl"
+    recs = [json.loads(l) for l in logf.read_text().splitlines()]
+    assert any("loss" in r and "lr" in r for r in recs)
