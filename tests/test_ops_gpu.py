@@ -11,6 +11,15 @@ import torch.nn.functional as F
 pytestmark = pytest.mark.gpu
 
 
+def _close(out, ref, rtol):
+    """max-norm comparison relative to the reference's max magnitude —
+    bf16 outputs quantize at ~0.8% of magnitude, so absolute tolerances
+    are shape-dependent; relative ones are not."""
+    scale = ref.abs().max().item() + 1e-6
+    err = (out.float() - ref.float()).abs().max().item()
+    assert err < rtol * scale, f"err={err:.4g} scale={scale:.4g} rtol={rtol}"
+
+
 @pytest.fixture(scope="module")
 def ext():
     from dcr_amd import ops
@@ -40,8 +49,7 @@ def test_groupnorm_silu_fwd(ext, dtype):
         b = torch.randn(C, device="cuda") * 0.1
         y, mean, rstd = ext.groupnorm_silu_fwd(x, w, b, G, 1e-5, True)
         ref = F.silu(F.group_norm(x.float(), G, w, b, 1e-5))
-        tol = 1e-5 if dtype == torch.float32 else 2e-2
-        assert (y.float() - ref).abs().max().item() < tol, (N, C, H, W, G, dtype)
+        _close(y, ref, 1e-5 if dtype == torch.float32 else 1e-2)
 
 
 @pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
@@ -85,13 +93,11 @@ def test_layernorm_fwd_bwd(ext, shape, dtype):
     y, mean, rstd = ext.layernorm_fwd(x, w.detach(), b.detach(), 1e-5)
     xr = x.float().detach().requires_grad_(True)
     ref = F.layer_norm(xr, (N,), w, b, 1e-5)
-    tol = 1e-5 if dtype == torch.float32 else 2e-2
-    assert (y.float() - ref).abs().max().item() < tol
+    _close(y, ref, 1e-5 if dtype == torch.float32 else 1e-2)
 
     ref.backward(dy.float())
     dx, dw, db = ext.layernorm_bwd(dy.contiguous(), x, w.detach(), mean, rstd)
-    atol_x = 1e-4 if dtype == torch.float32 else 5e-2
-    assert (dx.float() - xr.grad).abs().max().item() < atol_x
+    _close(dx, xr.grad, 1e-4 if dtype == torch.float32 else 3e-2)
     rel_w = (dw.float() - w.grad).abs().max() / (w.grad.abs().max() + 1e-6)
     assert rel_w.item() < (1e-4 if dtype == torch.float32 else 3e-2)
 
@@ -106,15 +112,13 @@ def test_geglu_fwd_bwd(ext, dtype):
     y = ext.geglu_fwd(x)
     a, g = x.float().chunk(2, dim=-1)
     ref = a * F.gelu(g)
-    tol = 1e-5 if dtype == torch.float32 else 2e-2
-    assert (y.float() - ref).abs().max().item() < tol
+    _close(y, ref, 1e-5 if dtype == torch.float32 else 1e-2)
 
     xr = x.float().detach().requires_grad_(True)
     ar, gr = xr.chunk(2, dim=-1)
     (ar * F.gelu(gr)).backward(dy.float())
     dx = ext.geglu_bwd(dy.contiguous(), x)
-    atol = 1e-5 if dtype == torch.float32 else 3e-2
-    assert (dx.float() - xr.grad).abs().max().item() < atol
+    _close(dx, xr.grad, 1e-5 if dtype == torch.float32 else 2e-2)
 
 
 def test_adamw_matches_torch(ext):
@@ -146,11 +150,10 @@ def test_add_noise_velocity(ext):
         xt = ext.add_noise(x0, noise, ac, t)
         acv = ac[t].view(-1, 1, 1, 1)
         ref = acv.sqrt() * x0.float() + (1 - acv).sqrt() * noise.float()
-        tol = 1e-5 if dtype == torch.float32 else 2e-2
-        assert (xt.float() - ref).abs().max().item() < tol
+        _close(xt, ref, 1e-5 if dtype == torch.float32 else 1e-2)
         v = ext.get_velocity(x0, noise, ac, t)
         refv = acv.sqrt() * noise.float() - (1 - acv).sqrt() * x0.float()
-        assert (v.float() - refv).abs().max().item() < tol
+        _close(v, refv, 1e-5 if dtype == torch.float32 else 1e-2)
 
 
 def test_cfg_combine(ext):
@@ -158,7 +161,7 @@ def test_cfg_combine(ext):
     et = torch.randn_like(eu)
     out = ext.cfg_combine(eu, et, 7.5)
     ref = eu.float() + 7.5 * (et.float() - eu.float())
-    assert (out.float() - ref).abs().max().item() < 3e-2
+    _close(out, ref, 1e-2)
 
 
 def test_ops_dispatch_uses_hip_on_gpu():
