@@ -38,6 +38,13 @@ class CLIPTextConfig:
         return cls()
 
     @classmethod
+    def sd14(cls) -> "CLIPTextConfig":
+        """CLIP ViT-L/14 text tower (SD-1.x conditioning)."""
+        return cls(hidden_size=768, intermediate_size=3072,
+                   num_hidden_layers=12, num_attention_heads=12,
+                   hidden_act="quick_gelu")
+
+    @classmethod
     def tiny(cls) -> "CLIPTextConfig":
         # full CLIP vocab so real tokenizer ids (bos 49406/eos 49407) stay valid
         return cls(vocab_size=49408, hidden_size=32, intermediate_size=64,

@@ -53,6 +53,13 @@ class UNetConfig:
         return cls()
 
     @classmethod
+    def sd14(cls) -> "UNetConfig":
+        """CompVis/stable-diffusion-v1-4 unet: 8 heads per block, 768-d
+        cross-attention (CLIP ViT-L), conv proj_in/out."""
+        return cls(sample_size=64, attention_head_dim=(8, 8, 8, 8),
+                   cross_attention_dim=768, use_linear_projection=False)
+
+    @classmethod
     def tiny(cls) -> "UNetConfig":
         """A tiny config for CPU tests — same topology, small widths."""
         return cls(

@@ -1,0 +1,227 @@
+#!/usr/bin/env python3
+"""Retrieval/metrics CLI — parity with /root/reference/diff_retrieval.py.
+
+Distributed feature extraction of generated + training images (SSCD /
+DINO / CLIP backbones), similarity matrices (rocBLAS GEMM), similarity
+stats + histograms, CLIP alignment score, image-complexity correlations,
+FID, duplication analysis and match-gallery plots; wandb-schema metrics
+with an always-on JSONL fallback.
+
+Distributed: launch under `python -m torch.distributed.run
+--nproc-per-node N diff_retrieval.py ...` (one process per GPU over
+RCCL; replaces the reference's mp.spawn path). Runs end-to-end on CPU
+for the no-GPU plumbing config (BASELINE config 1).
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+from pathlib import Path
+
+import numpy as np
+import torch
+from torch.utils.data import DataLoader
+from torch.utils.data.distributed import DistributedSampler
+
+from dcr_amd.data import SynthDataset, EvalTransform
+from dcr_amd.parallel import (barrier, get_rank, get_world_size,
+                              init_distributed_mode, is_main_process)
+from dcr_amd.retrieval import (extract_features, gen_clipscore, glcm_entropy,
+                               jpeg_size, l2_normalize, load_clip, load_dino,
+                               load_sscd, pearson, sim_matrix,
+                               similarity_histogram, top_matches, topk_stats,
+                               tv_loss)
+from dcr_amd.utils import Tracker
+
+
+def parse_args():
+    p = argparse.ArgumentParser("Generic image retrieval given a path")
+    p.add_argument("--query_dir", type=str, required=True, help="The inferences")
+    p.add_argument("--val_dir", type=str, required=True, help="The train data")
+    p.add_argument("--pt_style", default="sscd", type=str)
+    p.add_argument("-a", "--arch", default="resnet50", type=str)
+    p.add_argument("-j", "--workers", default=4, type=int)
+    p.add_argument("-b", "--batch-size", default=128, type=int)
+    p.add_argument("--world-size", default=-1, type=int)
+    p.add_argument("--rank", default=-1, type=int)
+    p.add_argument("--dist-url", default="env://", type=str)
+    p.add_argument("--dist-backend", default="nccl", type=str)
+    p.add_argument("--seed", default=None, type=int)
+    p.add_argument("--gpu", default=None, type=int)
+    p.add_argument("--multiprocessing-distributed", action="store_true")
+    p.add_argument("--multiscale", default=False, type=lambda s: s in ("1", "true", "True"))
+    p.add_argument("--pretrained", default="", type=str)
+    p.add_argument("--similarity_metric", default="dotproduct", type=str)
+    p.add_argument("--num_loss_chunks", default=1, type=int)
+    p.add_argument("--numpatches", default=1, type=int)
+    p.add_argument("--isvit", action="store_true")
+    p.add_argument("--layer", default=1, type=int)
+    p.add_argument("--stype", default="", type=str)
+    p.add_argument("--keephead", action="store_true")
+    p.add_argument("--keeppredictor", action="store_true")
+    p.add_argument("-ssp", "--sim_save_path", type=str, default="./similarityscores/")
+    p.add_argument("--einsum_chunks", default=30, type=int)
+    p.add_argument("--dontsave", action="store_true")
+    p.add_argument("--num_matches", default=4, type=int)
+    p.add_argument("--imsize", default=224, type=int)
+    p.add_argument("--noeval", action="store_true")
+    p.add_argument("--skip_fid", action="store_true")
+    p.add_argument("--project", default="imsimv2_retrieval", type=str)
+    return p.parse_args()
+
+
+def build_backbone(args, device):
+    """Backbone per --pt_style (reference diff_retrieval.py:249-285)."""
+    if args.pt_style.startswith("dino"):
+        arch = args.pt_style if args.pt_style != "dino" else "dino_vitb16"
+        return load_dino(arch, weights=args.pretrained or None, device=device)
+    if args.pt_style == "clip":
+        model, _ = load_clip(device=device)
+        return lambda x: model.encode_image(x)
+    return load_sscd(args.pt_style, device=device)  # sscd variants
+
+
+def main():
+    args = parse_args()
+    rank, world, local = init_distributed_mode()
+    device = torch.device("cuda", local) if torch.cuda.is_available() \
+        else torch.device("cpu")
+    if args.seed is not None:
+        torch.manual_seed(args.seed)
+        torch.backends.cudnn.deterministic = True
+
+    model = build_backbone(args, device)
+
+    tf = EvalTransform(args.imsize)
+    query_ds = SynthDataset(args.query_dir, transform=tf, with_prompts=True)
+    val_ds = SynthDataset(args.val_dir, transform=tf)
+    print(f"query: {len(query_ds)} imgs, values: {len(val_ds)} imgs")
+
+    def loader(ds):
+        sampler = DistributedSampler(ds, shuffle=False) if world > 1 else None
+        return DataLoader(ds, batch_size=min(64, max(1, args.batch_size // 2)),
+                          sampler=sampler, num_workers=args.workers,
+                          pin_memory=device.type == "cuda")
+
+    # HOT LOOP 1: distributed feature extraction + all-gather
+    query_f = extract_features(model, loader(query_ds), device,
+                               multiscale=args.multiscale)
+    val_f = extract_features(model, loader(val_ds), device,
+                             multiscale=args.multiscale)
+    barrier()
+    if not is_main_process():
+        return
+
+    out_dir = Path(args.sim_save_path)
+    out_dir.mkdir(parents=True, exist_ok=True)
+    tracker = Tracker(args.project, name=Path(args.query_dir).name,
+                      config=vars(args), out_dir=out_dir)
+
+    query_f = l2_normalize(query_f)
+    val_f = l2_normalize(val_f)
+
+    # HOT LOOP 2: dense GEMM similarity (rocBLAS)
+    sim = sim_matrix(val_f, query_f).t()        # [n_gen, n_train]
+    sim_tt = sim_matrix(val_f, val_f)           # [n_train, n_train]
+
+    if not args.dontsave:
+        torch.save(sim.cpu(), out_dir / "similarity.pth")
+        torch.save(sim_tt.cpu(), out_dir / "similarity_wtrain.pth")
+
+    stats = topk_stats(sim, sim_tt)
+    tracker.log(stats)
+    print(json.dumps(stats, indent=2))
+
+    top1 = sim.max(dim=1).values.float().cpu()
+    counts, bins = similarity_histogram(top1)
+    try:
+        import matplotlib
+        matplotlib.use("Agg")
+        import matplotlib.pyplot as plt
+        plt.figure()
+        plt.hist(top1.numpy(), bins=50, range=(0, 1), alpha=0.7, label="gen->train")
+        t2 = sim_tt.topk(2, dim=1).values[:, 1].float().cpu()
+        plt.hist(t2.numpy(), bins=50, range=(0, 1), alpha=0.5, label="train->train")
+        plt.legend()
+        plt.xlabel("top-1 SSCD similarity")
+        plt.savefig(out_dir / "similarity_hist.png", dpi=120)
+        plt.close()
+    except Exception as e:
+        print(f"histogram plot skipped: {e}")
+
+    # CLIP alignment score (reference :486-495) when prompts exist
+    if query_ds.prompts and not args.noeval:
+        clip_model, clip_tok = load_clip(device=device)
+        imgs = torch.stack([query_ds[i][0] for i in range(len(query_ds))])
+        cs = gen_clipscore(clip_model, clip_tok, imgs, query_ds.prompts,
+                           device=device)
+        tracker.log({"clipscore_mean": cs.mean().item(),
+                     "clipscore_std": cs.std().item()})
+
+    # complexity correlations (reference :498-540)
+    if not args.noeval:
+        from PIL import Image
+        ent, jpg, tv = [], [], []
+        for f in query_ds.files:
+            img = Image.open(f).convert("RGB")
+            arr = np.asarray(img)
+            ent.append(glcm_entropy(arr))
+            jpg.append(jpeg_size(img))
+            t = torch.from_numpy(arr.copy()).permute(2, 0, 1).float() / 255
+            tv.append(tv_loss(t).item())
+        sims_np = top1.numpy()
+        comp = {
+            "entropy_sim_pearson": pearson(ent, sims_np),
+            "jpegsize_sim_pearson": pearson(jpg, sims_np),
+            "tv_sim_pearson": pearson(tv, sims_np),
+        }
+        tracker.log(comp)
+        if not args.dontsave:
+            torch.save({"entropy": ent, "jpeg": jpg, "tv": tv},
+                       out_dir / "complexity.pth")
+
+    # FID (reference :597-600; HOT LOOP 3)
+    if not args.noeval and not args.skip_fid:
+        from dcr_amd.metrics import calculate_fid_given_paths
+        fid = calculate_fid_given_paths([args.query_dir, args.val_dir],
+                                        batch_size=50, device=str(device),
+                                        dims=2048)
+        tracker.log({"fid": fid})
+        print(f"FID: {fid:.3f}")
+
+    # gallery: top matches for the most-copied generations (reference :609-640)
+    if not args.dontsave:
+        try:
+            import matplotlib
+            matplotlib.use("Agg")
+            import matplotlib.pyplot as plt
+            from PIL import Image
+            k = args.num_matches
+            n_show = min(10, len(query_ds))
+            order = torch.argsort(top1, descending=True)[:n_show]
+            vals, idxs = top_matches(sim, k=k)
+            fig, axes = plt.subplots(n_show, k + 1,
+                                     figsize=(2 * (k + 1), 2 * n_show))
+            if n_show == 1:
+                axes = axes[None, :]
+            for row, qi in enumerate(order.tolist()):
+                axes[row, 0].imshow(Image.open(query_ds.files[qi]))
+                axes[row, 0].set_title(f"gen {qi}", fontsize=6)
+                for col in range(k):
+                    ti = int(idxs[qi, col])
+                    axes[row, col + 1].imshow(Image.open(val_ds.files[ti]))
+                    axes[row, col + 1].set_title(f"{vals[qi, col]:.2f}", fontsize=6)
+                for ax in axes[row]:
+                    ax.axis("off")
+            plt.tight_layout()
+            plt.savefig(out_dir / "gallery.png", dpi=100)
+            plt.close()
+        except Exception as e:
+            print(f"gallery plot skipped: {e}")
+
+    tracker.finish()
+
+
+if __name__ == "__main__":
+    main()

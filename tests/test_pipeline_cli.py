@@ -1,0 +1,135 @@
+"""Pipeline + CLI golden-file tests (CPU, tiny models).
+
+Covers the reference's CLI/artifact contracts: prompts.txt +
+generations/{i}.png naming (diff_inference.py:179-201), savepath
+derivation (:44-81), checkpoint_{step} consumption, retrieval CLI
+end-to-end on synthetic data (BASELINE config 1 shape)."""
+import json
+import subprocess
+import sys
+from pathlib import Path
+
+import numpy as np
+import pytest
+import torch
+from PIL import Image
+
+from dcr_amd.data import HashTokenizer
+from dcr_amd.models import (AutoencoderKL, CLIPTextModel, CLIPTextConfig,
+                            UNet2DConditionModel, UNetConfig, VAEConfig)
+from dcr_amd.pipelines import StableDiffusionPipeline
+from dcr_amd.schedulers import DDIMScheduler, DPMSolverMultistepScheduler
+
+
+def tiny_pipe(embed_noise_lam=0.0, scheduler=None):
+    torch.manual_seed(0)
+    return StableDiffusionPipeline(
+        UNet2DConditionModel(UNetConfig.tiny()),
+        AutoencoderKL(VAEConfig.tiny()),
+        CLIPTextModel(CLIPTextConfig.tiny()),
+        HashTokenizer(),
+        scheduler or DDIMScheduler(),
+        embed_noise_lam=embed_noise_lam)
+
+
+def test_pipeline_generates_images():
+    pipe = tiny_pipe()
+    out = pipe("a photo of a church", height=64, width=64,
+               num_inference_steps=3, num_images_per_prompt=2,
+               generator=torch.Generator().manual_seed(0))
+    assert len(out.images) == 2
+    assert out.images[0].size == (64, 64)
+
+
+def test_pipeline_dpm_solver():
+    pipe = tiny_pipe(scheduler=DPMSolverMultistepScheduler())
+    out = pipe("hello", height=64, width=64, num_inference_steps=4)
+    assert len(out.images) == 1
+
+
+def test_pipeline_embed_noise_changes_output():
+    g1 = torch.Generator().manual_seed(7)
+    g2 = torch.Generator().manual_seed(7)
+    p0 = tiny_pipe(0.0)
+    imgs_a = p0("x", height=64, width=64, num_inference_steps=2, generator=g1,
+                output_type="pt").images
+    p1 = tiny_pipe(0.5)  # same seed for weights (manual_seed(0) inside)
+    imgs_b = p1("x", height=64, width=64, num_inference_steps=2, generator=g2,
+                output_type="pt").images
+    assert not torch.allclose(imgs_a, imgs_b, atol=1e-4)
+
+
+def test_pipeline_save_load_roundtrip(tmp_path):
+    pipe = tiny_pipe()
+    pipe.save_pretrained(tmp_path / "checkpoint")
+    assert (tmp_path / "checkpoint" / "model_index.json").exists()
+    pipe2 = StableDiffusionPipeline.from_pretrained(tmp_path / "checkpoint")
+    g1 = torch.Generator().manual_seed(3)
+    g2 = torch.Generator().manual_seed(3)
+    a = pipe("same", height=64, width=64, num_inference_steps=2, generator=g1,
+             output_type="pt").images
+    b = pipe2("same", height=64, width=64, num_inference_steps=2, generator=g2,
+              output_type="pt").images
+    assert torch.allclose(a, b, atol=1e-4)
+
+
+def test_savepath_derivation():
+    import diff_inference as di
+
+    class A:
+        modelpath = "/models/myrun_imagenette_instancelevel_blip_nodup"
+        dataset = None
+        capstyle = None
+        iternum = 2000
+        modelstyle = "instancelevel_blip"
+        rand_noise_lam = 0.1
+        rand_augs = None
+        rand_aug_repeats = 2
+
+    sp = di.derive_savepath(A())
+    assert sp == ("./inferences/imagenette10_frozentext/"
+                  "myrun_imagenette_instancelevel_blip_nodup_2000/"
+                  "instancelevel_blip_ginfer0.1")
+
+
+def test_prompt_augmentation_modes():
+    import diff_inference as di
+    tok = HashTokenizer()
+    np.random.seed(0)
+    base = "a photo of a dog"
+    p1 = di.prompt_augmentation(base, "rand_numb_add", tok, 2)
+    assert len(p1.split()) == len(base.split()) + 2
+    p2 = di.prompt_augmentation(base, "rand_word_add", tok, 2)
+    assert len(p2.split()) >= len(base.split()) + 2
+    p3 = di.prompt_augmentation(base, "rand_word_repeat", tok, 2)
+    assert len(p3.split()) == len(base.split()) + 2
+    with pytest.raises(Exception):
+        di.prompt_augmentation(base, "nope", tok)
+
+
+@pytest.mark.timeout(600)
+def test_retrieval_cli_end_to_end(tmp_path):
+    """diff_retrieval.py over synthetic generations+train dirs, CPU."""
+    rng = np.random.default_rng(0)
+    qdir, vdir = tmp_path / "gens", tmp_path / "train"
+    qdir.mkdir(); vdir.mkdir()
+    for i in range(6):
+        Image.fromarray(rng.integers(0, 255, (64, 64, 3)).astype(np.uint8)) \
+            .save(qdir / f"{i}.png")
+    for i in range(8):
+        Image.fromarray(rng.integers(0, 255, (64, 64, 3)).astype(np.uint8)) \
+            .save(vdir / f"{i}.png")
+    (qdir / "prompts.txt").write_text("\n".join(f"prompt {i}" for i in range(6)))
+    out = tmp_path / "simscores"
+    r = subprocess.run(
+        [sys.executable, "diff_retrieval.py", "--query_dir", str(qdir),
+         "--val_dir", str(vdir), "--pt_style", "sscd", "-b", "8", "-j", "0",
+         "--imsize", "64", "-ssp", str(out), "--skip_fid", "--noeval"],
+        capture_output=True, text=True, cwd=str(Path(__file__).parent.parent),
+        timeout=570)
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert (out / "similarity.pth").exists()
+    assert (out / "similarity_wtrain.pth").exists()
+    log = out / "imsimv2_retrieval_log.jsonl"
+    recs = [json.loads(l) for l in log.read_text().splitlines()]
+    assert any("sim_mean" in r_ for r_ in recs)

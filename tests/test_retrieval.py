@@ -1,0 +1,168 @@
+"""Retrieval + metrics + search tests (CPU).
+
+BASELINE config 1: SSCD dot-product retrieval on 128 random 64x64
+images, CPU only — the full plumbing must run without a GPU."""
+import pickle
+
+import numpy as np
+import pytest
+import torch
+from PIL import Image
+from torch.utils.data import DataLoader
+
+from dcr_amd.retrieval import (SSCDModel, extract_features, l2_normalize,
+                               sim_matrix, topk_stats, glcm_entropy, jpeg_size,
+                               tv_loss, load_sscd)
+from dcr_amd.retrieval.similarity import einsum_in_chunks
+
+
+def test_sscd_descriptor_shape_and_norm():
+    torch.manual_seed(0)
+    m = SSCDModel().eval()
+    with torch.no_grad():
+        f = m(torch.randn(2, 3, 64, 64))
+    assert f.shape == (2, 512)
+    assert torch.allclose(f.norm(dim=-1), torch.ones(2), atol=1e-5)
+
+
+def test_baseline_config1_cpu_retrieval():
+    """128 random 64x64 images -> SSCD features -> dot-product sim -> stats."""
+    torch.manual_seed(1)
+    model = load_sscd("sscd", device="cpu")
+    imgs = torch.randn(128, 3, 64, 64)
+    ds = [(imgs[i], i) for i in range(128)]
+    loader = DataLoader(ds, batch_size=32)
+    feats = extract_features(model, loader, torch.device("cpu"))
+    assert feats.shape == (128, 512)
+    feats = l2_normalize(feats)
+    sim = sim_matrix(feats[:64], feats[64:])
+    assert sim.shape == (64, 64)
+    stats = topk_stats(sim, sim_matrix(feats[64:], feats[64:]))
+    for k in ("sim_mean", "sim_std", "sim_95pc", "sim_gt_05pc",
+              "trainsim_mean", "trainsim_gt_05pc"):
+        assert k in stats and np.isfinite(stats[k])
+    # identical sets => self-similarity 1.0 on the diagonal
+    sim_self = sim_matrix(feats, feats)
+    assert torch.allclose(sim_self.diag(), torch.ones(128), atol=1e-4)
+
+
+def test_extract_features_index_order():
+    """features land at their dataset indices regardless of batch order."""
+    model = torch.nn.Flatten()
+    data = torch.arange(8, dtype=torch.float32).view(8, 1).repeat(1, 4)
+    ds = [(data[i].view(1, 2, 2), i) for i in reversed(range(8))]
+    loader = DataLoader(ds, batch_size=3)
+    feats = extract_features(model, loader, torch.device("cpu"))
+    assert torch.equal(feats[5], data[5])
+
+
+def test_einsum_in_chunks_matches_dense():
+    torch.manual_seed(0)
+    a = torch.randn(6, 8, 3)
+    b = torch.randn(5, 8, 3)
+    out = einsum_in_chunks(a, b, chunk=2)
+    ref = torch.einsum("ncp,mcq->nmpq", a, b).amax(dim=(2, 3))
+    assert torch.allclose(out, ref, atol=1e-5)
+
+
+def test_complexity_metrics():
+    rng = np.random.default_rng(0)
+    noise = rng.integers(0, 255, (64, 64, 3)).astype(np.uint8)
+    flat = np.full((64, 64, 3), 128, np.uint8)
+    assert glcm_entropy(noise) > glcm_entropy(flat)
+    assert jpeg_size(Image.fromarray(noise)) > jpeg_size(Image.fromarray(flat))
+    t_noise = torch.rand(3, 32, 32)
+    t_flat = torch.full((3, 32, 32), 0.5)
+    assert tv_loss(t_noise) > tv_loss(t_flat)
+
+
+def test_fid_identical_sets_near_zero(tmp_path):
+    from dcr_amd.metrics import calculate_fid_given_paths
+    rng = np.random.default_rng(0)
+    d1, d2 = tmp_path / "a", tmp_path / "b"
+    d1.mkdir(); d2.mkdir()
+    for i in range(8):
+        arr = rng.integers(0, 255, (64, 64, 3)).astype(np.uint8)
+        Image.fromarray(arr).save(d1 / f"{i}.png")
+        Image.fromarray(arr).save(d2 / f"{i}.png")
+    fid = calculate_fid_given_paths([str(d1), str(d2)], batch_size=4,
+                                    device="cpu", dims=64)
+    assert fid < 1e-3, fid
+
+
+def test_fid_different_sets_positive(tmp_path):
+    from dcr_amd.metrics import calculate_fid_given_paths
+    rng = np.random.default_rng(0)
+    d1, d2 = tmp_path / "a", tmp_path / "b"
+    d1.mkdir(); d2.mkdir()
+    for i in range(8):
+        Image.fromarray(rng.integers(0, 128, (64, 64, 3)).astype(np.uint8)) \
+            .save(d1 / f"{i}.png")
+        Image.fromarray(rng.integers(128, 255, (64, 64, 3)).astype(np.uint8)) \
+            .save(d2 / f"{i}.png")
+    fid = calculate_fid_given_paths([str(d1), str(d2)], batch_size=4,
+                                    device="cpu", dims=64)
+    assert fid > 0.01
+
+
+def test_ipr_precision_recall():
+    from dcr_amd.metrics.ipr import IPR
+    torch.manual_seed(0)
+    ipr = IPR(batch_size=8, k=3)
+    ref = torch.rand(16, 3, 32, 32)
+    ipr.compute_manifold_ref(ref)
+    pr = ipr.precision_and_recall(ref + 0.01 * torch.randn_like(ref))
+    assert 0.0 <= pr.precision <= 1.0 and 0.0 <= pr.recall <= 1.0
+    assert pr.precision > 0.5  # nearly identical distributions overlap
+
+
+def test_embedding_pickle_contract(tmp_path):
+    """embedding.pkl = {'features': [N,D] f32, 'indexes': list}."""
+    from dcr_amd.search import generate_embeddings
+    blob = generate_embeddings(None, tmp_path / "embedding.pkl",
+                               batch_size=16, device="cpu", num_workers=0,
+                               synthetic_n=24, seed=3)
+    assert blob["features"].dtype == np.float32
+    assert blob["features"].shape[0] == 24
+    assert len(blob["indexes"]) == 24
+    with open(tmp_path / "embedding.pkl", "rb") as fh:
+        loaded = pickle.load(fh)
+    assert set(loaded.keys()) == {"features", "indexes"}
+
+
+def test_stream_top1_finds_planted_match(tmp_path):
+    from dcr_amd.search import stream_top1, dump_matches
+    rng = np.random.default_rng(0)
+    D = 16
+    # two chunks; plant exact query rows at known places
+    q = rng.normal(size=(4, D)).astype(np.float32)
+    q /= np.linalg.norm(q, axis=1, keepdims=True)
+    c1 = rng.normal(size=(50, D)).astype(np.float32)
+    c2 = rng.normal(size=(50, D)).astype(np.float32)
+    c1 /= np.linalg.norm(c1, axis=1, keepdims=True)
+    c2 /= np.linalg.norm(c2, axis=1, keepdims=True)
+    c1[7] = q[0]
+    c2[3] = q[2]
+    for name, feats in [("embedding_0.pkl", c1), ("embedding_1.pkl", c2)]:
+        with open(tmp_path / name, "wb") as fh:
+            pickle.dump({"features": feats,
+                         "indexes": [f"{name}:{i}" for i in range(50)]}, fh)
+    scores, keys = stream_top1(torch.from_numpy(q),
+                               sorted(tmp_path.glob("*.pkl")), device="cpu")
+    assert scores[0] > 0.999 and keys[0] == "embedding_0.pkl:7"
+    assert scores[2] > 0.999 and keys[2] == "embedding_1.pkl:3"
+    dump_matches(scores, keys, tmp_path / "match.pkl")
+    with open(tmp_path / "match.pkl", "rb") as fh:
+        m = pickle.load(fh)
+    assert list(m.keys()) == ["scores", "keys"]
+
+
+def test_sharded_topk_matches_full():
+    from dcr_amd.search import sharded_topk
+    torch.manual_seed(0)
+    q = torch.randn(5, 8)
+    shard = torch.randn(100, 8)
+    v, i = sharded_topk(q, shard, k=3, chunk=17, global_offset=1000)
+    ref = (q @ shard.t()).topk(3, dim=1)
+    assert torch.allclose(v, ref.values, atol=1e-5)
+    assert torch.equal(i, ref.indices + 1000)
