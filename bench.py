@@ -43,6 +43,8 @@ def main():
               file=sys.stderr)
     n_gpus = world if world > 1 else 1
 
+    if os.environ.get("DCR_CONV_BENCHMARK") == "1":
+        torch.backends.cudnn.benchmark = True  # MIOpen find-mode tuning
     use_cuda = torch.cuda.is_available()
     device = torch.device("cuda", local) if use_cuda else torch.device("cpu")
     if use_cuda:

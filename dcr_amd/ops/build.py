@@ -21,6 +21,7 @@ SOURCES = [
     HIP_DIR / "bindings.cpp",
     HIP_DIR / "norms.hip",
     HIP_DIR / "elementwise.hip",
+    HIP_DIR / "attention.hip",
 ]
 
 

@@ -131,8 +131,42 @@ def geglu(x: torch.Tensor) -> torch.Tensor:
 
 
 # --------------------------------------------------------------------------
-# Attention (flash-style). q,k,v: [B, H, Lq/Lk, D].
+# Attention (hand-written CDNA4 flash kernel). q,k,v: [B, H, L, D].
 # --------------------------------------------------------------------------
+class _FlashAttention(torch.autograd.Function):
+    """bf16 flash attention, head_dim 64 (dcr_amd/ops/hip/attention.hip)."""
+
+    @staticmethod
+    def forward(ctx, q, k, v, scale, causal):
+        m = require_hip("attn")
+        o, lse = m.attn_fwd(q, k, v, scale, causal)
+        ctx.save_for_backward(q, k, v, o, lse)
+        ctx.scale = scale
+        ctx.causal = causal
+        return o
+
+    @staticmethod
+    def backward(ctx, dO):
+        q, k, v, o, lse = ctx.saved_tensors
+        m = require_hip("attn")
+        dQ, dK, dV = m.attn_bwd(q, k, v, o, dO.contiguous(), lse,
+                                ctx.scale, ctx.causal)
+        return dQ, dK, dV, None, None
+
+
+def _attention_math(q, k, v, scale, causal):
+    """Composite path (rocBLAS GEMMs + native softmax) for shapes the
+    flash kernel does not cover (head_dim != 64, e.g. SD-1.4's 40/80/160
+    and the VAE's single 512-d head). No Triton involved."""
+    s = (q.float() @ k.float().transpose(-2, -1)) * scale
+    if causal:
+        Lq, Lk = q.shape[-2], k.shape[-2]
+        mask = torch.ones(Lq, Lk, dtype=torch.bool, device=q.device).tril_()
+        s = s.masked_fill(~mask, float("-inf"))
+    p = s.softmax(dim=-1)
+    return (p @ v.float()).to(q.dtype)
+
+
 def attention(
     q: torch.Tensor,
     k: torch.Tensor,
@@ -140,30 +174,19 @@ def attention(
     causal: bool = False,
     scale: Optional[float] = None,
 ) -> torch.Tensor:
-    from . import ext
-
-    m = ext()
-    if use_hip(q) and m is not None and hasattr(m, "attn_fwd") and not q.requires_grad \
-            and q.dtype == torch.bfloat16 and q.shape[-1] <= 128 and not causal:
-        # inference path: hand-written CDNA4 flash forward
-        return _AttnFwdOnly.apply(q, k, v, scale)
     if scale is None:
         scale = 1.0 / math.sqrt(q.shape[-1])
+    if use_hip(q):
+        if q.dtype == torch.bfloat16 and q.shape[-1] == 64 and k.shape[-1] == 64:
+            shp = q.shape
+            out = _FlashAttention.apply(
+                q.reshape(-1, shp[-2], 64).contiguous(),
+                k.reshape(-1, k.shape[-2], 64).contiguous(),
+                v.reshape(-1, v.shape[-2], 64).contiguous(),
+                scale, causal)
+            return out.reshape(shp)
+        return _attention_math(q, k, v, scale, causal)
     return F.scaled_dot_product_attention(q, k, v, is_causal=causal, scale=scale)
-
-
-class _AttnFwdOnly(torch.autograd.Function):
-    @staticmethod
-    def forward(ctx, q, k, v, scale):
-        from . import ext
-
-        m = ext()
-        s = scale if scale is not None else 1.0 / math.sqrt(q.shape[-1])
-        return m.attn_fwd(q.contiguous(), k.contiguous(), v.contiguous(), s)
-
-    @staticmethod
-    def backward(ctx, dy):  # pragma: no cover
-        raise RuntimeError("attn_fwd is inference-only; training uses SDPA for now")
 
 
 # --------------------------------------------------------------------------

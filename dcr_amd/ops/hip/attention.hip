@@ -1,0 +1,558 @@
+// Flash attention (forward + backward) for MI355X (gfx950), bf16, D=64.
+//
+// The hand-written CDNA4 FMHA required by the north star: LDS-staged
+// K/V tiles, MFMA (v_mfma_f32_16x16x32_bf16) for QK^T / PV and the
+// backward GEMMs, online softmax with fp32 row stats, wave64 quarter-
+// group row reductions. Replaces torch SDPA (AOTriton) for SD-2.1
+// attention (head_dim 64: latent self-attn L in {64,256,1024,4096},
+// text cross-attn Lk=77 — arbitrary lengths via tile masking) and CLIP
+// text (causal L=77). Reference ops covered: SURVEY.md §2.4.A FMHA rows.
+//
+// Fragment conventions (gfx950 v_mfma_f32_16x16x32_bf16, K-contiguous;
+// verified on hardware by the mfma_probe test in tests/test_ops_gpu.py):
+//   A[m][k]  : lane l holds row  m=l&15, k = (l>>4)*8 + j   (8 bf16)
+//   B[k][n]  : lane l holds col  n=l&15, k = (l>>4)*8 + j
+//   C/D[m][n]: lane l holds col  n=l&15, rows m=(l>>4)*4 + r (4 fp32)
+// Every LDS fragment read is "8 contiguous contraction elements at row
+// l&15", so tiles are stored with the contraction dim innermost; row
+// pitch 72 bf16 (144 B) puts the 16 rows of a ds_read_b128 lane group on
+// 16 distinct banks (36*r mod 64 is a 16-cycle) — conflict-free without
+// an XOR swizzle.
+//
+// Backward is the FlashAttention-2 split: attn_bwd_dkdv accumulates
+// dK/dV per key tile (loops q tiles), attn_bwd_dq accumulates dQ per
+// query tile (loops key tiles); both recompute P from (Q, K, LSE);
+// delta = rowsum(dO*O) precomputed by attn_bwd_delta.
+
+#include "dcr_common.h"
+
+namespace dcr_attn {
+
+typedef __attribute__((ext_vector_type(8))) short bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4_t;
+
+#define TILE 64
+#define PITCH 72
+#define DHEAD 64
+
+using bf16_t = __hip_bfloat16;
+
+__device__ __forceinline__ short f2bf_rne(float f) {
+  __hip_bfloat16 h = __float2bfloat16(f);
+  return *reinterpret_cast<short*>(&h);
+}
+
+// cooperative tile load (256 threads): dst[row][dim], zero-padded rows
+__device__ __forceinline__ void load_tile(const bf16_t* __restrict__ src,
+                                          int valid_rows,
+                                          short* __restrict__ dst) {
+  const int t = threadIdx.x;
+  const int row = t >> 2;
+  const int col = (t & 3) * 16;
+  uint4 a = make_uint4(0, 0, 0, 0), b = a;
+  if (row < valid_rows) {
+    const uint4* p = reinterpret_cast<const uint4*>(src + (long)row * DHEAD + col);
+    a = p[0];
+    b = p[1];
+  }
+  uint4* d = reinterpret_cast<uint4*>(dst + row * PITCH + col);
+  d[0] = a;
+  d[1] = b;
+}
+
+// transposed: dst[dim][row] = src[row][dim]
+__device__ __forceinline__ void load_tile_T(const bf16_t* __restrict__ src,
+                                            int valid_rows,
+                                            short* __restrict__ dst) {
+  const int t = threadIdx.x;
+  const int row = t >> 2;
+  const int col0 = (t & 3) * 16;
+  short v[16];
+  if (row < valid_rows) {
+    const uint4* p = reinterpret_cast<const uint4*>(src + (long)row * DHEAD + col0);
+    *reinterpret_cast<uint4*>(v) = p[0];
+    *reinterpret_cast<uint4*>(v + 8) = p[1];
+  } else {
+#pragma unroll
+    for (int j = 0; j < 16; ++j) v[j] = 0;
+  }
+#pragma unroll
+  for (int j = 0; j < 16; ++j) dst[(col0 + j) * PITCH + row] = v[j];
+}
+
+__device__ __forceinline__ bf16x8 frag(const short* lds, int row, int koff) {
+  return *reinterpret_cast<const bf16x8*>(lds + row * PITCH + koff);
+}
+
+// 16-lane (quarter-wave) reductions — rows are spread over lanes l^1..l^8
+__device__ __forceinline__ float qmax(float v) {
+#pragma unroll
+  for (int m = 8; m >= 1; m >>= 1) v = fmaxf(v, __shfl_xor(v, m, 64));
+  return v;
+}
+__device__ __forceinline__ float qsum(float v) {
+#pragma unroll
+  for (int m = 8; m >= 1; m >>= 1) v += __shfl_xor(v, m, 64);
+  return v;
+}
+
+// ==========================================================================
+// Forward. grid (ceil(Lq/64), BH), block 256.
+// ==========================================================================
+__global__ __launch_bounds__(256)
+void attn_fwd_kernel(const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
+                     const bf16_t* __restrict__ v, bf16_t* __restrict__ o,
+                     float* __restrict__ lse, int Lq, int Lk, float scale,
+                     int causal) {
+  __shared__ short sQ[TILE * PITCH];
+  __shared__ short sK[TILE * PITCH];
+  __shared__ short sVT[TILE * PITCH];   // [dim][key]
+  __shared__ short sP[TILE * PITCH];    // [qrow][key]
+
+  const int bh = blockIdx.y;
+  const int q0 = blockIdx.x * TILE;
+  const bf16_t* qp = q + ((long)bh * Lq + q0) * DHEAD;
+  const bf16_t* kp = k + (long)bh * Lk * DHEAD;
+  const bf16_t* vp = v + (long)bh * Lk * DHEAD;
+  bf16_t* op = o + (long)bh * Lq * DHEAD;
+
+  load_tile(qp, Lq - q0, sQ);
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int l16 = lane & 15;
+  const int kgrp = lane >> 4;
+  const int wrow0 = wid * 16;
+
+  float row_max[4] = {-1e30f, -1e30f, -1e30f, -1e30f};
+  float row_sum[4] = {0.f, 0.f, 0.f, 0.f};
+  f32x4_t acc_o[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) acc_o[i] = {0.f, 0.f, 0.f, 0.f};
+
+  const int kv_end = causal ? min(Lk, q0 + TILE) : Lk;
+
+  for (int kv0 = 0; kv0 < kv_end; kv0 += TILE) {
+    __syncthreads();
+    load_tile(kp + (long)kv0 * DHEAD, Lk - kv0, sK);
+    load_tile_T(vp + (long)kv0 * DHEAD, Lk - kv0, sVT);
+    __syncthreads();
+
+    // S = scale * Q K^T  (wave: 16 q-rows x 64 keys)
+    bf16x8 qf0 = frag(sQ, wrow0 + l16, kgrp * 8);
+    bf16x8 qf1 = frag(sQ, wrow0 + l16, kgrp * 8 + 32);
+    f32x4_t s_frag[4];
+#pragma unroll
+    for (int ns = 0; ns < 4; ++ns) {
+      f32x4_t acc = {0.f, 0.f, 0.f, 0.f};
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          qf0, frag(sK, ns * 16 + l16, kgrp * 8), acc, 0, 0, 0);
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          qf1, frag(sK, ns * 16 + l16, kgrp * 8 + 32), acc, 0, 0, 0);
+      s_frag[ns] = acc;
+    }
+
+    // mask + online softmax
+    float tile_max[4] = {-1e30f, -1e30f, -1e30f, -1e30f};
+#pragma unroll
+    for (int ns = 0; ns < 4; ++ns) {
+      const int key = kv0 + ns * 16 + l16;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float sv = s_frag[ns][r] * scale;
+        const int qrow = q0 + wrow0 + kgrp * 4 + r;
+        if (key >= Lk || (causal && key > qrow)) sv = -1e30f;
+        s_frag[ns][r] = sv;
+        tile_max[r] = fmaxf(tile_max[r], sv);
+      }
+    }
+    float alpha[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      float tm = qmax(tile_max[r]);
+      float mnew = fmaxf(row_max[r], tm);
+      alpha[r] = (mnew <= -1e29f) ? 1.f : __expf(row_max[r] - mnew);
+      row_max[r] = mnew;
+    }
+
+    float psum[4] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int ns = 0; ns < 4; ++ns) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float pv = (s_frag[ns][r] <= -1e29f)
+                       ? 0.f : __expf(s_frag[ns][r] - row_max[r]);
+        psum[r] += pv;
+        sP[(wrow0 + kgrp * 4 + r) * PITCH + ns * 16 + l16] = f2bf_rne(pv);
+      }
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) row_sum[r] = row_sum[r] * alpha[r] + qsum(psum[r]);
+
+#pragma unroll
+    for (int ds_ = 0; ds_ < 4; ++ds_)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) acc_o[ds_][r] *= alpha[r];
+
+    // O += P V   (contraction 64 keys, two 32-chunks; same-wave P reuse)
+    bf16x8 pf0 = frag(sP, wrow0 + l16, kgrp * 8);
+    bf16x8 pf1 = frag(sP, wrow0 + l16, kgrp * 8 + 32);
+#pragma unroll
+    for (int ds_ = 0; ds_ < 4; ++ds_) {
+      acc_o[ds_] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          pf0, frag(sVT, ds_ * 16 + l16, kgrp * 8), acc_o[ds_], 0, 0, 0);
+      acc_o[ds_] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          pf1, frag(sVT, ds_ * 16 + l16, kgrp * 8 + 32), acc_o[ds_], 0, 0, 0);
+    }
+  }
+
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int qrow = q0 + wrow0 + kgrp * 4 + r;
+    if (qrow >= Lq) continue;
+    const float inv = (row_sum[r] > 0.f) ? 1.f / row_sum[r] : 0.f;
+#pragma unroll
+    for (int ds_ = 0; ds_ < 4; ++ds_)
+      op[(long)qrow * DHEAD + ds_ * 16 + l16] = __float2bfloat16(acc_o[ds_][r] * inv);
+    if (l16 == 0 && lse != nullptr)
+      lse[(long)bh * Lq + qrow] = row_max[r] + __logf(fmaxf(row_sum[r], 1e-30f));
+  }
+}
+
+// ==========================================================================
+// Backward preprocess: delta[row] = sum_d dO[row,d]*O[row,d]
+// ==========================================================================
+__global__ void attn_bwd_delta_kernel(const bf16_t* __restrict__ dO,
+                                      const bf16_t* __restrict__ O,
+                                      float* __restrict__ delta, long total_rows) {
+  long row = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (row >= total_rows) return;
+  const bf16_t* a = dO + row * DHEAD;
+  const bf16_t* b = O + row * DHEAD;
+  float s = 0.f;
+#pragma unroll
+  for (int i = 0; i < DHEAD; i += 4) {
+    dcr::f32x4 av = dcr::load4<__hip_bfloat16>(a + i);
+    dcr::f32x4 bv = dcr::load4<__hip_bfloat16>(b + i);
+    s += av.x * bv.x + av.y * bv.y + av.z * bv.z + av.w * bv.w;
+  }
+  delta[row] = s;
+}
+
+// ==========================================================================
+// Backward dK/dV. grid (ceil(Lk/64), BH); wave owns 16 keys.
+// ==========================================================================
+__global__ __launch_bounds__(256)
+void attn_bwd_dkdv_kernel(const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
+                          const bf16_t* __restrict__ v, const bf16_t* __restrict__ dO,
+                          const float* __restrict__ lse, const float* __restrict__ delta,
+                          bf16_t* __restrict__ dK, bf16_t* __restrict__ dV,
+                          int Lq, int Lk, float scale, int causal) {
+  __shared__ short sK[TILE * PITCH];     // [key][dim]
+  __shared__ short sV[TILE * PITCH];     // [key][dim]
+  __shared__ short sQ[TILE * PITCH];     // [qrow][dim]
+  __shared__ short sQT[TILE * PITCH];    // [dim][qrow]
+  __shared__ short sdO[TILE * PITCH];    // [qrow][dim]
+  __shared__ short sdOT[TILE * PITCH];   // [dim][qrow]
+  __shared__ short sPT[TILE * PITCH];    // [key][qrow]: P^T, then dS^T
+  __shared__ float sLse[TILE];
+  __shared__ float sDelta[TILE];
+
+  const int bh = blockIdx.y;
+  const int k0 = blockIdx.x * TILE;
+  const bf16_t* qp = q + (long)bh * Lq * DHEAD;
+  const bf16_t* kp = k + ((long)bh * Lk + k0) * DHEAD;
+  const bf16_t* vp = v + ((long)bh * Lk + k0) * DHEAD;
+  const bf16_t* dop = dO + (long)bh * Lq * DHEAD;
+  const float* lsep = lse + (long)bh * Lq;
+  const float* delp = delta + (long)bh * Lq;
+
+  load_tile(kp, Lk - k0, sK);
+  load_tile(vp, Lk - k0, sV);
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int l16 = lane & 15;
+  const int kgrp = lane >> 4;
+  const int wkey0 = wid * 16;
+
+  f32x4_t acc_dk[4], acc_dv[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    acc_dk[i] = {0.f, 0.f, 0.f, 0.f};
+    acc_dv[i] = {0.f, 0.f, 0.f, 0.f};
+  }
+
+  const int q_start = causal ? k0 : 0;
+
+  for (int q0 = q_start; q0 < Lq; q0 += TILE) {
+    __syncthreads();
+    load_tile(qp + (long)q0 * DHEAD, Lq - q0, sQ);
+    load_tile_T(qp + (long)q0 * DHEAD, Lq - q0, sQT);
+    load_tile(dop + (long)q0 * DHEAD, Lq - q0, sdO);
+    load_tile_T(dop + (long)q0 * DHEAD, Lq - q0, sdOT);
+    if (threadIdx.x < TILE) {
+      const int qr = q0 + threadIdx.x;
+      sLse[threadIdx.x] = (qr < Lq) ? lsep[qr] : 1e30f;
+      sDelta[threadIdx.x] = (qr < Lq) ? delp[qr] : 0.f;
+    }
+    __syncthreads();
+
+    bf16x8 kf0 = frag(sK, wkey0 + l16, kgrp * 8);
+    bf16x8 kf1 = frag(sK, wkey0 + l16, kgrp * 8 + 32);
+    bf16x8 vf0 = frag(sV, wkey0 + l16, kgrp * 8);
+    bf16x8 vf1 = frag(sV, wkey0 + l16, kgrp * 8 + 32);
+
+    f32x4_t dst_all[4];                  // dS^T frags per query subtile
+
+#pragma unroll
+    for (int ms = 0; ms < 4; ++ms) {
+      // S^T[key][m] = K Q^T: A=K (k=dim), B=Q rows (k=dim)
+      f32x4_t st = {0.f, 0.f, 0.f, 0.f};
+      st = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          kf0, frag(sQ, ms * 16 + l16, kgrp * 8), st, 0, 0, 0);
+      st = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          kf1, frag(sQ, ms * 16 + l16, kgrp * 8 + 32), st, 0, 0, 0);
+      // dP^T[key][m] = V dO^T: A=V (k=dv), B=dO rows (k=dv)
+      f32x4_t dpt = {0.f, 0.f, 0.f, 0.f};
+      dpt = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          vf0, frag(sdO, ms * 16 + l16, kgrp * 8), dpt, 0, 0, 0);
+      dpt = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          vf1, frag(sdO, ms * 16 + l16, kgrp * 8 + 32), dpt, 0, 0, 0);
+
+      const int mcol = ms * 16 + l16;
+      const float lse_m = sLse[mcol];
+      const float del_m = sDelta[mcol];
+      const int qrow = q0 + mcol;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int key = k0 + wkey0 + kgrp * 4 + r;
+        const bool masked = (key >= Lk) || (causal && key > qrow);
+        const float pt = masked ? 0.f : __expf(st[r] * scale - lse_m);
+        sPT[(wkey0 + kgrp * 4 + r) * PITCH + mcol] = f2bf_rne(pt);
+        dst_all[ms][r] = pt * (dpt[r] - del_m) * scale;
+      }
+    }
+
+    // dV[key][dv] += P^T dO: A = P^T (k=m, sPT rows), B = dO^T rows (k=m)
+#pragma unroll
+    for (int kc = 0; kc < 2; ++kc) {
+      bf16x8 ptf = frag(sPT, wkey0 + l16, kgrp * 8 + kc * 32);
+#pragma unroll
+      for (int ds_ = 0; ds_ < 4; ++ds_)
+        acc_dv[ds_] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            ptf, frag(sdOT, ds_ * 16 + l16, kgrp * 8 + kc * 32),
+            acc_dv[ds_], 0, 0, 0);
+    }
+
+    // overwrite sPT with dS^T (wave-local rows), then
+    // dK[key][dim] += dS^T Q: A = dS^T (k=m), B = Q^T rows (k=m)
+#pragma unroll
+    for (int ms = 0; ms < 4; ++ms)
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        sPT[(wkey0 + kgrp * 4 + r) * PITCH + ms * 16 + l16] =
+            f2bf_rne(dst_all[ms][r]);
+
+#pragma unroll
+    for (int kc = 0; kc < 2; ++kc) {
+      bf16x8 dsf = frag(sPT, wkey0 + l16, kgrp * 8 + kc * 32);
+#pragma unroll
+      for (int ds_ = 0; ds_ < 4; ++ds_)
+        acc_dk[ds_] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            dsf, frag(sQT, ds_ * 16 + l16, kgrp * 8 + kc * 32),
+            acc_dk[ds_], 0, 0, 0);
+    }
+  }
+
+  // store dK/dV rows (key-guarded)
+  bf16_t* dkp = dK + ((long)bh * Lk + k0) * DHEAD;
+  bf16_t* dvp = dV + ((long)bh * Lk + k0) * DHEAD;
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int key = wkey0 + kgrp * 4 + r;
+    if (k0 + key >= Lk) continue;
+#pragma unroll
+    for (int ds_ = 0; ds_ < 4; ++ds_) {
+      dkp[(long)key * DHEAD + ds_ * 16 + l16] = __float2bfloat16(acc_dk[ds_][r]);
+      dvp[(long)key * DHEAD + ds_ * 16 + l16] = __float2bfloat16(acc_dv[ds_][r]);
+    }
+  }
+}
+
+// ==========================================================================
+// Backward dQ. grid (ceil(Lq/64), BH); wave owns 16 q rows.
+// ==========================================================================
+__global__ __launch_bounds__(256)
+void attn_bwd_dq_kernel(const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
+                        const bf16_t* __restrict__ v, const bf16_t* __restrict__ dO,
+                        const float* __restrict__ lse, const float* __restrict__ delta,
+                        bf16_t* __restrict__ dQ, int Lq, int Lk, float scale,
+                        int causal) {
+  __shared__ short sQ[TILE * PITCH];
+  __shared__ short sdO[TILE * PITCH];
+  __shared__ short sK[TILE * PITCH];     // [key][dim]
+  __shared__ short sKT[TILE * PITCH];    // [dim][key]
+  __shared__ short sV[TILE * PITCH];     // [key][dim]
+  __shared__ short sDS[TILE * PITCH];    // [qrow][key]
+
+  const int bh = blockIdx.y;
+  const int q0 = blockIdx.x * TILE;
+  const bf16_t* qp = q + ((long)bh * Lq + q0) * DHEAD;
+  const bf16_t* kp = k + (long)bh * Lk * DHEAD;
+  const bf16_t* vp = v + (long)bh * Lk * DHEAD;
+  const bf16_t* dop = dO + ((long)bh * Lq + q0) * DHEAD;
+
+  load_tile(qp, Lq - q0, sQ);
+  load_tile(dop, Lq - q0, sdO);
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int l16 = lane & 15;
+  const int kgrp = lane >> 4;
+  const int wrow0 = wid * 16;
+
+  // per-lane row constants (rows m = kgrp*4 + r)
+  float lse_r[4], del_r[4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int qrow = q0 + wrow0 + kgrp * 4 + r;
+    lse_r[r] = (qrow < Lq) ? lse[(long)bh * Lq + qrow] : 1e30f;
+    del_r[r] = (qrow < Lq) ? delta[(long)bh * Lq + qrow] : 0.f;
+  }
+
+  f32x4_t acc_dq[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) acc_dq[i] = {0.f, 0.f, 0.f, 0.f};
+
+  const int kv_end = causal ? min(Lk, q0 + TILE) : Lk;
+
+  for (int kv0 = 0; kv0 < kv_end; kv0 += TILE) {
+    __syncthreads();
+    load_tile(kp + (long)kv0 * DHEAD, Lk - kv0, sK);
+    load_tile_T(kp + (long)kv0 * DHEAD, Lk - kv0, sKT);
+    load_tile(vp + (long)kv0 * DHEAD, Lk - kv0, sV);
+    __syncthreads();
+
+    bf16x8 qf0 = frag(sQ, wrow0 + l16, kgrp * 8);
+    bf16x8 qf1 = frag(sQ, wrow0 + l16, kgrp * 8 + 32);
+    bf16x8 df0 = frag(sdO, wrow0 + l16, kgrp * 8);
+    bf16x8 df1 = frag(sdO, wrow0 + l16, kgrp * 8 + 32);
+
+#pragma unroll
+    for (int ns = 0; ns < 4; ++ns) {
+      // S[m][key] = Q K^T: A=Q (k=dim), B=K rows (k=dim)
+      f32x4_t s = {0.f, 0.f, 0.f, 0.f};
+      s = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          qf0, frag(sK, ns * 16 + l16, kgrp * 8), s, 0, 0, 0);
+      s = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          qf1, frag(sK, ns * 16 + l16, kgrp * 8 + 32), s, 0, 0, 0);
+      // dP[m][key] = dO V^T: A=dO (k=dv), B=V rows (k=dv)
+      f32x4_t dp = {0.f, 0.f, 0.f, 0.f};
+      dp = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          df0, frag(sV, ns * 16 + l16, kgrp * 8), dp, 0, 0, 0);
+      dp = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          df1, frag(sV, ns * 16 + l16, kgrp * 8 + 32), dp, 0, 0, 0);
+
+      const int key = kv0 + ns * 16 + l16;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int qrow = q0 + wrow0 + kgrp * 4 + r;
+        const bool masked = (key >= Lk) || (causal && key > qrow);
+        const float p = masked ? 0.f : __expf(s[r] * scale - lse_r[r]);
+        sDS[(wrow0 + kgrp * 4 + r) * PITCH + ns * 16 + l16] =
+            f2bf_rne(p * (dp[r] - del_r[r]) * scale);
+      }
+    }
+
+    // dQ[m][dim] += dS K: A = dS (k=key, sDS rows), B = K^T rows (k=key)
+#pragma unroll
+    for (int kc = 0; kc < 2; ++kc) {
+      bf16x8 dsf = frag(sDS, wrow0 + l16, kgrp * 8 + kc * 32);
+#pragma unroll
+      for (int ds_ = 0; ds_ < 4; ++ds_)
+        acc_dq[ds_] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            dsf, frag(sKT, ds_ * 16 + l16, kgrp * 8 + kc * 32),
+            acc_dq[ds_], 0, 0, 0);
+    }
+  }
+
+  bf16_t* dqp = dQ + ((long)bh * Lq + q0) * DHEAD;
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int row = wrow0 + kgrp * 4 + r;
+    if (q0 + row >= Lq) continue;
+#pragma unroll
+    for (int ds_ = 0; ds_ < 4; ++ds_)
+      dqp[(long)row * DHEAD + ds_ * 16 + l16] = __float2bfloat16(acc_dq[ds_][r]);
+  }
+}
+
+// ==========================================================================
+// mfma layout probe: C[16][16] = A[16][32] @ B[32][16], my frag conventions.
+// Used by tests to pin the hardware fragment layout.
+// ==========================================================================
+__global__ void mfma_probe_kernel(const bf16_t* __restrict__ A,
+                                  const bf16_t* __restrict__ B,
+                                  float* __restrict__ C) {
+  const int l = threadIdx.x & 63;
+  bf16x8 a, b;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    a[j] = *reinterpret_cast<const short*>(A + (l & 15) * 32 + (l >> 4) * 8 + j);
+    b[j] = *reinterpret_cast<const short*>(B + ((l >> 4) * 8 + j) * 16 + (l & 15));
+  }
+  f32x4_t c = {0.f, 0.f, 0.f, 0.f};
+  c = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+#pragma unroll
+  for (int r = 0; r < 4; ++r) C[((l >> 4) * 4 + r) * 16 + (l & 15)] = c[r];
+}
+
+}  // namespace dcr_attn
+
+// ==========================================================================
+// Host launchers
+// ==========================================================================
+#include "dcr_launchers.h"
+
+namespace dcr {
+
+void attn_fwd_launch(const void* q, const void* k, const void* v, void* o,
+                     float* lse, int BH, int Lq, int Lk, float scale,
+                     bool causal, hipStream_t s) {
+  dim3 grid((Lq + TILE - 1) / TILE, BH), block(256);
+  hipLaunchKernelGGL(dcr_attn::attn_fwd_kernel, grid, block, 0, s,
+                     (const dcr_attn::bf16_t*)q, (const dcr_attn::bf16_t*)k,
+                     (const dcr_attn::bf16_t*)v, (dcr_attn::bf16_t*)o, lse,
+                     Lq, Lk, scale, causal ? 1 : 0);
+}
+
+void attn_bwd_launch(const void* q, const void* k, const void* v,
+                     const void* o, const void* dO, const float* lse,
+                     float* delta, void* dQ, void* dK, void* dV, int BH,
+                     int Lq, int Lk, float scale, bool causal, hipStream_t s) {
+  long rows = (long)BH * Lq;
+  dim3 gd((rows + 255) / 256), bd(256);
+  hipLaunchKernelGGL(dcr_attn::attn_bwd_delta_kernel, gd, bd, 0, s,
+                     (const dcr_attn::bf16_t*)dO, (const dcr_attn::bf16_t*)o,
+                     delta, rows);
+  dim3 g1((Lk + TILE - 1) / TILE, BH), b1(256);
+  hipLaunchKernelGGL(dcr_attn::attn_bwd_dkdv_kernel, g1, b1, 0, s,
+                     (const dcr_attn::bf16_t*)q, (const dcr_attn::bf16_t*)k,
+                     (const dcr_attn::bf16_t*)v, (const dcr_attn::bf16_t*)dO,
+                     lse, delta, (dcr_attn::bf16_t*)dK, (dcr_attn::bf16_t*)dV,
+                     Lq, Lk, scale, causal ? 1 : 0);
+  dim3 g2((Lq + TILE - 1) / TILE, BH), b2(256);
+  hipLaunchKernelGGL(dcr_attn::attn_bwd_dq_kernel, g2, b2, 0, s,
+                     (const dcr_attn::bf16_t*)q, (const dcr_attn::bf16_t*)k,
+                     (const dcr_attn::bf16_t*)v, (const dcr_attn::bf16_t*)dO,
+                     lse, delta, (dcr_attn::bf16_t*)dQ, Lq, Lk, scale,
+                     causal ? 1 : 0);
+}
+
+void mfma_probe_launch(const void* A, const void* B, float* C, hipStream_t s) {
+  hipLaunchKernelGGL(dcr_attn::mfma_probe_kernel, dim3(1), dim3(64), 0, s,
+                     (const dcr_attn::bf16_t*)A, (const dcr_attn::bf16_t*)B, C);
+}
+
+}  // namespace dcr

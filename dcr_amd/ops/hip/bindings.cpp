@@ -172,7 +172,54 @@ at::Tensor cfg_combine(at::Tensor eu, at::Tensor et, double scale) {
   return out;
 }
 
+// ---------------------------------------------------------------- attention
+std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
+                                 double scale, bool causal) {
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16,
+              "attn: bf16 CUDA only");
+  TORCH_CHECK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
+  TORCH_CHECK(q.size(-1) == 64 && k.size(-1) == 64, "attn: head_dim 64 only");
+  const int64_t Lq = q.size(-2), Lk = k.size(-2);
+  const int64_t BH = q.numel() / (Lq * 64);
+  auto o = at::empty_like(q);
+  auto lse = at::empty({BH, Lq}, q.options().dtype(at::kFloat));
+  attn_fwd_launch(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
+                  lse.data_ptr<float>(), (int)BH, (int)Lq, (int)Lk,
+                  (float)scale, causal, cur_stream());
+  return {o, lse};
+}
+
+std::vector<at::Tensor> attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
+                                 at::Tensor o, at::Tensor dO, at::Tensor lse,
+                                 double scale, bool causal) {
+  const int64_t Lq = q.size(-2), Lk = k.size(-2);
+  const int64_t BH = q.numel() / (Lq * 64);
+  auto dQ = at::empty_like(q);
+  auto dK = at::empty_like(k);
+  auto dV = at::empty_like(v);
+  auto delta = at::empty({BH, Lq}, q.options().dtype(at::kFloat));
+  attn_bwd_launch(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
+                  dO.contiguous().data_ptr(), lse.data_ptr<float>(),
+                  delta.data_ptr<float>(), dQ.data_ptr(), dK.data_ptr(),
+                  dV.data_ptr(), (int)BH, (int)Lq, (int)Lk, (float)scale,
+                  causal, cur_stream());
+  return {dQ, dK, dV};
+}
+
+at::Tensor mfma_probe(at::Tensor A, at::Tensor B) {
+  TORCH_CHECK(A.is_cuda() && A.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(A.sizes() == at::IntArrayRef({16, 32}) &&
+              B.sizes() == at::IntArrayRef({32, 16}));
+  auto C = at::zeros({16, 16}, A.options().dtype(at::kFloat));
+  mfma_probe_launch(A.contiguous().data_ptr(), B.contiguous().data_ptr(),
+                    C.data_ptr<float>(), cur_stream());
+  return C;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("attn_fwd", &attn_fwd);
+  mod.def("attn_bwd", &attn_bwd);
+  mod.def("mfma_probe", &mfma_probe);
   mod.def("groupnorm_silu_fwd", &groupnorm_silu_fwd);
   mod.def("groupnorm_silu_bwd", &groupnorm_silu_bwd);
   mod.def("layernorm_fwd", &layernorm_fwd);

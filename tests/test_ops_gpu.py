@@ -174,3 +174,99 @@ def test_ops_dispatch_uses_hip_on_gpu():
     y = ops.group_norm_silu(x, w, b, 8, 1e-5, True)
     y.sum().backward()
     assert x.grad is not None and torch.isfinite(x.grad).all()
+
+
+# ---------------------------------------------------------------- attention
+def _attn_ref(q, k, v, scale, causal):
+    s = (q.float() @ k.float().transpose(-2, -1)) * scale
+    if causal:
+        Lq, Lk = q.shape[-2], k.shape[-2]
+        mask = torch.ones(Lq, Lk, dtype=torch.bool, device=q.device).tril_()
+        s = s.masked_fill(~mask, float("-inf"))
+    p = s.softmax(dim=-1)
+    return p @ v.float()
+
+
+def test_mfma_probe_layout(ext):
+    """Pins the assumed gfx950 16x16x32 bf16 A/B/C fragment layouts."""
+    torch.manual_seed(0)
+    A = torch.randn(16, 32, device="cuda").to(torch.bfloat16)
+    B = torch.randn(32, 16, device="cuda").to(torch.bfloat16)  # asymmetric
+    C = ext.mfma_probe(A, B)
+    ref = A.float() @ B.float()
+    err = (C - ref).abs().max().item()
+    assert err < 0.1 * ref.abs().max().item(), f"mfma layout wrong: err={err}"
+
+
+@pytest.mark.parametrize("Lq,Lk,causal", [
+    (1024, 1024, False),   # SD self-attn @ 32x32 latents
+    (256, 256, False),
+    (64, 64, False),
+    (1024, 77, False),     # cross-attn, odd kv length
+    (77, 77, True),        # CLIP text, causal, odd length
+    (4096, 4096, False),   # 512px latents
+    (100, 200, False),     # both odd
+])
+def test_flash_attention_fwd(ext, Lq, Lk, causal):
+    torch.manual_seed(0)
+    B, H, D = 2, 5, 64
+    q = torch.randn(B, H, Lq, D, device="cuda").to(torch.bfloat16)
+    k = torch.randn(B, H, Lk, D, device="cuda").to(torch.bfloat16)
+    v = torch.randn(B, H, Lk, D, device="cuda").to(torch.bfloat16)
+    scale = 1.0 / D ** 0.5
+    o, lse = ext.attn_fwd(q.reshape(B * H, Lq, D), k.reshape(B * H, Lk, D),
+                          v.reshape(B * H, Lk, D), scale, causal)
+    ref = _attn_ref(q, k, v, scale, causal).reshape(B * H, Lq, D)
+    _close(o, ref, 2e-2)
+    # LSE check (non-masked rows)
+    s = (q.float() @ k.float().transpose(-2, -1)) * scale
+    if causal:
+        mask = torch.ones(Lq, Lk, dtype=torch.bool, device=q.device).tril_()
+        s = s.masked_fill(~mask, float("-inf"))
+    lse_ref = s.logsumexp(dim=-1).reshape(B * H, Lq)
+    assert (lse - lse_ref).abs().max().item() < 5e-2
+
+
+@pytest.mark.parametrize("Lq,Lk,causal", [
+    (256, 256, False),
+    (1024, 77, False),
+    (77, 77, True),
+    (100, 200, False),
+])
+def test_flash_attention_bwd(ext, Lq, Lk, causal):
+    torch.manual_seed(1)
+    B, H, D = 1, 4, 64
+    qf = torch.randn(B * H, Lq, D, device="cuda")
+    kf = torch.randn(B * H, Lk, D, device="cuda")
+    vf = torch.randn(B * H, Lk, D, device="cuda")
+    dO = torch.randn(B * H, Lq, D, device="cuda")
+    scale = 1.0 / D ** 0.5
+
+    qr = qf.clone().requires_grad_(True)
+    kr = kf.clone().requires_grad_(True)
+    vr = vf.clone().requires_grad_(True)
+    _attn_ref(qr, kr, vr, scale, causal).backward(dO)
+
+    q = qf.to(torch.bfloat16)
+    k = kf.to(torch.bfloat16)
+    v = vf.to(torch.bfloat16)
+    o, lse = ext.attn_fwd(q, k, v, scale, causal)
+    dQ, dK, dV = ext.attn_bwd(q, k, v, o, dO.to(torch.bfloat16), lse,
+                              scale, causal)
+    _close(dQ, qr.grad, 5e-2)
+    _close(dK, kr.grad, 5e-2)
+    _close(dV, vr.grad, 5e-2)
+
+
+def test_flash_attention_autograd_wrapper(ext):
+    from dcr_amd import ops
+    torch.manual_seed(2)
+    q = torch.randn(2, 5, 256, 64, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    k = torch.randn(2, 5, 77, 64, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    v = torch.randn_like(k, requires_grad=True)
+    out = ops.attention(q, k, v)
+    out.sum().backward()
+    for g in (q.grad, k.grad, v.grad):
+        assert g is not None and torch.isfinite(g.float()).all()
