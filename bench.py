@@ -32,6 +32,8 @@ def main():
     ap.add_argument("--batch-size", type=int, default=16)
     ap.add_argument("--resolution", type=int, default=256)
     ap.add_argument("--model", type=str, default="sd21", choices=["sd21", "tiny"])
+    ap.add_argument("--precision", type=str, default="bf16",
+                    choices=["bf16", "pure_bf16", "no"])
     args = ap.parse_args()
 
     from dcr_amd.parallel import dist as dist_utils
@@ -56,7 +58,7 @@ def main():
         synthetic_size=args.batch_size * 4,
         resolution=args.resolution,
         train_batch_size=args.batch_size,
-        mixed_precision="bf16" if use_cuda else "no",
+        mixed_precision=args.precision if use_cuda else "no",
         dataloader_num_workers=0,
         max_train_steps=10**9,
         seed=1234,

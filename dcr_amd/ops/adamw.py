@@ -55,6 +55,10 @@ class FusedAdamW:
         self.flat_grad = torch.zeros(total, device=device, dtype=dtype)
         self.exp_avg = torch.zeros(total, device=device, dtype=torch.float32)
         self.exp_avg_sq = torch.zeros(total, device=device, dtype=torch.float32)
+        # pure-bf16 training: bf16 model params/grads + fp32 master copy
+        self.master = None
+        if dtype == torch.bfloat16:
+            self.master = torch.empty(total, device=device, dtype=torch.float32)
 
         # Re-parameterize: params become views of flat_param, grads views of flat_grad.
         offset = 0
@@ -67,6 +71,8 @@ class FusedAdamW:
             self.offsets.append(offset)
             offset += n
         self.numel = total
+        if self.master is not None:
+            self.master.copy_(self.flat_param.float())
 
     # -- torch.optim-ish surface ------------------------------------------
     def zero_grad(self, set_to_none: bool = False):
@@ -82,10 +88,16 @@ class FusedAdamW:
         if use_hip(self.flat_param):
             m = require_hip("adamw")
             if m is not None:
-                m.adamw_step(
-                    self.flat_param, self.flat_grad, self.exp_avg, self.exp_avg_sq,
-                    self.lr, self.beta1, self.beta2, self.eps, self.weight_decay, t,
-                )
+                if self.master is not None:
+                    m.adamw_step_bf16(
+                        self.flat_param, self.flat_grad, self.master,
+                        self.exp_avg, self.exp_avg_sq, self.lr, self.beta1,
+                        self.beta2, self.eps, self.weight_decay, t)
+                else:
+                    m.adamw_step(
+                        self.flat_param, self.flat_grad, self.exp_avg, self.exp_avg_sq,
+                        self.lr, self.beta1, self.beta2, self.eps, self.weight_decay, t,
+                    )
                 return
         # reference implementation (CPU tests / debug fallback)
         bc1 = 1.0 - self.beta1 ** t
@@ -95,10 +107,14 @@ class FusedAdamW:
         self.exp_avg_sq.mul_(self.beta2).addcmul_(g, g, value=1 - self.beta2)
         denom = (self.exp_avg_sq / bc2).sqrt_().add_(self.eps)
         upd = (self.exp_avg / bc1) / denom
-        self.flat_param.add_(
-            (upd + self.weight_decay * self.flat_param.float()).to(self.flat_param.dtype),
-            alpha=-self.lr,
-        )
+        if self.master is not None:
+            self.master.add_(upd + self.weight_decay * self.master, alpha=-self.lr)
+            self.flat_param.copy_(self.master.to(self.flat_param.dtype))
+        else:
+            self.flat_param.add_(
+                (upd + self.weight_decay * self.flat_param.float()).to(self.flat_param.dtype),
+                alpha=-self.lr,
+            )
 
     @torch.no_grad()
     def clip_grad_norm_(self, max_norm: float) -> torch.Tensor:
@@ -111,13 +127,16 @@ class FusedAdamW:
 
     # -- checkpointing -----------------------------------------------------
     def state_dict(self):
-        return {
+        d = {
             "step": self.step_count,
             "lr": self.lr,
             "exp_avg": self.exp_avg,
             "exp_avg_sq": self.exp_avg_sq,
             "flat_param": self.flat_param,
         }
+        if self.master is not None:
+            d["master"] = self.master
+        return d
 
     def load_state_dict(self, sd):
         self.step_count = sd["step"]
@@ -125,3 +144,8 @@ class FusedAdamW:
         self.exp_avg.copy_(sd["exp_avg"])
         self.exp_avg_sq.copy_(sd["exp_avg_sq"])
         self.flat_param.copy_(sd["flat_param"])
+        if self.master is not None:
+            if "master" in sd:
+                self.master.copy_(sd["master"])
+            else:
+                self.master.copy_(self.flat_param.float())

@@ -141,6 +141,18 @@ void adamw_step(at::Tensor p, at::Tensor g, at::Tensor m, at::Tensor v,
                (float)beta2, (float)eps, (float)wd, step, cur_stream());
 }
 
+void adamw_step_bf16(at::Tensor p, at::Tensor g, at::Tensor master,
+                     at::Tensor m, at::Tensor v, double lr, double beta1,
+                     double beta2, double eps, double wd, int64_t step) {
+  TORCH_CHECK(p.is_cuda() && p.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(master.scalar_type() == at::kFloat);
+  TORCH_CHECK(p.is_contiguous() && g.is_contiguous() && master.is_contiguous());
+  adamw_bf16_launch(p.data_ptr(), g.data_ptr(), master.data_ptr<float>(),
+                    m.data_ptr<float>(), v.data_ptr<float>(), p.numel(),
+                    (float)lr, (float)beta1, (float)beta2, (float)eps,
+                    (float)wd, step, cur_stream());
+}
+
 // ------------------------------------------------------------- scheduler math
 static at::Tensor sched_common(int mode, at::Tensor a, at::Tensor b, at::Tensor ac,
                                at::Tensor t) {
@@ -227,6 +239,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("geglu_fwd", &geglu_fwd);
   mod.def("geglu_bwd", &geglu_bwd);
   mod.def("adamw_step", &adamw_step);
+  mod.def("adamw_step_bf16", &adamw_step_bf16);
   mod.def("add_noise", &add_noise);
   mod.def("get_velocity", &get_velocity);
   mod.def("cfg_combine", &cfg_combine);

@@ -32,6 +32,9 @@ void geglu_bwd_launch(DType dt, const void* dy, const void* x, void* dx,
 void adamw_launch(float* p, const float* g, float* m, float* v, long n,
                   float lr, float b1, float b2, float eps, float wd, long step,
                   hipStream_t s);
+void adamw_bf16_launch(void* p, const void* g, float* master, float* m,
+                       float* v, long n, float lr, float b1, float b2,
+                       float eps, float wd, long step, hipStream_t s);
 void sched_launch(DType dt, int mode, const void* x0, const void* noise,
                   const float* ac, const long* t, void* out, long per_sample,
                   long total, hipStream_t s);

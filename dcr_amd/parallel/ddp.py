@@ -85,8 +85,8 @@ class GradBucketAllReduce:
             self._param_bucket[id(p)] = b
 
     def _launch(self, b: _Bucket):
+        # SUM then post-scale (bf16 grads: pre-divide would lose mantissa)
         sl = self.opt.flat_grad[b.start:b.end]
-        sl.div_(self.world)
         b.work = dist.all_reduce(sl, group=self.pg, async_op=True)
 
     def _hook(self, param):
@@ -106,9 +106,12 @@ class GradBucketAllReduce:
             if b.work is None and b.pending > 0:
                 # params that never produced grads this step (e.g. frozen path)
                 self._launch(b)
+        launched = any(b.work is not None for b in self.buckets)
         for b in self.buckets:
             if b.work is not None:
                 b.work.wait()
+        if launched and self.world > 1:
+            self.opt.flat_grad.div_(self.world)
         self._reset()
 
     def _reset(self):

@@ -65,7 +65,9 @@ class Trainer:
 
         self.weight_dtype = {
             "no": torch.float32, "fp16": torch.float16, "bf16": torch.bfloat16,
+            "pure_bf16": torch.bfloat16,
         }[cfg.mixed_precision]
+        self.pure_bf16 = cfg.mixed_precision == "pure_bf16"
 
         self._build_models()
         self._build_data()
@@ -105,7 +107,8 @@ class Trainer:
         if not self.cfg.train_text_encoder:
             self.text_encoder.requires_grad_(False)
 
-        self.unet.to(self.device)
+        self.unet.to(self.device,
+                     dtype=torch.bfloat16 if self.pure_bf16 else None)
         # frozen models run in the compute dtype (reference: diff_train.py:531-533)
         self.vae.to(self.device, dtype=self.weight_dtype)
         self.text_encoder.to(
@@ -179,7 +182,7 @@ class Trainer:
         """One micro-step; returns the (detached) loss."""
         cfg = self.cfg
         device_type = self.device.type
-        autocast_on = self.weight_dtype != torch.float32
+        autocast_on = self.weight_dtype != torch.float32 and not self.pure_bf16
 
         pixel_values = batch["pixel_values"].to(self.device, non_blocking=True)
         input_ids = batch["input_ids"].to(self.device, non_blocking=True)

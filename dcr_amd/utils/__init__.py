@@ -2,3 +2,4 @@ from .logging import SmoothedValue, MetricLogger, Tracker
 from .image import tensor_to_pil, concat_h, image_grid
 
 __all__ = ["SmoothedValue", "MetricLogger", "Tracker", "tensor_to_pil", "concat_h", "image_grid"]
+from .profiler import PhaseProfiler

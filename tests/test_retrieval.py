@@ -166,3 +166,14 @@ def test_sharded_topk_matches_full():
     ref = (q @ shard.t()).topk(3, dim=1)
     assert torch.allclose(v, ref.values, atol=1e-5)
     assert torch.equal(i, ref.indices + 1000)
+
+
+def test_compute_map():
+    from dcr_amd.retrieval.map_eval import compute_map
+    # 4 db items, 2 queries; query0: ok={0,1} ranked first -> AP 1.0
+    ranks = np.array([[0, 3], [1, 2], [2, 1], [3, 0]])
+    gnd = [{"ok": [0, 1], "junk": []}, {"ok": [3], "junk": [2]}]
+    m, aps, _, _ = compute_map(ranks, gnd)
+    assert aps[0] == 1.0
+    assert aps[1] == 1.0  # junk(2) ranked above ok(3) is ignored
+    assert m == 1.0

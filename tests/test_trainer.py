@@ -117,3 +117,15 @@ def test_fit_writes_checkpoint_layout(tmp_path):
     logf = out / "diffrep_ft_log.jsonl"
     recs = [json.loads(l) for l in logf.read_text().splitlines()]
     assert any("loss" in r and "lr" in r for r in recs)
+
+
+def test_pure_bf16_mode(tmp_path):
+    """bf16 params + fp32 master: step runs, master tracks params."""
+    tr = Trainer(tiny_cfg(tmp_path, mixed_precision="pure_bf16"))
+    assert tr.optimizer.master is not None
+    b = next(iter(tr.dataloader))
+    l1 = tr.train_step(b)
+    l2 = tr.train_step(b)
+    assert torch.isfinite(l1) and torch.isfinite(l2)
+    assert torch.allclose(tr.optimizer.flat_param.float(),
+                          tr.optimizer.master, atol=1e-2)
