@@ -60,6 +60,11 @@ class Trainer:
             device = torch.device("cuda", dist_utils.get_local_rank()) \
                 if torch.cuda.is_available() else torch.device("cpu")
         self.device = device
+        if self.device.type == "cuda":
+            # MIOpen exhaustive find on first occurrence of each conv shape:
+            # measured +33% step throughput vs FAST find mode on MI355X
+            # (profiles/r01: workspace-limited fallback solvers otherwise).
+            torch.backends.cudnn.benchmark = True
         if cfg.seed is not None:
             set_seed(cfg.seed + self.rank)
 
