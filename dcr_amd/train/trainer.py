@@ -127,12 +127,20 @@ class Trainer:
             self._enable_grad_ckpt()
 
     def _enable_grad_ckpt(self):
+        # wrap pure submodules (resnets/attentions) — whole up/down blocks
+        # mutate the skip-connection list, which breaks checkpoint replay
         from torch.utils.checkpoint import checkpoint
-        for blk in list(self.unet.down_blocks) + [self.unet.mid_block] + list(self.unet.up_blocks):
-            orig = blk.forward
-            def wrapped(*args, _orig=orig, **kw):
-                return checkpoint(_orig, *args, use_reentrant=False, **kw)
-            blk.forward = wrapped
+        from ..models.attention import Transformer2DModel
+        from ..models.resnet import ResnetBlock2D
+
+        for mod in self.unet.modules():
+            if isinstance(mod, (ResnetBlock2D, Transformer2DModel)):
+                orig = mod.forward
+
+                def wrapped(*args, _orig=orig, **kw):
+                    return checkpoint(_orig, *args, use_reentrant=False, **kw)
+
+                mod.forward = wrapped
 
     def _build_data(self):
         cfg = self.cfg

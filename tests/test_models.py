@@ -97,3 +97,23 @@ def test_checkpoint_roundtrip(tmp_path):
     assert sd1.keys() == sd2.keys()
     for k in sd1:
         assert torch.equal(sd1[k], sd2[k])
+
+
+def test_sd14_unet_family():
+    """SD-1.4 config: 8 heads/block, 768-d cross-attn, conv projections."""
+    cfg = UNetConfig.sd14()
+    assert cfg.cross_attention_dim == 768 and not cfg.use_linear_projection
+    # param count of CompVis/stable-diffusion-v1-4 unet
+    unet = UNet2DConditionModel(cfg)
+    assert sum(p.numel() for p in unet.parameters()) == 859_520_964
+
+
+def test_sd14_tiny_forward_shape():
+    import dcr_amd.models.unet as U
+    cfg = U.UNetConfig(sample_size=8, block_out_channels=(32, 64, 64, 64),
+                       attention_head_dim=(8, 8, 8, 8), cross_attention_dim=48,
+                       norm_num_groups=8, layers_per_block=1,
+                       use_linear_projection=False)
+    unet = U.UNet2DConditionModel(cfg)
+    out = unet(torch.randn(1, 4, 8, 8), torch.tensor([5]), torch.randn(1, 7, 48))
+    assert out.shape == (1, 4, 8, 8)

@@ -129,3 +129,26 @@ def test_pure_bf16_mode(tmp_path):
     assert torch.isfinite(l1) and torch.isfinite(l2)
     assert torch.allclose(tr.optimizer.flat_param.float(),
                           tr.optimizer.master, atol=1e-2)
+
+
+def test_determinism_seeded_losses(tmp_path):
+    """seeded runs reproduce the loss sequence."""
+    def run():
+        cfg = tiny_cfg(tmp_path, seed=123)
+        tr = Trainer(cfg)
+        out = []
+        for batch in tr.dataloader:
+            out.append(tr.train_step(batch).item())
+            if tr.global_step >= 3:
+                break
+        return out
+
+    assert run() == run()
+
+
+def test_gradient_checkpointing(tmp_path):
+    tr = Trainer(tiny_cfg(tmp_path, gradient_checkpointing=True))
+    b = next(iter(tr.dataloader))
+    loss = tr.train_step(b)
+    assert torch.isfinite(loss)
+    assert tr.optimizer.flat_grad.abs().sum() == 0  # zeroed after step
