@@ -181,6 +181,31 @@ def main():
             torch.save({"entropy": ent, "jpeg": jpg, "tv": tv},
                        out_dir / "complexity.pth")
 
+    # duplication analysis (reference :562-583): are the up-weighted
+    # (duplicated) training images matched more often / more strongly?
+    if not args.noeval:
+        import glob as _glob
+        import pickle as _pickle
+        wpicks = sorted(_glob.glob(str(Path(args.val_dir) / "weights_*.pickle")))
+        if wpicks:
+            with open(wpicks[0], "rb") as fh:
+                weights = _pickle.load(fh)
+            if len(weights) == sim.shape[1]:
+                w = torch.tensor([float(x) for x in weights])
+                dup_mask = w > 1
+                match_idx = sim.argmax(dim=1)
+                matched_dup = dup_mask[match_idx].float().mean().item()
+                dup_stats = {
+                    "dup_frac_of_train": dup_mask.float().mean().item(),
+                    "dup_matched_frac": matched_dup,
+                    "sim_to_dup_mean": sim[:, dup_mask].max(dim=1).values.mean().item()
+                        if dup_mask.any() else 0.0,
+                    "sim_to_nodup_mean": sim[:, ~dup_mask].max(dim=1).values.mean().item()
+                        if (~dup_mask).any() else 0.0,
+                }
+                tracker.log(dup_stats)
+                print(json.dumps(dup_stats, indent=2))
+
     # FID (reference :597-600; HOT LOOP 3)
     if not args.noeval and not args.skip_fid:
         from dcr_amd.metrics import calculate_fid_given_paths
