@@ -107,6 +107,11 @@ def main():
         torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
         elapsed = float(t.item())
 
+    phases = trainer.prof.summary()
+    if phases and rank == 0:
+        print("phase ms/step:", json.dumps({k: round(v, 2) for k, v in phases.items()}),
+              file=sys.stderr)
+
     ms_per_step = elapsed / args.steps * 1000.0
     imgs_per_sec = args.batch_size * n_gpus * args.steps / elapsed
 

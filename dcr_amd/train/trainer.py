@@ -40,6 +40,7 @@ from ..ops.adamw import FusedAdamW
 from ..parallel import GradBucketAllReduce, dist as dist_utils
 from ..schedulers import DDPMScheduler
 from ..utils import Tracker
+from ..utils.profiler import PhaseProfiler
 from .config import TrainConfig, get_lr
 
 
@@ -84,6 +85,7 @@ class Trainer:
 
         self.global_step = 0
         self.tracker = None
+        self.prof = PhaseProfiler()  # enabled via DCR_PROFILE=1
 
     # ------------------------------------------------------------------
     def _build_models(self):
@@ -203,7 +205,7 @@ class Trainer:
         self.ddp.require_backward_grad_sync = sync_gradients
 
         with torch.autocast(device_type, dtype=self.weight_dtype, enabled=autocast_on):
-            with torch.no_grad():
+            with torch.no_grad(), self.prof.phase("vae_encode"):
                 latents = self.vae.encode(
                     pixel_values.to(self.weight_dtype)).latent_dist.sample()
                 latents = latents * self.vae.config.scaling_factor
@@ -215,7 +217,8 @@ class Trainer:
                     device=self.device, dtype=torch.long)
                 noisy_latents = self.noise_scheduler.add_noise(latents, noise, timesteps)
 
-            with torch.set_grad_enabled(cfg.train_text_encoder):
+            with torch.set_grad_enabled(cfg.train_text_encoder), \
+                    self.prof.phase("text_encode"):
                 encoder_hidden_states = self.text_encoder(input_ids)[0]
             if cfg.rand_noise_lam > 0:
                 encoder_hidden_states = encoder_hidden_states + \
@@ -226,7 +229,8 @@ class Trainer:
                 encoder_hidden_states = lam * encoder_hidden_states + \
                     (1 - lam) * encoder_hidden_states[index]
 
-            model_pred = self.unet(noisy_latents, timesteps, encoder_hidden_states)
+            with self.prof.phase("unet_fwd"):
+                model_pred = self.unet(noisy_latents, timesteps, encoder_hidden_states)
 
             if self.noise_scheduler.prediction_type == "epsilon":
                 target = noise
@@ -236,13 +240,15 @@ class Trainer:
                 raise ValueError(self.noise_scheduler.prediction_type)
 
         loss = F.mse_loss(model_pred.float(), target.float(), reduction="mean")
-        loss.backward()
+        with self.prof.phase("backward"):
+            loss.backward()
 
         if sync_gradients:
-            self.ddp.finalize()
-            self.optimizer.clip_grad_norm_(cfg.max_grad_norm)
-            self.optimizer.step(lr=get_lr(cfg, self.global_step))
-            self.optimizer.zero_grad()
+            with self.prof.phase("optimizer"):
+                self.ddp.finalize()
+                self.optimizer.clip_grad_norm_(cfg.max_grad_norm)
+                self.optimizer.step(lr=get_lr(cfg, self.global_step))
+                self.optimizer.zero_grad()
             self.global_step += 1
         return loss.detach()
 
