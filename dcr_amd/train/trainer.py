@@ -254,6 +254,22 @@ class Trainer:
 
     # ------------------------------------------------------------------
     def fit(self, max_steps: Optional[int] = None, sample_fn=None):
+        """Train loop. Failure policy: any exception is
+        logged and the process group torn down so peer ranks abort on
+        their next collective instead of hanging until the RCCL timeout;
+        torchrun then propagates the non-zero exit."""
+        try:
+            self._fit_inner(max_steps, sample_fn)
+        except Exception:
+            import traceback
+            print(f"rank {self.rank}: training failed\n{traceback.format_exc()}",
+                  flush=True)
+            if dist_utils.is_dist():
+                import torch.distributed as dist
+                dist.destroy_process_group()
+            raise
+
+    def _fit_inner(self, max_steps: Optional[int] = None, sample_fn=None):
         cfg = self.cfg
         max_steps = max_steps or cfg.max_train_steps
         out_dir = Path(cfg.output_dir)
