@@ -34,6 +34,7 @@ def main():
     ap.add_argument("--model", type=str, default="sd21", choices=["sd21", "tiny"])
     ap.add_argument("--precision", type=str, default="bf16",
                     choices=["bf16", "pure_bf16", "no"])
+    ap.add_argument("--channels-last", action="store_true")
     args = ap.parse_args()
 
     from dcr_amd.parallel import dist as dist_utils
@@ -65,6 +66,7 @@ def main():
         class_prompt="instancelevel_blip",
         duplication="nodup",
         output_dir="/tmp/dcr_bench_out",
+        channels_last=args.channels_last,
     )
     trainer = Trainer(cfg, device=device)
     trainer.unet.train()

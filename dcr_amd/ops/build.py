@@ -20,6 +20,7 @@ TARGET = HERE / "_dcr_hip.so"
 SOURCES = [
     HIP_DIR / "bindings.cpp",
     HIP_DIR / "norms.hip",
+    HIP_DIR / "norms_nhwc.hip",
     HIP_DIR / "elementwise.hip",
     HIP_DIR / "attention.hip",
 ]

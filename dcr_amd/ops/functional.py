@@ -26,6 +26,30 @@ from . import use_hip, require_hip
 # --------------------------------------------------------------------------
 # GroupNorm (+ optional fused SiLU)
 # --------------------------------------------------------------------------
+class _GroupNormSiLUNHWC(torch.autograd.Function):
+    """channels_last path: stats/apply kernels over the native NHWC walk
+    (dcr_amd/ops/hip/norms_nhwc.hip) — no transpose round-trips."""
+
+    @staticmethod
+    def forward(ctx, x, weight, bias, num_groups, eps, apply_silu):
+        m = require_hip("group_norm_silu_nhwc")
+        y, mean, rstd = m.groupnorm_silu_nhwc_fwd(x, weight, bias, num_groups,
+                                                  eps, apply_silu)
+        ctx.save_for_backward(x, weight, bias, mean, rstd)
+        ctx.num_groups = num_groups
+        ctx.apply_silu = apply_silu
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight, bias, mean, rstd = ctx.saved_tensors
+        m = require_hip("group_norm_silu_nhwc")
+        dx, dw, db = m.groupnorm_silu_nhwc_bwd(
+            dy.contiguous(memory_format=torch.channels_last), x, weight, bias,
+            mean, rstd, ctx.num_groups, ctx.apply_silu)
+        return dx, dw, db, None, None, None
+
+
 class _GroupNormSiLU(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, bias, num_groups, eps, apply_silu):
@@ -65,6 +89,10 @@ def group_norm_silu(
     apply_silu: bool = True,
 ) -> torch.Tensor:
     if use_hip(x):
+        if x.dim() == 4 and x.is_contiguous(memory_format=torch.channels_last) \
+                and not x.is_contiguous() and x.shape[1] % 4 == 0:
+            return _GroupNormSiLUNHWC.apply(x, weight, bias, num_groups, eps,
+                                            apply_silu)
         return _GroupNormSiLU.apply(x.contiguous(), weight, bias, num_groups, eps, apply_silu)
     return _gn_silu_ref(x, weight, bias, num_groups, eps, apply_silu)
 

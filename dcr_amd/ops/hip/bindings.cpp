@@ -69,6 +69,51 @@ std::vector<at::Tensor> groupnorm_silu_bwd(at::Tensor dy, at::Tensor x, at::Tens
   return {dx, dw.to(w.scalar_type()), db.to(b.scalar_type())};
 }
 
+// channels_last GroupNorm: x is NHWC-contiguous, passed as [N, C, H, W]
+// logical sizes with channels_last memory format.
+std::vector<at::Tensor> groupnorm_silu_nhwc_fwd(at::Tensor x, at::Tensor w,
+                                                at::Tensor b, int64_t groups,
+                                                double eps, bool silu) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4 &&
+              x.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "gn_nhwc: channels_last 4D expected");
+  const int64_t N = x.size(0), C = x.size(1);
+  const int64_t R = x.size(2) * x.size(3);
+  TORCH_CHECK(C % groups == 0 && C % 4 == 0);
+  auto wf = as_f32(w), bf = as_f32(b);
+  auto y = at::empty_like(x);  // preserves channels_last
+  auto ws = at::zeros({N * groups * 2}, x.options().dtype(at::kFloat));
+  auto mean = at::empty({N * groups}, x.options().dtype(at::kFloat));
+  auto rstd = at::empty_like(mean);
+  gn_nhwc_fwd_launch(dtype_of(x), x.data_ptr(), wf.data_ptr<float>(),
+                     bf.data_ptr<float>(), y.data_ptr(), ws.data_ptr<float>(),
+                     mean.data_ptr<float>(), rstd.data_ptr<float>(), (int)N,
+                     (int)R, (int)C, (int)groups, (float)eps, silu, cur_stream());
+  return {y, mean, rstd};
+}
+
+std::vector<at::Tensor> groupnorm_silu_nhwc_bwd(at::Tensor dy, at::Tensor x,
+                                                at::Tensor w, at::Tensor b,
+                                                at::Tensor mean, at::Tensor rstd,
+                                                int64_t groups, bool silu) {
+  TORCH_CHECK(dy.is_contiguous(at::MemoryFormat::ChannelsLast) &&
+              x.is_contiguous(at::MemoryFormat::ChannelsLast));
+  const int64_t N = x.size(0), C = x.size(1);
+  const int64_t R = x.size(2) * x.size(3);
+  auto wf = as_f32(w), bf = as_f32(b);
+  auto dx = at::empty_like(x);
+  auto ws = at::zeros({N * groups * 2}, x.options().dtype(at::kFloat));
+  auto dw = at::zeros({C}, x.options().dtype(at::kFloat));
+  auto db = at::zeros({C}, x.options().dtype(at::kFloat));
+  gn_nhwc_bwd_launch(dtype_of(x), dy.data_ptr(), x.data_ptr(),
+                     wf.data_ptr<float>(), bf.data_ptr<float>(),
+                     mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                     ws.data_ptr<float>(), dx.data_ptr(), dw.data_ptr<float>(),
+                     db.data_ptr<float>(), (int)N, (int)R, (int)C, (int)groups,
+                     silu, cur_stream());
+  return {dx, dw.to(w.scalar_type()), db.to(b.scalar_type())};
+}
+
 // ---------------------------------------------------------------- LayerNorm
 std::vector<at::Tensor> layernorm_fwd(at::Tensor x, at::Tensor w, at::Tensor b,
                                       double eps) {
@@ -234,6 +279,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("mfma_probe", &mfma_probe);
   mod.def("groupnorm_silu_fwd", &groupnorm_silu_fwd);
   mod.def("groupnorm_silu_bwd", &groupnorm_silu_bwd);
+  mod.def("groupnorm_silu_nhwc_fwd", &groupnorm_silu_nhwc_fwd);
+  mod.def("groupnorm_silu_nhwc_bwd", &groupnorm_silu_nhwc_bwd);
   mod.def("layernorm_fwd", &layernorm_fwd);
   mod.def("layernorm_bwd", &layernorm_bwd);
   mod.def("geglu_fwd", &geglu_fwd);

@@ -24,6 +24,15 @@ void ln_bwd_launch(DType dt, const void* dy, const void* x, const float* w,
                    const float* mean, const float* rstd, void* dx, float* dw,
                    float* db, long M, int N, hipStream_t s);
 
+// norms_nhwc.hip (channels_last GroupNorm)
+void gn_nhwc_fwd_launch(DType dt, const void* x, const float* w, const float* b,
+                        void* y, float* ws, float* mean, float* rstd, int N,
+                        int R, int C, int G, float eps, bool silu, hipStream_t s);
+void gn_nhwc_bwd_launch(DType dt, const void* dy, const void* x, const float* w,
+                        const float* b, const float* mean, const float* rstd,
+                        float* ws, void* dx, float* dw, float* db, int N,
+                        int R, int C, int G, bool silu, hipStream_t s);
+
 // elementwise.hip
 void geglu_fwd_launch(DType dt, const void* x, void* y, long M, long N,
                       hipStream_t s);

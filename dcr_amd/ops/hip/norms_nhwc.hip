@@ -1,0 +1,276 @@
+// GroupNorm(+fused SiLU) for channels_last (NHWC) tensors on MI355X.
+//
+// Rationale: MIOpen's fast igemm conv kernels are NHWC; with NCHW models
+// every conv pays batched_transpose round-trips (profiles/prof4: ~5.6
+// ms/step). channels_last removes them — but then GroupNorm must reduce
+// over a strided [HW, Cg] slab. Design:
+//   stats:   grid (row-chunks, N); threads own CHANNELS (consecutive c
+//            => coalesced), register-accumulate over the chunk's rows,
+//            LDS per-channel sums -> per-group partials -> global atomics
+//            into a [N, G, 2] fp32 workspace. No per-element atomics.
+//   finalize:[N*G] -> mean/rstd.
+//   apply:   flat vec4 elementwise (fully coalesced NHWC walk).
+// Backward mirrors it (stats also emit per-channel dw/db).
+
+#include "dcr_common.h"
+
+using namespace dcr;
+
+namespace dcr_nhwc {
+
+// --------------------------------------------------------------- stats fwd
+template <typename T>
+__global__ void gn_nhwc_stats_kernel(const T* __restrict__ x, float* __restrict__ ws,
+                                     int N, int R, int C, int G, int rows_per_blk) {
+  extern __shared__ float smem[];          // [2*C]
+  float* s1 = smem;
+  float* s2 = smem + C;
+  const int n = blockIdx.y;
+  const int r0 = blockIdx.x * rows_per_blk;
+  const int r1 = min(R, r0 + rows_per_blk);
+  const long base = (long)n * R * C;
+  const int Cg = C / G;
+
+  for (int c = threadIdx.x; c < C; c += blockDim.x) {
+    float a = 0.f, b = 0.f;
+    for (int r = r0; r < r1; ++r) {
+      float v = to_f32<T>(x[base + (long)r * C + c]);
+      a += v;
+      b += v * v;
+    }
+    s1[c] = a;
+    s2[c] = b;
+  }
+  __syncthreads();
+  for (int g = threadIdx.x; g < G; g += blockDim.x) {
+    float a = 0.f, b = 0.f;
+    for (int c = g * Cg; c < (g + 1) * Cg; ++c) { a += s1[c]; b += s2[c]; }
+    atomicAdd(&ws[((long)n * G + g) * 2], a);
+    atomicAdd(&ws[((long)n * G + g) * 2 + 1], b);
+  }
+}
+
+__global__ void gn_finalize_kernel(const float* __restrict__ ws,
+                                   float* __restrict__ mean, float* __restrict__ rstd,
+                                   long NG, float inv_L, float eps) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= NG) return;
+  float m = ws[i * 2] * inv_L;
+  float var = fmaxf(ws[i * 2 + 1] * inv_L - m * m, 0.f);
+  mean[i] = m;
+  rstd[i] = rsqrtf(var + eps);
+}
+
+// --------------------------------------------------------------- apply fwd
+template <typename T, bool SILU>
+__global__ void gn_nhwc_apply_kernel(const T* __restrict__ x, const float* __restrict__ w,
+                                     const float* __restrict__ b,
+                                     const float* __restrict__ mean,
+                                     const float* __restrict__ rstd,
+                                     T* __restrict__ y, long total, int R, int C, int G) {
+  const int Cg = C / G;
+  const long nvec = total / 4;
+  const long RC = (long)R * C;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
+       i += (long)gridDim.x * blockDim.x) {
+    long e = i * 4;
+    long n = e / RC;
+    int c = (int)(e % C);                  // C % 4 == 0: 4 consecutive c
+    f32x4 xv = load4<T>(x + e);
+    f32x4 wv = load4<float>(w + c);
+    f32x4 bv = load4<float>(b + c);
+    f32x4 o;
+#pragma unroll
+    for (int k = 0; k < 4; ++k) {
+      int g = (c + k) / Cg;
+      float m = mean[n * G + g];
+      float rs = rstd[n * G + g];
+      float z = (( &xv.x)[k] - m) * rs * (&wv.x)[k] + (&bv.x)[k];
+      (&o.x)[k] = SILU ? silu(z) : z;
+    }
+    store4<T>(y + e, o);
+  }
+}
+
+// --------------------------------------------------------------- stats bwd
+// per-channel dw/db; per-group S1 = sum(w*dz), S2 = sum(w*dz*yhat)
+template <typename T, bool SILU>
+__global__ void gn_nhwc_bwd_stats_kernel(const T* __restrict__ dy, const T* __restrict__ x,
+                                         const float* __restrict__ w, const float* __restrict__ b_,
+                                         const float* __restrict__ mean,
+                                         const float* __restrict__ rstd,
+                                         float* __restrict__ ws, float* __restrict__ dw,
+                                         float* __restrict__ db,
+                                         int N, int R, int C, int G, int rows_per_blk) {
+  extern __shared__ float smem[];          // [2*C]
+  float* sa = smem;                        // S1 per channel (w*dz)
+  float* sb = smem + C;                    // S2 per channel (w*dz*yhat)
+  const int n = blockIdx.y;
+  const int r0 = blockIdx.x * rows_per_blk;
+  const int r1 = min(R, r0 + rows_per_blk);
+  const long base = (long)n * R * C;
+  const int Cg = C / G;
+
+  for (int c = threadIdx.x; c < C; c += blockDim.x) {
+    const int g = c / Cg;
+    const float m = mean[(long)n * G + g];
+    const float rs = rstd[(long)n * G + g];
+    const float wc = w[c], bc = b_[c];
+    float a = 0.f, bb = 0.f, dwc = 0.f, dbc = 0.f;
+    for (int r = r0; r < r1; ++r) {
+      long idx = base + (long)r * C + c;
+      float yh = (to_f32<T>(x[idx]) - m) * rs;
+      float dz = to_f32<T>(dy[idx]);
+      if (SILU) dz *= dsilu(yh * wc + bc);
+      float gx = dz * wc;
+      a += gx;
+      bb += gx * yh;
+      dwc += dz * yh;
+      dbc += dz;
+    }
+    sa[c] = a;
+    sb[c] = bb;
+    atomicAdd(&dw[c], dwc);
+    atomicAdd(&db[c], dbc);
+  }
+  __syncthreads();
+  for (int g = threadIdx.x; g < G; g += blockDim.x) {
+    float a = 0.f, bb = 0.f;
+    for (int c = g * Cg; c < (g + 1) * Cg; ++c) { a += sa[c]; bb += sb[c]; }
+    atomicAdd(&ws[((long)n * G + g) * 2], a);
+    atomicAdd(&ws[((long)n * G + g) * 2 + 1], bb);
+  }
+}
+
+// --------------------------------------------------------------- apply bwd
+template <typename T, bool SILU>
+__global__ void gn_nhwc_bwd_apply_kernel(const T* __restrict__ dy, const T* __restrict__ x,
+                                         const float* __restrict__ w, const float* __restrict__ b_,
+                                         const float* __restrict__ mean,
+                                         const float* __restrict__ rstd,
+                                         const float* __restrict__ ws,
+                                         T* __restrict__ dx, long total, int R, int C,
+                                         int G, float inv_L) {
+  const int Cg = C / G;
+  const long nvec = total / 4;
+  const long RC = (long)R * C;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
+       i += (long)gridDim.x * blockDim.x) {
+    long e = i * 4;
+    long n = e / RC;
+    int c = (int)(e % C);
+    f32x4 xv = load4<T>(x + e);
+    f32x4 gv = load4<T>(dy + e);
+    f32x4 wv = load4<float>(w + c);
+    f32x4 bv = load4<float>(b_ + c);
+    f32x4 o;
+#pragma unroll
+    for (int k = 0; k < 4; ++k) {
+      int g = (c + k) / Cg;
+      float m = mean[n * G + g];
+      float rs = rstd[n * G + g];
+      float m1 = ws[(n * G + g) * 2] * inv_L;
+      float m2 = ws[(n * G + g) * 2 + 1] * inv_L;
+      float yh = ((&xv.x)[k] - m) * rs;
+      float dz = (&gv.x)[k];
+      if (SILU) dz *= dsilu(yh * (&wv.x)[k] + (&bv.x)[k]);
+      float gx = dz * (&wv.x)[k];
+      (&o.x)[k] = rs * (gx - m1 - yh * m2);
+    }
+    store4<T>(dx + e, o);
+  }
+}
+
+}  // namespace dcr_nhwc
+
+// ==========================================================================
+// Host launchers
+// ==========================================================================
+#include "dcr_launchers.h"
+
+namespace dcr {
+
+static inline int nhwc_row_chunks(int N, int R) {
+  // target >= 512 workgroups to fill 256 CUs / 8 XCDs
+  int chunks = (512 + N - 1) / N;
+  if (chunks > R) chunks = R;
+  if (chunks < 1) chunks = 1;
+  return chunks;
+}
+
+template <typename T>
+static void gn_nhwc_fwd_t(const void* x, const float* w, const float* b, void* y,
+                          float* ws, float* mean, float* rstd, int N, int R,
+                          int C, int G, float eps, bool silu, hipStream_t s) {
+  int chunks = nhwc_row_chunks(N, R);
+  int rows_per_blk = (R + chunks - 1) / chunks;
+  dim3 grid(chunks, N), block(256);
+  size_t lds = 2 * (size_t)C * sizeof(float);
+  hipLaunchKernelGGL((dcr_nhwc::gn_nhwc_stats_kernel<T>), grid, block, lds, s,
+                     (const T*)x, ws, N, R, C, G, rows_per_blk);
+  long NG = (long)N * G;
+  float inv_L = 1.f / ((float)R * (C / G));
+  hipLaunchKernelGGL(dcr_nhwc::gn_finalize_kernel, dim3((NG + 255) / 256),
+                     dim3(256), 0, s, ws, mean, rstd, NG, inv_L, eps);
+  long total = (long)N * R * C;
+  dim3 agrid((int)min((total / 4 + 255) / 256, (long)8192)), ablock(256);
+  if (silu)
+    hipLaunchKernelGGL((dcr_nhwc::gn_nhwc_apply_kernel<T, true>), agrid, ablock, 0, s,
+                       (const T*)x, w, b, mean, rstd, (T*)y, total, R, C, G);
+  else
+    hipLaunchKernelGGL((dcr_nhwc::gn_nhwc_apply_kernel<T, false>), agrid, ablock, 0, s,
+                       (const T*)x, w, b, mean, rstd, (T*)y, total, R, C, G);
+}
+
+void gn_nhwc_fwd_launch(DType dt, const void* x, const float* w, const float* b,
+                        void* y, float* ws, float* mean, float* rstd, int N,
+                        int R, int C, int G, float eps, bool silu, hipStream_t s) {
+  switch (dt) {
+    case DT_F32: gn_nhwc_fwd_t<float>(x, w, b, y, ws, mean, rstd, N, R, C, G, eps, silu, s); break;
+    case DT_F16: gn_nhwc_fwd_t<__half>(x, w, b, y, ws, mean, rstd, N, R, C, G, eps, silu, s); break;
+    case DT_BF16: gn_nhwc_fwd_t<__hip_bfloat16>(x, w, b, y, ws, mean, rstd, N, R, C, G, eps, silu, s); break;
+  }
+}
+
+template <typename T>
+static void gn_nhwc_bwd_t(const void* dy, const void* x, const float* w,
+                          const float* b, const float* mean, const float* rstd,
+                          float* ws, void* dx, float* dw, float* db, int N,
+                          int R, int C, int G, bool silu, hipStream_t s) {
+  int chunks = nhwc_row_chunks(N, R);
+  int rows_per_blk = (R + chunks - 1) / chunks;
+  dim3 grid(chunks, N), block(256);
+  size_t lds = 2 * (size_t)C * sizeof(float);
+  if (silu)
+    hipLaunchKernelGGL((dcr_nhwc::gn_nhwc_bwd_stats_kernel<T, true>), grid, block, lds, s,
+                       (const T*)dy, (const T*)x, w, b, mean, rstd, ws, dw, db,
+                       N, R, C, G, rows_per_blk);
+  else
+    hipLaunchKernelGGL((dcr_nhwc::gn_nhwc_bwd_stats_kernel<T, false>), grid, block, lds, s,
+                       (const T*)dy, (const T*)x, w, b, mean, rstd, ws, dw, db,
+                       N, R, C, G, rows_per_blk);
+  long total = (long)N * R * C;
+  float inv_L = 1.f / ((float)R * (C / G));
+  dim3 agrid((int)min((total / 4 + 255) / 256, (long)8192)), ablock(256);
+  if (silu)
+    hipLaunchKernelGGL((dcr_nhwc::gn_nhwc_bwd_apply_kernel<T, true>), agrid, ablock, 0, s,
+                       (const T*)dy, (const T*)x, w, b, mean, rstd, ws, (T*)dx,
+                       total, R, C, G, inv_L);
+  else
+    hipLaunchKernelGGL((dcr_nhwc::gn_nhwc_bwd_apply_kernel<T, false>), agrid, ablock, 0, s,
+                       (const T*)dy, (const T*)x, w, b, mean, rstd, ws, (T*)dx,
+                       total, R, C, G, inv_L);
+}
+
+void gn_nhwc_bwd_launch(DType dt, const void* dy, const void* x, const float* w,
+                        const float* b, const float* mean, const float* rstd,
+                        float* ws, void* dx, float* dw, float* db, int N,
+                        int R, int C, int G, bool silu, hipStream_t s) {
+  switch (dt) {
+    case DT_F32: gn_nhwc_bwd_t<float>(dy, x, w, b, mean, rstd, ws, dx, dw, db, N, R, C, G, silu, s); break;
+    case DT_F16: gn_nhwc_bwd_t<__half>(dy, x, w, b, mean, rstd, ws, dx, dw, db, N, R, C, G, silu, s); break;
+    case DT_BF16: gn_nhwc_bwd_t<__hip_bfloat16>(dy, x, w, b, mean, rstd, ws, dx, dw, db, N, R, C, G, silu, s); break;
+  }
+}
+
+}  // namespace dcr

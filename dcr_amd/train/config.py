@@ -54,7 +54,8 @@ class TrainConfig:
     adam_weight_decay: float = 1e-2
     adam_epsilon: float = 1e-8
     max_grad_norm: float = 1.0
-    mixed_precision: str = "bf16"           # no | fp16 | bf16
+    mixed_precision: str = "bf16"           # no | fp16 | bf16 | pure_bf16
+    channels_last: bool = False             # NHWC convs (MI355X igemm path)
     seed: Optional[int] = None
 
     # mitigations (train-time)

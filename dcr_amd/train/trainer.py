@@ -116,6 +116,9 @@ class Trainer:
 
         self.unet.to(self.device,
                      dtype=torch.bfloat16 if self.pure_bf16 else None)
+        if self.cfg.channels_last and self.device.type == "cuda":
+            self.unet.to(memory_format=torch.channels_last)
+            self.vae.to(memory_format=torch.channels_last)
         # frozen models run in the compute dtype (reference: diff_train.py:531-533)
         self.vae.to(self.device, dtype=self.weight_dtype)
         self.text_encoder.to(
@@ -200,6 +203,8 @@ class Trainer:
         autocast_on = self.weight_dtype != torch.float32 and not self.pure_bf16
 
         pixel_values = batch["pixel_values"].to(self.device, non_blocking=True)
+        if cfg.channels_last and self.device.type == "cuda":
+            pixel_values = pixel_values.to(memory_format=torch.channels_last)
         input_ids = batch["input_ids"].to(self.device, non_blocking=True)
 
         self.ddp.require_backward_grad_sync = sync_gradients

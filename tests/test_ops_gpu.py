@@ -270,3 +270,44 @@ def test_flash_attention_autograd_wrapper(ext):
     out.sum().backward()
     for g in (q.grad, k.grad, v.grad):
         assert g is not None and torch.isfinite(g.float()).all()
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_groupnorm_nhwc_fwd_bwd(ext, dtype):
+    """channels_last GroupNorm vs the fp32 reference."""
+    torch.manual_seed(5)
+    for (N, C, H, W, G) in [(2, 320, 32, 32, 32), (4, 1280, 8, 8, 32),
+                            (2, 128, 64, 64, 32), (1, 512, 9, 7, 32)]:
+        x = torch.randn(N, C, H, W, device="cuda", dtype=dtype) \
+            .to(memory_format=torch.channels_last)
+        w = (torch.randn(C, device="cuda") * 0.5 + 1).requires_grad_(True)
+        b = (torch.randn(C, device="cuda") * 0.1).requires_grad_(True)
+        dy = torch.randn(N, C, H, W, device="cuda", dtype=dtype) \
+            .to(memory_format=torch.channels_last)
+
+        y, mean, rstd = ext.groupnorm_silu_nhwc_fwd(x, w.detach(), b.detach(),
+                                                    G, 1e-5, True)
+        xr = x.float().detach().requires_grad_(True)
+        ref = F.silu(F.group_norm(xr, G, w, b, 1e-5))
+        _close(y, ref, 1e-4 if dtype == torch.float32 else 1e-2)
+
+        ref.backward(dy.float())
+        dx, dw, db = ext.groupnorm_silu_nhwc_bwd(dy, x, w.detach(), b.detach(),
+                                                 mean, rstd, G, True)
+        _close(dx, xr.grad, 2e-3 if dtype == torch.float32 else 4e-2)
+        rel_w = (dw.float() - w.grad).abs().max() / (w.grad.abs().max() + 1e-6)
+        assert rel_w.item() < (1e-3 if dtype == torch.float32 else 3e-2)
+
+
+def test_group_norm_dispatch_channels_last(ext):
+    """ops.group_norm_silu must route channels_last input through the
+    NHWC kernels and return a channels_last tensor."""
+    from dcr_amd import ops
+    x = torch.randn(2, 64, 16, 16, device="cuda", dtype=torch.bfloat16) \
+        .to(memory_format=torch.channels_last).requires_grad_(True)
+    w = torch.ones(64, device="cuda")
+    b = torch.zeros(64, device="cuda")
+    y = ops.group_norm_silu(x, w, b, 8, 1e-5, True)
+    assert y.is_contiguous(memory_format=torch.channels_last)
+    y.sum().backward()
+    assert torch.isfinite(x.grad.float()).all()
