@@ -12,6 +12,11 @@ import argparse
 import json
 import time
 
+import sys
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
+
 import torch
 
 
