@@ -256,6 +256,23 @@ def get_velocity(
     return (sa * noise.float() - sb * x0.float()).to(x0.dtype)
 
 
+def lincomb(x: torch.Tensor, y: torch.Tensor, a: float, b: float,
+            z: Optional[torch.Tensor] = None, c: float = 0.0) -> torch.Tensor:
+    """out = a*x + b*y (+ c*z) — one fused HIP kernel; the DDIM and
+    DPM-Solver++ sampler updates are expressed through this."""
+    from . import ext
+
+    m = ext()
+    if use_hip(x) and m is not None and x.numel() % 4 == 0:
+        return m.lincomb(x.contiguous(), y.contiguous(),
+                         z.contiguous() if z is not None else None,
+                         float(a), float(b), float(c))
+    out = a * x.float() + b * y.float()
+    if z is not None:
+        out = out + c * z.float()
+    return out.to(x.dtype)
+
+
 def cfg_combine(eps_uncond: torch.Tensor, eps_text: torch.Tensor, scale: float) -> torch.Tensor:
     """Classifier-free guidance: eps_u + s * (eps_t - eps_u)."""
     from . import ext

@@ -220,6 +220,22 @@ at::Tensor get_velocity(at::Tensor x0, at::Tensor noise, at::Tensor ac, at::Tens
   return sched_common(1, x0, noise, ac, t);
 }
 
+at::Tensor lincomb(at::Tensor X, at::Tensor Y,
+                   c10::optional<at::Tensor> Z, double a, double b, double c) {
+  TORCH_CHECK(X.is_cuda() && X.is_contiguous() && Y.is_contiguous());
+  TORCH_CHECK(X.numel() % 4 == 0);
+  auto out = at::empty_like(X);
+  const void* zp = nullptr;
+  at::Tensor Zc;
+  if (Z.has_value()) {
+    Zc = Z->contiguous();
+    zp = Zc.data_ptr();
+  }
+  lincomb_launch(dtype_of(X), X.data_ptr(), Y.data_ptr(), zp, out.data_ptr(),
+                 (float)a, (float)b, (float)c, X.numel(), cur_stream());
+  return out;
+}
+
 at::Tensor cfg_combine(at::Tensor eu, at::Tensor et, double scale) {
   TORCH_CHECK(eu.is_cuda() && eu.is_contiguous() && et.is_contiguous());
   TORCH_CHECK(eu.numel() % 4 == 0);
@@ -290,4 +306,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("add_noise", &add_noise);
   mod.def("get_velocity", &get_velocity);
   mod.def("cfg_combine", &cfg_combine);
+  mod.def("lincomb", &lincomb);
 }

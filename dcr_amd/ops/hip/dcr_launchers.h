@@ -49,6 +49,9 @@ void sched_launch(DType dt, int mode, const void* x0, const void* noise,
                   long total, hipStream_t s);
 void cfg_launch(DType dt, const void* eu, const void* et, void* out, float s_,
                 long total, hipStream_t s);
+void lincomb_launch(DType dt, const void* X, const void* Y, const void* Z,
+                    void* out, float a, float b, float c, long total,
+                    hipStream_t s);
 
 // attention.hip (bf16, head_dim 64)
 void attn_fwd_launch(const void* q, const void* k, const void* v, void* o,

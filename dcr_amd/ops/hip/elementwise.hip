@@ -169,6 +169,34 @@ __global__ void sched_kernel(const T* __restrict__ x0, const T* __restrict__ noi
   }
 }
 
+// ------------------------------------------------------- sampler lincomb
+// out = a*X + b*Y (+ c*Z). Every DDIM / DPM-Solver++ update is one such
+// fused elementwise op with host-side scalar coefficients — the whole
+// sampling step is ONE kernel (north star: scheduler math as CDNA4 HIP).
+template <typename T, bool HASZ>
+__global__ void lincomb_kernel(const T* __restrict__ X, const T* __restrict__ Y,
+                               const T* __restrict__ Z, T* __restrict__ out,
+                               float a, float b, float c, long total) {
+  const long nvec = total / 4;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
+       i += (long)gridDim.x * blockDim.x) {
+    long e = i * 4;
+    f32x4 xv = load4<T>(X + e);
+    f32x4 yv = load4<T>(Y + e);
+    f32x4 o;
+#pragma unroll
+    for (int k = 0; k < 4; ++k)
+      (&o.x)[k] = a * (&xv.x)[k] + b * (&yv.x)[k];
+    if (HASZ) {
+      f32x4 zv = load4<T>(Z + e);
+#pragma unroll
+      for (int k = 0; k < 4; ++k)
+        (&o.x)[k] += c * (&zv.x)[k];
+    }
+    store4<T>(out + e, o);
+  }
+}
+
 // ---------------------------------------------------------------- CFG
 template <typename T>
 __global__ void cfg_kernel(const T* __restrict__ eu, const T* __restrict__ et,
@@ -195,7 +223,11 @@ __global__ void cfg_kernel(const T* __restrict__ eu, const T* __restrict__ et,
       const long*, T*, long, long);                                               \
   template __global__ void sched_kernel<T, 1>(const T*, const T*, const float*,   \
       const long*, T*, long, long);                                               \
-  template __global__ void cfg_kernel<T>(const T*, const T*, T*, float, long);
+  template __global__ void cfg_kernel<T>(const T*, const T*, T*, float, long); \
+  template __global__ void lincomb_kernel<T, true>(const T*, const T*, const T*, \
+      T*, float, float, float, long);                                          \
+  template __global__ void lincomb_kernel<T, false>(const T*, const T*, const T*, \
+      T*, float, float, float, long);
 
 DCR_INST_EW(float)
 DCR_INST_EW(__hip_bfloat16)
@@ -273,6 +305,27 @@ void sched_launch(DType dt, int mode, const void* x0, const void* noise,
     case DT_BF16: { SCHED_CASE(__hip_bfloat16) break; }
   }
 #undef SCHED_CASE
+}
+
+void lincomb_launch(DType dt, const void* X, const void* Y, const void* Z,
+                    void* out, float a, float b, float c, long total,
+                    hipStream_t s) {
+  dim3 grid(ew_blocks(total / 4)), block(256);
+#define LC_CASE(T)                                                                \
+  if (Z)                                                                          \
+    hipLaunchKernelGGL((lincomb_kernel<T, true>), grid, block, 0, s,              \
+                       (const T*)X, (const T*)Y, (const T*)Z, (T*)out, a, b, c,   \
+                       total);                                                    \
+  else                                                                            \
+    hipLaunchKernelGGL((lincomb_kernel<T, false>), grid, block, 0, s,             \
+                       (const T*)X, (const T*)Y, (const T*)Z, (T*)out, a, b, c,   \
+                       total);
+  switch (dt) {
+    case DT_F32: { LC_CASE(float) break; }
+    case DT_F16: { LC_CASE(__half) break; }
+    case DT_BF16: { LC_CASE(__hip_bfloat16) break; }
+  }
+#undef LC_CASE
 }
 
 void cfg_launch(DType dt, const void* eu, const void* et, void* out, float s_,

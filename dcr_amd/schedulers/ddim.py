@@ -13,6 +13,7 @@ from typing import Optional
 
 import torch
 
+from .. import ops
 from .ddpm import make_betas
 
 
@@ -66,6 +67,24 @@ class DDIMScheduler:
         prev_t = t - self.config.num_train_timesteps // self.num_inference_steps
         ac_t = self.alphas_cumprod[t]
         ac_prev = self.alphas_cumprod[prev_t] if prev_t >= 0 else self.final_alpha_cumprod
+
+        # fast path (GPU): the eta=0 update is linear in (sample,
+        # model_output) -> one fused HIP kernel (ops.lincomb)
+        if eta == 0.0 and not self.config.clip_sample and sample.is_cuda:
+            a_t = float(ac_t.sqrt())
+            s_t = float((1 - ac_t).sqrt())
+            a_p = float(ac_prev.sqrt())
+            s_p = float((1 - ac_prev).sqrt())     # dir coefficient, sigma = 0
+            if self.prediction_type == "epsilon":
+                A = a_p / a_t
+                B = s_p - a_p * s_t / a_t
+            elif self.prediction_type == "v_prediction":
+                A = a_p * a_t + s_p * s_t
+                B = s_p * a_t - a_p * s_t
+            else:
+                raise ValueError(self.prediction_type)
+            return _StepOut(ops.lincomb(sample, model_output, A, B))
+
         ac_t = ac_t.to(sample.device)
         ac_prev = ac_prev.to(sample.device)
 

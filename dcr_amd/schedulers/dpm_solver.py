@@ -15,6 +15,7 @@ from typing import List, Optional
 
 import torch
 
+from .. import ops
 from .ddpm import make_betas
 
 
@@ -66,12 +67,16 @@ class DPMSolverMultistepScheduler:
         return sample
 
     def _to_x0(self, model_output: torch.Tensor, t: int, sample: torch.Tensor):
-        a, s = self.alpha_t[t].to(sample.device), self.sigma_t[t].to(sample.device)
+        a, s = float(self.alpha_t[t]), float(self.sigma_t[t])
         if self.prediction_type == "epsilon":
-            return (sample.float() - s * model_output.float()) / a
-        if self.prediction_type == "v_prediction":
-            return a * sample.float() - s * model_output.float()
-        raise ValueError(self.prediction_type)
+            cx, cm = 1.0 / a, -s / a
+        elif self.prediction_type == "v_prediction":
+            cx, cm = a, -s
+        else:
+            raise ValueError(self.prediction_type)
+        if sample.is_cuda:
+            return ops.lincomb(sample, model_output, cx, cm)
+        return cx * sample.float() + cm * model_output.float()
 
     def step(self, model_output: torch.Tensor, timestep: int, sample: torch.Tensor,
              generator=None) -> _StepOut:
@@ -89,16 +94,24 @@ class DPMSolverMultistepScheduler:
         s_prev = self.sigma_t[prev_t].to(sample.device)
         s_t = self.sigma_t[t].to(sample.device)
 
+        K = float(-a_prev * torch.expm1(-h))
+        c_x = float(s_prev / s_t)
         if self.lower_order_nums < 1 or self.model_outputs[-2] is None or i + 1 >= len(self.timesteps):
             # first-order (DDIM-like) update in x0-parameterization
-            prev = (s_prev / s_t) * sample.float() - a_prev * (torch.expm1(-h)) * x0
+            if sample.is_cuda:
+                prev = ops.lincomb(sample, x0, c_x, K)
+            else:
+                prev = c_x * sample.float() + K * x0
         else:
             t_prev2 = int(self.timesteps[i - 1])
-            h_last = (lam_t - self.lambda_t[t_prev2]).to(sample.device)
-            r = h_last / h
+            h_last = lam_t - self.lambda_t[t_prev2]
+            r = float(h_last / h)
             x0_prev = self.model_outputs[-2]
-            D = (1 + 1 / (2 * r)) * x0 - (1 / (2 * r)) * x0_prev
-            prev = (s_prev / s_t) * sample.float() - a_prev * (torch.expm1(-h)) * D
+            g = 1.0 / (2 * r)
+            if sample.is_cuda:
+                prev = ops.lincomb(sample, x0, c_x, K * (1 + g), x0_prev, -K * g)
+            else:
+                prev = c_x * sample.float() + K * ((1 + g) * x0 - g * x0_prev)
 
         self.lower_order_nums = min(self.lower_order_nums + 1, self.config.solver_order)
         self._step_index += 1
