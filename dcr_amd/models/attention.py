@@ -45,11 +45,13 @@ class Attention(nn.Module):
     def forward(self, x: torch.Tensor, context: Optional[torch.Tensor] = None) -> torch.Tensor:
         B, L, _ = x.shape
         ctx = context if context is not None else x
-        q = self.to_q(x).view(B, L, self.heads, self.dim_head).transpose(1, 2)
-        k = self.to_k(ctx).view(B, ctx.shape[1], self.heads, self.dim_head).transpose(1, 2)
-        v = self.to_v(ctx).view(B, ctx.shape[1], self.heads, self.dim_head).transpose(1, 2)
-        out = ops.attention(q, k, v)
-        out = out.transpose(1, 2).reshape(B, L, self.heads * self.dim_head)
+        # transpose-free [B, L, H, D] layout: the flash kernel reads it
+        # with a strided row pitch, so no permute copies are materialized
+        q = self.to_q(x).view(B, L, self.heads, self.dim_head)
+        k = self.to_k(ctx).view(B, ctx.shape[1], self.heads, self.dim_head)
+        v = self.to_v(ctx).view(B, ctx.shape[1], self.heads, self.dim_head)
+        out = ops.attention(q, k, v, layout="blhd")
+        out = out.reshape(B, L, self.heads * self.dim_head)
         out = self.to_out[0](out)
         return self.to_out[1](out)
 

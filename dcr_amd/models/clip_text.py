@@ -74,11 +74,11 @@ class CLIPAttention(nn.Module):
 
     def forward(self, x):
         B, L, C = x.shape
-        q = self.q_proj(x).view(B, L, self.heads, self.dim_head).transpose(1, 2)
-        k = self.k_proj(x).view(B, L, self.heads, self.dim_head).transpose(1, 2)
-        v = self.v_proj(x).view(B, L, self.heads, self.dim_head).transpose(1, 2)
-        out = ops.attention(q, k, v, causal=True)
-        return self.out_proj(out.transpose(1, 2).reshape(B, L, C))
+        q = self.q_proj(x).view(B, L, self.heads, self.dim_head)
+        k = self.k_proj(x).view(B, L, self.heads, self.dim_head)
+        v = self.v_proj(x).view(B, L, self.heads, self.dim_head)
+        out = ops.attention(q, k, v, causal=True, layout="blhd")
+        return self.out_proj(out.reshape(B, L, C))
 
 
 class CLIPMLP(nn.Module):

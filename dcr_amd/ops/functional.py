@@ -201,11 +201,18 @@ def attention(
     v: torch.Tensor,
     causal: bool = False,
     scale: Optional[float] = None,
+    layout: str = "bhld",
 ) -> torch.Tensor:
+    """layout "bhld": q,k,v are [B, H, L, D]. layout "blhd": [B, L, H, D] —
+    the transpose-free layout the models use (the flash kernel walks it
+    with a strided row pitch, so no head-split permute copies happen)."""
     if scale is None:
         scale = 1.0 / math.sqrt(q.shape[-1])
     if use_hip(q):
         if q.dtype == torch.bfloat16 and q.shape[-1] == 64 and k.shape[-1] == 64:
+            if layout == "blhd":
+                return _FlashAttention.apply(q.contiguous(), k.contiguous(),
+                                             v.contiguous(), scale, causal)
             shp = q.shape
             out = _FlashAttention.apply(
                 q.reshape(-1, shp[-2], 64).contiguous(),
@@ -213,7 +220,16 @@ def attention(
                 v.reshape(-1, v.shape[-2], 64).contiguous(),
                 scale, causal)
             return out.reshape(shp)
+        if layout == "blhd":
+            out = _attention_math(q.permute(0, 2, 1, 3), k.permute(0, 2, 1, 3),
+                                  v.permute(0, 2, 1, 3), scale, causal)
+            return out.permute(0, 2, 1, 3)
         return _attention_math(q, k, v, scale, causal)
+    if layout == "blhd":
+        out = F.scaled_dot_product_attention(
+            q.permute(0, 2, 1, 3), k.permute(0, 2, 1, 3), v.permute(0, 2, 1, 3),
+            is_causal=causal, scale=scale)
+        return out.permute(0, 2, 1, 3)
     return F.scaled_dot_product_attention(q, k, v, is_causal=causal, scale=scale)
 
 

@@ -42,16 +42,18 @@ __device__ __forceinline__ short f2bf_rne(float f) {
   return *reinterpret_cast<short*>(&h);
 }
 
-// cooperative tile load (256 threads): dst[row][dim], zero-padded rows
+// cooperative tile load (256 threads): dst[row][dim], zero-padded rows.
+// rs = element stride between consecutive sequence rows (64 for [BH,L,D]
+// packed, H*64 for the transpose-free [B,L,H,D] layout).
 __device__ __forceinline__ void load_tile(const bf16_t* __restrict__ src,
-                                          int valid_rows,
+                                          long rs, int valid_rows,
                                           short* __restrict__ dst) {
   const int t = threadIdx.x;
   const int row = t >> 2;
   const int col = (t & 3) * 16;
   uint4 a = make_uint4(0, 0, 0, 0), b = a;
   if (row < valid_rows) {
-    const uint4* p = reinterpret_cast<const uint4*>(src + (long)row * DHEAD + col);
+    const uint4* p = reinterpret_cast<const uint4*>(src + (long)row * rs + col);
     a = p[0];
     b = p[1];
   }
@@ -62,14 +64,14 @@ __device__ __forceinline__ void load_tile(const bf16_t* __restrict__ src,
 
 // transposed: dst[dim][row] = src[row][dim]
 __device__ __forceinline__ void load_tile_T(const bf16_t* __restrict__ src,
-                                            int valid_rows,
+                                            long rs, int valid_rows,
                                             short* __restrict__ dst) {
   const int t = threadIdx.x;
   const int row = t >> 2;
   const int col0 = (t & 3) * 16;
   short v[16];
   if (row < valid_rows) {
-    const uint4* p = reinterpret_cast<const uint4*>(src + (long)row * DHEAD + col0);
+    const uint4* p = reinterpret_cast<const uint4*>(src + (long)row * rs + col0);
     *reinterpret_cast<uint4*>(v) = p[0];
     *reinterpret_cast<uint4*>(v + 8) = p[1];
   } else {
@@ -102,21 +104,23 @@ __device__ __forceinline__ float qsum(float v) {
 __global__ __launch_bounds__(256)
 void attn_fwd_kernel(const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
                      const bf16_t* __restrict__ v, bf16_t* __restrict__ o,
-                     float* __restrict__ lse, int Lq, int Lk, float scale,
-                     int causal) {
+                     float* __restrict__ lse, int Lq, int Lk, int H,
+                     float scale, int causal) {
   __shared__ short sQ[TILE * PITCH];
   __shared__ short sK[TILE * PITCH];
   __shared__ short sVT[TILE * PITCH];   // [dim][key]
   __shared__ short sP[TILE * PITCH];    // [qrow][key]
 
   const int bh = blockIdx.y;
+  const int b = bh / H, h = bh % H;
   const int q0 = blockIdx.x * TILE;
-  const bf16_t* qp = q + ((long)bh * Lq + q0) * DHEAD;
-  const bf16_t* kp = k + (long)bh * Lk * DHEAD;
-  const bf16_t* vp = v + (long)bh * Lk * DHEAD;
-  bf16_t* op = o + (long)bh * Lq * DHEAD;
+  const long qrs = (long)H * DHEAD;       // row stride in the [B,L,H,D] walk
+  const bf16_t* qp = q + (((long)b * Lq + q0) * H + h) * DHEAD;
+  const bf16_t* kp = k + ((long)b * Lk * H + h) * DHEAD;
+  const bf16_t* vp = v + ((long)b * Lk * H + h) * DHEAD;
+  bf16_t* op = o + ((long)b * Lq * H + h) * DHEAD;
 
-  load_tile(qp, Lq - q0, sQ);
+  load_tile(qp, qrs, Lq - q0, sQ);
 
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
@@ -134,8 +138,8 @@ void attn_fwd_kernel(const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
 
   for (int kv0 = 0; kv0 < kv_end; kv0 += TILE) {
     __syncthreads();
-    load_tile(kp + (long)kv0 * DHEAD, Lk - kv0, sK);
-    load_tile_T(vp + (long)kv0 * DHEAD, Lk - kv0, sVT);
+    load_tile(kp + (long)kv0 * qrs, qrs, Lk - kv0, sK);
+    load_tile_T(vp + (long)kv0 * qrs, qrs, Lk - kv0, sVT);
     __syncthreads();
 
     // S = scale * Q K^T  (wave: 16 q-rows x 64 keys)
@@ -213,7 +217,7 @@ void attn_fwd_kernel(const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
     const float inv = (row_sum[r] > 0.f) ? 1.f / row_sum[r] : 0.f;
 #pragma unroll
     for (int ds_ = 0; ds_ < 4; ++ds_)
-      op[(long)qrow * DHEAD + ds_ * 16 + l16] = __float2bfloat16(acc_o[ds_][r] * inv);
+      op[(long)qrow * qrs + ds_ * 16 + l16] = __float2bfloat16(acc_o[ds_][r] * inv);
     if (l16 == 0 && lse != nullptr)
       lse[(long)bh * Lq + qrow] = row_max[r] + __logf(fmaxf(row_sum[r], 1e-30f));
   }
@@ -224,11 +228,15 @@ void attn_fwd_kernel(const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
 // ==========================================================================
 __global__ void attn_bwd_delta_kernel(const bf16_t* __restrict__ dO,
                                       const bf16_t* __restrict__ O,
-                                      float* __restrict__ delta, long total_rows) {
-  long row = (long)blockIdx.x * blockDim.x + threadIdx.x;
+                                      float* __restrict__ delta, long total_rows,
+                                      int Lq, int H) {
+  long row = (long)blockIdx.x * blockDim.x + threadIdx.x;  // (b*H + h)*Lq + l
   if (row >= total_rows) return;
-  const bf16_t* a = dO + row * DHEAD;
-  const bf16_t* b = O + row * DHEAD;
+  const long bh = row / Lq, l = row % Lq;
+  const long b_ = bh / H, h = bh % H;
+  const long off = ((b_ * Lq + l) * H + h) * DHEAD;
+  const bf16_t* a = dO + off;
+  const bf16_t* b = O + off;
   float s = 0.f;
 #pragma unroll
   for (int i = 0; i < DHEAD; i += 4) {
@@ -247,7 +255,7 @@ void attn_bwd_dkdv_kernel(const bf16_t* __restrict__ q, const bf16_t* __restrict
                           const bf16_t* __restrict__ v, const bf16_t* __restrict__ dO,
                           const float* __restrict__ lse, const float* __restrict__ delta,
                           bf16_t* __restrict__ dK, bf16_t* __restrict__ dV,
-                          int Lq, int Lk, float scale, int causal) {
+                          int Lq, int Lk, int H, float scale, int causal) {
   __shared__ short sK[TILE * PITCH];     // [key][dim]
   __shared__ short sV[TILE * PITCH];     // [key][dim]
   __shared__ short sQ[TILE * PITCH];     // [qrow][dim]
@@ -259,16 +267,18 @@ void attn_bwd_dkdv_kernel(const bf16_t* __restrict__ q, const bf16_t* __restrict
   __shared__ float sDelta[TILE];
 
   const int bh = blockIdx.y;
+  const int b_ = bh / H, h = bh % H;
   const int k0 = blockIdx.x * TILE;
-  const bf16_t* qp = q + (long)bh * Lq * DHEAD;
-  const bf16_t* kp = k + ((long)bh * Lk + k0) * DHEAD;
-  const bf16_t* vp = v + ((long)bh * Lk + k0) * DHEAD;
-  const bf16_t* dop = dO + (long)bh * Lq * DHEAD;
+  const long rs = (long)H * DHEAD;
+  const bf16_t* qp = q + ((long)b_ * Lq * H + h) * DHEAD;
+  const bf16_t* kp = k + (((long)b_ * Lk + k0) * H + h) * DHEAD;
+  const bf16_t* vp = v + (((long)b_ * Lk + k0) * H + h) * DHEAD;
+  const bf16_t* dop = dO + ((long)b_ * Lq * H + h) * DHEAD;
   const float* lsep = lse + (long)bh * Lq;
   const float* delp = delta + (long)bh * Lq;
 
-  load_tile(kp, Lk - k0, sK);
-  load_tile(vp, Lk - k0, sV);
+  load_tile(kp, rs, Lk - k0, sK);
+  load_tile(vp, rs, Lk - k0, sV);
 
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
@@ -287,10 +297,10 @@ void attn_bwd_dkdv_kernel(const bf16_t* __restrict__ q, const bf16_t* __restrict
 
   for (int q0 = q_start; q0 < Lq; q0 += TILE) {
     __syncthreads();
-    load_tile(qp + (long)q0 * DHEAD, Lq - q0, sQ);
-    load_tile_T(qp + (long)q0 * DHEAD, Lq - q0, sQT);
-    load_tile(dop + (long)q0 * DHEAD, Lq - q0, sdO);
-    load_tile_T(dop + (long)q0 * DHEAD, Lq - q0, sdOT);
+    load_tile(qp + (long)q0 * rs, rs, Lq - q0, sQ);
+    load_tile_T(qp + (long)q0 * rs, rs, Lq - q0, sQT);
+    load_tile(dop + (long)q0 * rs, rs, Lq - q0, sdO);
+    load_tile_T(dop + (long)q0 * rs, rs, Lq - q0, sdOT);
     if (threadIdx.x < TILE) {
       const int qr = q0 + threadIdx.x;
       sLse[threadIdx.x] = (qr < Lq) ? lsep[qr] : 1e30f;
@@ -366,16 +376,16 @@ void attn_bwd_dkdv_kernel(const bf16_t* __restrict__ q, const bf16_t* __restrict
   }
 
   // store dK/dV rows (key-guarded)
-  bf16_t* dkp = dK + ((long)bh * Lk + k0) * DHEAD;
-  bf16_t* dvp = dV + ((long)bh * Lk + k0) * DHEAD;
+  bf16_t* dkp = dK + (((long)b_ * Lk + k0) * H + h) * DHEAD;
+  bf16_t* dvp = dV + (((long)b_ * Lk + k0) * H + h) * DHEAD;
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
     const int key = wkey0 + kgrp * 4 + r;
     if (k0 + key >= Lk) continue;
 #pragma unroll
     for (int ds_ = 0; ds_ < 4; ++ds_) {
-      dkp[(long)key * DHEAD + ds_ * 16 + l16] = __float2bfloat16(acc_dk[ds_][r]);
-      dvp[(long)key * DHEAD + ds_ * 16 + l16] = __float2bfloat16(acc_dv[ds_][r]);
+      dkp[(long)key * rs + ds_ * 16 + l16] = __float2bfloat16(acc_dk[ds_][r]);
+      dvp[(long)key * rs + ds_ * 16 + l16] = __float2bfloat16(acc_dv[ds_][r]);
     }
   }
 }
@@ -387,8 +397,8 @@ __global__ __launch_bounds__(256)
 void attn_bwd_dq_kernel(const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
                         const bf16_t* __restrict__ v, const bf16_t* __restrict__ dO,
                         const float* __restrict__ lse, const float* __restrict__ delta,
-                        bf16_t* __restrict__ dQ, int Lq, int Lk, float scale,
-                        int causal) {
+                        bf16_t* __restrict__ dQ, int Lq, int Lk, int H,
+                        float scale, int causal) {
   __shared__ short sQ[TILE * PITCH];
   __shared__ short sdO[TILE * PITCH];
   __shared__ short sK[TILE * PITCH];     // [key][dim]
@@ -397,14 +407,16 @@ void attn_bwd_dq_kernel(const bf16_t* __restrict__ q, const bf16_t* __restrict__
   __shared__ short sDS[TILE * PITCH];    // [qrow][key]
 
   const int bh = blockIdx.y;
+  const int b_ = bh / H, h = bh % H;
   const int q0 = blockIdx.x * TILE;
-  const bf16_t* qp = q + ((long)bh * Lq + q0) * DHEAD;
-  const bf16_t* kp = k + (long)bh * Lk * DHEAD;
-  const bf16_t* vp = v + (long)bh * Lk * DHEAD;
-  const bf16_t* dop = dO + ((long)bh * Lq + q0) * DHEAD;
+  const long rs = (long)H * DHEAD;
+  const bf16_t* qp = q + (((long)b_ * Lq + q0) * H + h) * DHEAD;
+  const bf16_t* kp = k + ((long)b_ * Lk * H + h) * DHEAD;
+  const bf16_t* vp = v + ((long)b_ * Lk * H + h) * DHEAD;
+  const bf16_t* dop = dO + (((long)b_ * Lq + q0) * H + h) * DHEAD;
 
-  load_tile(qp, Lq - q0, sQ);
-  load_tile(dop, Lq - q0, sdO);
+  load_tile(qp, rs, Lq - q0, sQ);
+  load_tile(dop, rs, Lq - q0, sdO);
 
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
@@ -429,9 +441,9 @@ void attn_bwd_dq_kernel(const bf16_t* __restrict__ q, const bf16_t* __restrict__
 
   for (int kv0 = 0; kv0 < kv_end; kv0 += TILE) {
     __syncthreads();
-    load_tile(kp + (long)kv0 * DHEAD, Lk - kv0, sK);
-    load_tile_T(kp + (long)kv0 * DHEAD, Lk - kv0, sKT);
-    load_tile(vp + (long)kv0 * DHEAD, Lk - kv0, sV);
+    load_tile(kp + (long)kv0 * rs, rs, Lk - kv0, sK);
+    load_tile_T(kp + (long)kv0 * rs, rs, Lk - kv0, sKT);
+    load_tile(vp + (long)kv0 * rs, rs, Lk - kv0, sV);
     __syncthreads();
 
     bf16x8 qf0 = frag(sQ, wrow0 + l16, kgrp * 8);
@@ -477,14 +489,14 @@ void attn_bwd_dq_kernel(const bf16_t* __restrict__ q, const bf16_t* __restrict__
     }
   }
 
-  bf16_t* dqp = dQ + ((long)bh * Lq + q0) * DHEAD;
+  bf16_t* dqp = dQ + (((long)b_ * Lq + q0) * H + h) * DHEAD;
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
     const int row = wrow0 + kgrp * 4 + r;
     if (q0 + row >= Lq) continue;
 #pragma unroll
     for (int ds_ = 0; ds_ < 4; ++ds_)
-      dqp[(long)row * DHEAD + ds_ * 16 + l16] = __float2bfloat16(acc_dq[ds_][r]);
+      dqp[(long)row * rs + ds_ * 16 + l16] = __float2bfloat16(acc_dq[ds_][r]);
   }
 }
 
@@ -518,35 +530,36 @@ __global__ void mfma_probe_kernel(const bf16_t* __restrict__ A,
 namespace dcr {
 
 void attn_fwd_launch(const void* q, const void* k, const void* v, void* o,
-                     float* lse, int BH, int Lq, int Lk, float scale,
+                     float* lse, int BH, int Lq, int Lk, int H, float scale,
                      bool causal, hipStream_t s) {
   dim3 grid((Lq + TILE - 1) / TILE, BH), block(256);
   hipLaunchKernelGGL(dcr_attn::attn_fwd_kernel, grid, block, 0, s,
                      (const dcr_attn::bf16_t*)q, (const dcr_attn::bf16_t*)k,
                      (const dcr_attn::bf16_t*)v, (dcr_attn::bf16_t*)o, lse,
-                     Lq, Lk, scale, causal ? 1 : 0);
+                     Lq, Lk, H, scale, causal ? 1 : 0);
 }
 
 void attn_bwd_launch(const void* q, const void* k, const void* v,
                      const void* o, const void* dO, const float* lse,
                      float* delta, void* dQ, void* dK, void* dV, int BH,
-                     int Lq, int Lk, float scale, bool causal, hipStream_t s) {
+                     int Lq, int Lk, int H, float scale, bool causal,
+                     hipStream_t s) {
   long rows = (long)BH * Lq;
   dim3 gd((rows + 255) / 256), bd(256);
   hipLaunchKernelGGL(dcr_attn::attn_bwd_delta_kernel, gd, bd, 0, s,
                      (const dcr_attn::bf16_t*)dO, (const dcr_attn::bf16_t*)o,
-                     delta, rows);
+                     delta, rows, Lq, H);
   dim3 g1((Lk + TILE - 1) / TILE, BH), b1(256);
   hipLaunchKernelGGL(dcr_attn::attn_bwd_dkdv_kernel, g1, b1, 0, s,
                      (const dcr_attn::bf16_t*)q, (const dcr_attn::bf16_t*)k,
                      (const dcr_attn::bf16_t*)v, (const dcr_attn::bf16_t*)dO,
                      lse, delta, (dcr_attn::bf16_t*)dK, (dcr_attn::bf16_t*)dV,
-                     Lq, Lk, scale, causal ? 1 : 0);
+                     Lq, Lk, H, scale, causal ? 1 : 0);
   dim3 g2((Lq + TILE - 1) / TILE, BH), b2(256);
   hipLaunchKernelGGL(dcr_attn::attn_bwd_dq_kernel, g2, b2, 0, s,
                      (const dcr_attn::bf16_t*)q, (const dcr_attn::bf16_t*)k,
                      (const dcr_attn::bf16_t*)v, (const dcr_attn::bf16_t*)dO,
-                     lse, delta, (dcr_attn::bf16_t*)dQ, Lq, Lk, scale,
+                     lse, delta, (dcr_attn::bf16_t*)dQ, Lq, Lk, H, scale,
                      causal ? 1 : 0);
 }
 

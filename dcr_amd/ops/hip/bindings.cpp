@@ -246,36 +246,49 @@ at::Tensor cfg_combine(at::Tensor eu, at::Tensor et, double scale) {
 }
 
 // ---------------------------------------------------------------- attention
+// q/k/v: [B, L, H, 64] (transpose-free BLHD layout; pass 3D [BH, L, 64]
+// for the packed per-head layout — treated as B'=BH, H=1).
+static void attn_dims(const at::Tensor& q, const at::Tensor& k,
+                      int64_t& B, int64_t& H, int64_t& Lq, int64_t& Lk) {
+  TORCH_CHECK(q.size(-1) == 64 && k.size(-1) == 64, "attn: head_dim 64 only");
+  if (q.dim() == 4) {
+    B = q.size(0); Lq = q.size(1); H = q.size(2); Lk = k.size(1);
+    TORCH_CHECK(k.dim() == 4 && k.size(2) == H, "attn: head mismatch");
+  } else {
+    TORCH_CHECK(q.dim() == 3);
+    B = q.size(0); Lq = q.size(1); H = 1; Lk = k.size(1);
+  }
+}
+
 std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
                                  double scale, bool causal) {
   TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16,
               "attn: bf16 CUDA only");
   TORCH_CHECK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
-  TORCH_CHECK(q.size(-1) == 64 && k.size(-1) == 64, "attn: head_dim 64 only");
-  const int64_t Lq = q.size(-2), Lk = k.size(-2);
-  const int64_t BH = q.numel() / (Lq * 64);
+  int64_t B, H, Lq, Lk;
+  attn_dims(q, k, B, H, Lq, Lk);
   auto o = at::empty_like(q);
-  auto lse = at::empty({BH, Lq}, q.options().dtype(at::kFloat));
+  auto lse = at::empty({B * H, Lq}, q.options().dtype(at::kFloat));
   attn_fwd_launch(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
-                  lse.data_ptr<float>(), (int)BH, (int)Lq, (int)Lk,
-                  (float)scale, causal, cur_stream());
+                  lse.data_ptr<float>(), (int)(B * H), (int)Lq, (int)Lk,
+                  (int)H, (float)scale, causal, cur_stream());
   return {o, lse};
 }
 
 std::vector<at::Tensor> attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
                                  at::Tensor o, at::Tensor dO, at::Tensor lse,
                                  double scale, bool causal) {
-  const int64_t Lq = q.size(-2), Lk = k.size(-2);
-  const int64_t BH = q.numel() / (Lq * 64);
+  int64_t B, H, Lq, Lk;
+  attn_dims(q, k, B, H, Lq, Lk);
   auto dQ = at::empty_like(q);
   auto dK = at::empty_like(k);
   auto dV = at::empty_like(v);
-  auto delta = at::empty({BH, Lq}, q.options().dtype(at::kFloat));
+  auto delta = at::empty({B * H, Lq}, q.options().dtype(at::kFloat));
   attn_bwd_launch(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
                   dO.contiguous().data_ptr(), lse.data_ptr<float>(),
                   delta.data_ptr<float>(), dQ.data_ptr(), dK.data_ptr(),
-                  dV.data_ptr(), (int)BH, (int)Lq, (int)Lk, (float)scale,
-                  causal, cur_stream());
+                  dV.data_ptr(), (int)(B * H), (int)Lq, (int)Lk, (int)H,
+                  (float)scale, causal, cur_stream());
   return {dQ, dK, dV};
 }
 

@@ -55,12 +55,13 @@ void lincomb_launch(DType dt, const void* X, const void* Y, const void* Z,
 
 // attention.hip (bf16, head_dim 64)
 void attn_fwd_launch(const void* q, const void* k, const void* v, void* o,
-                     float* lse, int BH, int Lq, int Lk, float scale,
+                     float* lse, int BH, int Lq, int Lk, int H, float scale,
                      bool causal, hipStream_t s);
 void attn_bwd_launch(const void* q, const void* k, const void* v,
                      const void* o, const void* dO, const float* lse,
                      float* delta, void* dQ, void* dK, void* dV, int BH,
-                     int Lq, int Lk, float scale, bool causal, hipStream_t s);
+                     int Lq, int Lk, int H, float scale, bool causal,
+                     hipStream_t s);
 void mfma_probe_launch(const void* A, const void* B, float* C, hipStream_t s);
 
 }  // namespace dcr

@@ -31,15 +31,19 @@ __global__ void gn_nhwc_stats_kernel(const T* __restrict__ x, float* __restrict_
   const long base = (long)n * R * C;
   const int Cg = C / G;
 
-  for (int c = threadIdx.x; c < C; c += blockDim.x) {
-    float a = 0.f, b = 0.f;
+  // vec4 over channels: thread t owns channels [4t, 4t+4) per stripe
+  for (int c4 = threadIdx.x * 4; c4 < C; c4 += blockDim.x * 4) {
+    f32x4 a = {0.f, 0.f, 0.f, 0.f}, b = {0.f, 0.f, 0.f, 0.f};
     for (int r = r0; r < r1; ++r) {
-      float v = to_f32<T>(x[base + (long)r * C + c]);
-      a += v;
-      b += v * v;
+      f32x4 v = load4<T>(x + base + (long)r * C + c4);
+#pragma unroll
+      for (int k = 0; k < 4; ++k) {
+        (&a.x)[k] += (&v.x)[k];
+        (&b.x)[k] += (&v.x)[k] * (&v.x)[k];
+      }
     }
-    s1[c] = a;
-    s2[c] = b;
+#pragma unroll
+    for (int k = 0; k < 4; ++k) { s1[c4 + k] = (&a.x)[k]; s2[c4 + k] = (&b.x)[k]; }
   }
   __syncthreads();
   for (int g = threadIdx.x; g < G; g += blockDim.x) {
@@ -111,27 +115,41 @@ __global__ void gn_nhwc_bwd_stats_kernel(const T* __restrict__ dy, const T* __re
   const long base = (long)n * R * C;
   const int Cg = C / G;
 
-  for (int c = threadIdx.x; c < C; c += blockDim.x) {
-    const int g = c / Cg;
-    const float m = mean[(long)n * G + g];
-    const float rs = rstd[(long)n * G + g];
-    const float wc = w[c], bc = b_[c];
-    float a = 0.f, bb = 0.f, dwc = 0.f, dbc = 0.f;
-    for (int r = r0; r < r1; ++r) {
-      long idx = base + (long)r * C + c;
-      float yh = (to_f32<T>(x[idx]) - m) * rs;
-      float dz = to_f32<T>(dy[idx]);
-      if (SILU) dz *= dsilu(yh * wc + bc);
-      float gx = dz * wc;
-      a += gx;
-      bb += gx * yh;
-      dwc += dz * yh;
-      dbc += dz;
+  for (int c4 = threadIdx.x * 4; c4 < C; c4 += blockDim.x * 4) {
+    float m[4], rs[4], wc[4], bc[4];
+    float a[4] = {0, 0, 0, 0}, bb[4] = {0, 0, 0, 0};
+    float dwc[4] = {0, 0, 0, 0}, dbc[4] = {0, 0, 0, 0};
+#pragma unroll
+    for (int k = 0; k < 4; ++k) {
+      const int g = (c4 + k) / Cg;
+      m[k] = mean[(long)n * G + g];
+      rs[k] = rstd[(long)n * G + g];
+      wc[k] = w[c4 + k];
+      bc[k] = b_[c4 + k];
     }
-    sa[c] = a;
-    sb[c] = bb;
-    atomicAdd(&dw[c], dwc);
-    atomicAdd(&db[c], dbc);
+    for (int r = r0; r < r1; ++r) {
+      long idx = base + (long)r * C + c4;
+      f32x4 xv = load4<T>(x + idx);
+      f32x4 gv = load4<T>(dy + idx);
+#pragma unroll
+      for (int k = 0; k < 4; ++k) {
+        float yh = ((&xv.x)[k] - m[k]) * rs[k];
+        float dz = (&gv.x)[k];
+        if (SILU) dz *= dsilu(yh * wc[k] + bc[k]);
+        float gx = dz * wc[k];
+        a[k] += gx;
+        bb[k] += gx * yh;
+        dwc[k] += dz * yh;
+        dbc[k] += dz;
+      }
+    }
+#pragma unroll
+    for (int k = 0; k < 4; ++k) {
+      sa[c4 + k] = a[k];
+      sb[c4 + k] = bb[k];
+      atomicAdd(&dw[c4 + k], dwc[k]);
+      atomicAdd(&db[c4 + k], dbc[k]);
+    }
   }
   __syncthreads();
   for (int g = threadIdx.x; g < G; g += blockDim.x) {

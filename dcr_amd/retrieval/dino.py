@@ -43,11 +43,12 @@ class ViTAttention(nn.Module):
 
     def forward(self, x):
         B, N, C = x.shape
-        qkv = self.qkv(x).reshape(B, N, 3, self.num_heads, self.head_dim) \
-            .permute(2, 0, 3, 1, 4)
-        q, k, v = qkv[0], qkv[1], qkv[2]
-        out = ops.attention(q, k, v)
-        return self.proj(out.transpose(1, 2).reshape(B, N, C))
+        qkv = self.qkv(x).reshape(B, N, 3, self.num_heads, self.head_dim)
+        q = qkv[:, :, 0]
+        k = qkv[:, :, 1]
+        v = qkv[:, :, 2]
+        out = ops.attention(q, k, v, layout="blhd")
+        return self.proj(out.reshape(B, N, C))
 
 
 class ViTMlp(nn.Module):
