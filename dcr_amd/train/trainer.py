@@ -74,6 +74,10 @@ class Trainer:
             "pure_bf16": torch.bfloat16,
         }[cfg.mixed_precision]
         self.pure_bf16 = cfg.mixed_precision == "pure_bf16"
+        # fp16 needs loss scaling (reference parity: utils_ret.py:834-860);
+        # bf16 does not.
+        self.scaler = torch.amp.GradScaler("cuda") \
+            if cfg.mixed_precision == "fp16" and torch.cuda.is_available() else None
 
         self._build_models()
         self._build_data()
@@ -246,11 +250,18 @@ class Trainer:
 
         loss = F.mse_loss(model_pred.float(), target.float(), reduction="mean")
         with self.prof.phase("backward"):
-            loss.backward()
+            if self.scaler is not None:
+                self.scaler.scale(loss).backward()
+            else:
+                loss.backward()
 
         if sync_gradients:
             with self.prof.phase("optimizer"):
                 self.ddp.finalize()
+                if self.scaler is not None:
+                    inv = 1.0 / self.scaler.get_scale()
+                    self.optimizer.flat_grad.mul_(inv)
+                    self.scaler.update()
                 self.optimizer.clip_grad_norm_(cfg.max_grad_norm)
                 self.optimizer.step(lr=get_lr(cfg, self.global_step))
                 self.optimizer.zero_grad()

@@ -132,3 +132,26 @@ def test_synthetic_dataset_deterministic():
     ds = SyntheticImageDataset(4, size=32, tokenizer=tok, seed=5)
     a, b = ds[2], ds[2]
     assert torch.equal(a["instance_images"], b["instance_images"])
+
+
+def test_cutmix_mixup():
+    from dcr_amd.data import cutmix_data, mixup_data, rand_bbox
+    torch.manual_seed(0)
+    np.random.seed(0)
+    x = torch.rand(4, 3, 32, 32)
+    y = torch.arange(4)
+    mx, ya, yb, lam = cutmix_data(x, y, alpha=1.0)
+    assert mx.shape == x.shape and 0.0 <= lam <= 1.0
+    mx2, _, _, lam2 = mixup_data(x, y, alpha=1.0)
+    assert torch.isfinite(mx2).all() and 0.0 <= lam2 <= 1.0
+    b = rand_bbox(x.shape, 0.5)
+    assert b[0] <= b[2] and b[1] <= b[3]
+
+
+def test_bool_flag():
+    import pytest as _pytest
+    from dcr_amd.utils import bool_flag
+    assert bool_flag("true") and bool_flag("on") and bool_flag("1")
+    assert not bool_flag("false") and not bool_flag("off")
+    with _pytest.raises(Exception):
+        bool_flag("maybe")
