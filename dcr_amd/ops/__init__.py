@@ -92,5 +92,6 @@ from .functional import (  # noqa: E402,F401
     add_noise,
     get_velocity,
     cfg_combine,
+    lincomb,
 )
 from .adamw import FusedAdamW  # noqa: E402,F401

@@ -61,15 +61,16 @@ def topk_stats(sim_gen_train: torch.Tensor, sim_train_train: Optional[torch.Tens
         "sim_gt_05pc": (top1 > threshold).float().mean().item(),
     }
     if sim_train_train is not None:
-        # top-2 because top-1 is the self-match (diff_retrieval.py:419)
+        # top-2 because top-1 is the self-match (diff_retrieval.py:419);
+        # 'bg_*' = train-train background, reference wandb key names
+        # (diff_retrieval.py:456-468)
         t2 = sim_train_train.topk(2, dim=1).values[:, 1].float()
         out.update({
-            "trainsim_mean": t2.mean().item(),
-            "trainsim_std": t2.std().item(),
-            "trainsim_75pc": t2.quantile(0.75).item(),
-            "trainsim_90pc": t2.quantile(0.90).item(),
-            "trainsim_95pc": t2.quantile(0.95).item(),
-            "trainsim_gt_05pc": (t2 > threshold).float().mean().item(),
+            "bg_mean": t2.mean().item(),
+            "bg_std": t2.std().item(),
+            "bg_75pc": t2.quantile(0.75).item(),
+            "bg_90pc": t2.quantile(0.90).item(),
+            "bg_95pc": t2.quantile(0.95).item(),
         })
     return out
 

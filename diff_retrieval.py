@@ -156,8 +156,13 @@ def main():
         imgs = torch.stack([query_ds[i][0] for i in range(len(query_ds))])
         cs = gen_clipscore(clip_model, clip_tok, imgs, query_ds.prompts,
                            device=device)
-        tracker.log({"clipscore_mean": cs.mean().item(),
-                     "clipscore_std": cs.std().item()})
+        rec = {"clipscore": cs.mean().item()}
+        if val_ds.prompts:
+            vimgs = torch.stack([val_ds[i][0] for i in range(len(val_ds))])
+            csb = gen_clipscore(clip_model, clip_tok, vimgs, val_ds.prompts,
+                                device=device)
+            rec["clipscore_bg"] = csb.mean().item()
+        tracker.log(rec)
 
     # complexity correlations (reference :498-540)
     if not args.noeval:

@@ -39,7 +39,7 @@ def test_baseline_config1_cpu_retrieval():
     assert sim.shape == (64, 64)
     stats = topk_stats(sim, sim_matrix(feats[64:], feats[64:]))
     for k in ("sim_mean", "sim_std", "sim_95pc", "sim_gt_05pc",
-              "trainsim_mean", "trainsim_gt_05pc"):
+              "bg_mean", "bg_95pc"):
         assert k in stats and np.isfinite(stats[k])
     # identical sets => self-similarity 1.0 on the diagonal
     sim_self = sim_matrix(feats, feats)
