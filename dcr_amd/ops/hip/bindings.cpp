@@ -30,6 +30,21 @@ at::Tensor as_f32(const at::Tensor& t) {
                                        : t.to(at::kFloat).contiguous();
 }
 
+// weights may stay in the activation dtype (pure-bf16 training: avoids a
+// cast kernel per norm call) or be fp32 (autocast); anything else -> fp32.
+at::Tensor norm_weight(const at::Tensor& w, const at::Tensor& x, bool& w_f32) {
+  if (w.scalar_type() == at::kFloat) {
+    w_f32 = true;
+    return w.contiguous();
+  }
+  if (w.scalar_type() == x.scalar_type()) {
+    w_f32 = false;
+    return w.contiguous();
+  }
+  w_f32 = true;
+  return w.to(at::kFloat).contiguous();
+}
+
 }  // namespace
 
 // ---------------------------------------------------------------- GroupNorm
@@ -40,11 +55,15 @@ std::vector<at::Tensor> groupnorm_silu_fwd(at::Tensor x, at::Tensor w, at::Tenso
   const int64_t N = x.size(0), C = x.size(1);
   const int64_t HW = x.numel() / (N * C);
   TORCH_CHECK(C % groups == 0, "gn: C % groups != 0");
-  auto wf = as_f32(w), bf = as_f32(b);
+  bool wf32;
+  auto wf = norm_weight(w, x, wf32);
+  bool bf32;
+  auto bf = norm_weight(b, x, bf32);
+  TORCH_CHECK(wf32 == bf32);
   auto y = at::empty_like(x);
   auto mean = at::empty({N * groups}, x.options().dtype(at::kFloat));
   auto rstd = at::empty_like(mean);
-  gn_fwd_launch(dtype_of(x), x.data_ptr(), wf.data_ptr<float>(), bf.data_ptr<float>(),
+  gn_fwd_launch(dtype_of(x), x.data_ptr(), wf.data_ptr(), bf.data_ptr(), wf32,
                 y.data_ptr(), mean.data_ptr<float>(), rstd.data_ptr<float>(),
                 (int)(N * groups), (int)groups, (int)(C / groups), (int)HW,
                 (float)eps, silu, cur_stream());
@@ -57,12 +76,16 @@ std::vector<at::Tensor> groupnorm_silu_bwd(at::Tensor dy, at::Tensor x, at::Tens
   TORCH_CHECK(dy.is_contiguous() && x.is_contiguous());
   const int64_t N = x.size(0), C = x.size(1);
   const int64_t HW = x.numel() / (N * C);
-  auto wf = as_f32(w), bf = as_f32(b);
+  bool wf32;
+  auto wf = norm_weight(w, x, wf32);
+  bool bf32;
+  auto bf = norm_weight(b, x, bf32);
   auto dx = at::empty_like(x);
-  auto dw = at::zeros({C}, x.options().dtype(at::kFloat));
-  auto db = at::zeros({C}, x.options().dtype(at::kFloat));
-  gn_bwd_launch(dtype_of(x), dy.data_ptr(), x.data_ptr(), wf.data_ptr<float>(),
-                bf.data_ptr<float>(), mean.data_ptr<float>(), rstd.data_ptr<float>(),
+  auto dwdb = at::zeros({2 * C}, x.options().dtype(at::kFloat));
+  auto dw = dwdb.narrow(0, 0, C);
+  auto db = dwdb.narrow(0, C, C);
+  gn_bwd_launch(dtype_of(x), dy.data_ptr(), x.data_ptr(), wf.data_ptr(),
+                bf.data_ptr(), wf32, mean.data_ptr<float>(), rstd.data_ptr<float>(),
                 dx.data_ptr(), dw.data_ptr<float>(), db.data_ptr<float>(),
                 (int)(N * groups), (int)groups, (int)(C / groups), (int)HW,
                 silu, cur_stream());
@@ -80,13 +103,17 @@ std::vector<at::Tensor> groupnorm_silu_nhwc_fwd(at::Tensor x, at::Tensor w,
   const int64_t N = x.size(0), C = x.size(1);
   const int64_t R = x.size(2) * x.size(3);
   TORCH_CHECK(C % groups == 0 && C % 4 == 0);
-  auto wf = as_f32(w), bf = as_f32(b);
+  bool wf32;
+  auto wf = norm_weight(w, x, wf32);
+  bool bf32;
+  auto bf = norm_weight(b, x, bf32);
+  TORCH_CHECK(wf32 == bf32);
   auto y = at::empty_like(x);  // preserves channels_last
   auto ws = at::zeros({N * groups * 2}, x.options().dtype(at::kFloat));
   auto mean = at::empty({N * groups}, x.options().dtype(at::kFloat));
   auto rstd = at::empty_like(mean);
-  gn_nhwc_fwd_launch(dtype_of(x), x.data_ptr(), wf.data_ptr<float>(),
-                     bf.data_ptr<float>(), y.data_ptr(), ws.data_ptr<float>(),
+  gn_nhwc_fwd_launch(dtype_of(x), x.data_ptr(), wf.data_ptr(), bf.data_ptr(),
+                     wf32, y.data_ptr(), ws.data_ptr<float>(),
                      mean.data_ptr<float>(), rstd.data_ptr<float>(), (int)N,
                      (int)R, (int)C, (int)groups, (float)eps, silu, cur_stream());
   return {y, mean, rstd};
@@ -100,13 +127,18 @@ std::vector<at::Tensor> groupnorm_silu_nhwc_bwd(at::Tensor dy, at::Tensor x,
               x.is_contiguous(at::MemoryFormat::ChannelsLast));
   const int64_t N = x.size(0), C = x.size(1);
   const int64_t R = x.size(2) * x.size(3);
-  auto wf = as_f32(w), bf = as_f32(b);
+  bool wf32;
+  auto wf = norm_weight(w, x, wf32);
+  bool bf32;
+  auto bf = norm_weight(b, x, bf32);
   auto dx = at::empty_like(x);
-  auto ws = at::zeros({N * groups * 2}, x.options().dtype(at::kFloat));
-  auto dw = at::zeros({C}, x.options().dtype(at::kFloat));
-  auto db = at::zeros({C}, x.options().dtype(at::kFloat));
+  auto scratch = at::zeros({N * groups * 2 + 2 * C},
+                           x.options().dtype(at::kFloat));
+  auto ws = scratch.narrow(0, 0, N * groups * 2);
+  auto dw = scratch.narrow(0, N * groups * 2, C);
+  auto db = scratch.narrow(0, N * groups * 2 + C, C);
   gn_nhwc_bwd_launch(dtype_of(x), dy.data_ptr(), x.data_ptr(),
-                     wf.data_ptr<float>(), bf.data_ptr<float>(),
+                     wf.data_ptr(), bf.data_ptr(), wf32,
                      mean.data_ptr<float>(), rstd.data_ptr<float>(),
                      ws.data_ptr<float>(), dx.data_ptr(), dw.data_ptr<float>(),
                      db.data_ptr<float>(), (int)N, (int)R, (int)C, (int)groups,
@@ -120,11 +152,15 @@ std::vector<at::Tensor> layernorm_fwd(at::Tensor x, at::Tensor w, at::Tensor b,
   TORCH_CHECK(x.is_cuda() && x.is_contiguous());
   const int64_t Nd = x.size(-1);
   const int64_t M = x.numel() / Nd;
-  auto wf = as_f32(w), bf = as_f32(b);
+  bool wf32;
+  auto wf = norm_weight(w, x, wf32);
+  bool bf32;
+  auto bf = norm_weight(b, x, bf32);
+  TORCH_CHECK(wf32 == bf32);
   auto y = at::empty_like(x);
   auto mean = at::empty({M}, x.options().dtype(at::kFloat));
   auto rstd = at::empty_like(mean);
-  ln_fwd_launch(dtype_of(x), x.data_ptr(), wf.data_ptr<float>(), bf.data_ptr<float>(),
+  ln_fwd_launch(dtype_of(x), x.data_ptr(), wf.data_ptr(), bf.data_ptr(), wf32,
                 y.data_ptr(), mean.data_ptr<float>(), rstd.data_ptr<float>(),
                 M, (int)Nd, (float)eps, cur_stream());
   return {y, mean, rstd};
@@ -136,11 +172,13 @@ std::vector<at::Tensor> layernorm_bwd(at::Tensor dy, at::Tensor x, at::Tensor w,
   const int64_t Nd = x.size(-1);
   const int64_t M = x.numel() / Nd;
   TORCH_CHECK(Nd <= 16384, "ln_bwd: N too large for LDS accumulator");
-  auto wf = as_f32(w);
+  bool wf32;
+  auto wf = norm_weight(w, x, wf32);
   auto dx = at::empty_like(x);
-  auto dw = at::zeros({Nd}, x.options().dtype(at::kFloat));
-  auto db = at::zeros({Nd}, x.options().dtype(at::kFloat));
-  ln_bwd_launch(dtype_of(x), dy.data_ptr(), x.data_ptr(), wf.data_ptr<float>(),
+  auto dwdb = at::zeros({2 * Nd}, x.options().dtype(at::kFloat));
+  auto dw = dwdb.narrow(0, 0, Nd);
+  auto db = dwdb.narrow(0, Nd, Nd);
+  ln_bwd_launch(dtype_of(x), dy.data_ptr(), x.data_ptr(), wf.data_ptr(), wf32,
                 mean.data_ptr<float>(), rstd.data_ptr<float>(), dx.data_ptr(),
                 dw.data_ptr<float>(), db.data_ptr<float>(), M, (int)Nd,
                 cur_stream());

@@ -9,29 +9,31 @@ namespace dcr {
 
 enum DType { DT_F32 = 0, DT_F16 = 1, DT_BF16 = 2 };
 
-// norms.hip
-void gn_fwd_launch(DType dt, const void* x, const float* w, const float* b,
-                   void* y, float* mean, float* rstd, int NG, int G, int Cg,
-                   int HW, float eps, bool silu, hipStream_t s);
-void gn_bwd_launch(DType dt, const void* dy, const void* x, const float* w,
-                   const float* b, const float* mean, const float* rstd,
-                   void* dx, float* dw, float* db, int NG, int G, int Cg,
-                   int HW, bool silu, hipStream_t s);
-void ln_fwd_launch(DType dt, const void* x, const float* w, const float* b,
-                   void* y, float* mean, float* rstd, long M, int N, float eps,
-                   hipStream_t s);
-void ln_bwd_launch(DType dt, const void* dy, const void* x, const float* w,
-                   const float* mean, const float* rstd, void* dx, float* dw,
-                   float* db, long M, int N, hipStream_t s);
+// norms.hip (w_f32: weights/bias are fp32; else same dtype as x)
+void gn_fwd_launch(DType dt, const void* x, const void* w, const void* b,
+                   bool w_f32, void* y, float* mean, float* rstd, int NG,
+                   int G, int Cg, int HW, float eps, bool silu, hipStream_t s);
+void gn_bwd_launch(DType dt, const void* dy, const void* x, const void* w,
+                   const void* b, bool w_f32, const float* mean,
+                   const float* rstd, void* dx, float* dw, float* db, int NG,
+                   int G, int Cg, int HW, bool silu, hipStream_t s);
+void ln_fwd_launch(DType dt, const void* x, const void* w, const void* b,
+                   bool w_f32, void* y, float* mean, float* rstd, long M,
+                   int N, float eps, hipStream_t s);
+void ln_bwd_launch(DType dt, const void* dy, const void* x, const void* w,
+                   bool w_f32, const float* mean, const float* rstd, void* dx,
+                   float* dw, float* db, long M, int N, hipStream_t s);
 
-// norms_nhwc.hip (channels_last GroupNorm)
-void gn_nhwc_fwd_launch(DType dt, const void* x, const float* w, const float* b,
-                        void* y, float* ws, float* mean, float* rstd, int N,
-                        int R, int C, int G, float eps, bool silu, hipStream_t s);
-void gn_nhwc_bwd_launch(DType dt, const void* dy, const void* x, const float* w,
-                        const float* b, const float* mean, const float* rstd,
-                        float* ws, void* dx, float* dw, float* db, int N,
-                        int R, int C, int G, bool silu, hipStream_t s);
+// norms_nhwc.hip (channels_last GroupNorm; w_f32 as above)
+void gn_nhwc_fwd_launch(DType dt, const void* x, const void* w, const void* b,
+                        bool w_f32, void* y, float* ws, float* mean, float* rstd,
+                        int N, int R, int C, int G, float eps, bool silu,
+                        hipStream_t s);
+void gn_nhwc_bwd_launch(DType dt, const void* dy, const void* x, const void* w,
+                        const void* b, bool w_f32, const float* mean,
+                        const float* rstd, float* ws, void* dx, float* dw,
+                        float* db, int N, int R, int C, int G, bool silu,
+                        hipStream_t s);
 
 // elementwise.hip
 void geglu_fwd_launch(DType dt, const void* x, void* y, long M, long N,

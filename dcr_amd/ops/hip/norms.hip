@@ -18,9 +18,9 @@ using namespace dcr;
 // ===========================================================================
 // GroupNorm + SiLU forward
 // ===========================================================================
-template <typename T, bool SILU>
-__global__ void gn_fwd_kernel(const T* __restrict__ x, const float* __restrict__ w,
-                              const float* __restrict__ b, T* __restrict__ y,
+template <typename T, typename WT, bool SILU>
+__global__ void gn_fwd_kernel(const T* __restrict__ x, const WT* __restrict__ w,
+                              const WT* __restrict__ b, T* __restrict__ y,
                               float* __restrict__ mean_out, float* __restrict__ rstd_out,
                               int G, int Cg, int HW, float eps) {
   const long base = (long)blockIdx.x * Cg * HW;
@@ -49,12 +49,12 @@ __global__ void gn_fwd_kernel(const T* __restrict__ x, const float* __restrict__
   const float rs = rsqrtf(var + eps);
   if (tid == 0) { mean_out[blockIdx.x] = m; rstd_out[blockIdx.x] = rs; }
 
-  const float* wg = w + (long)g * Cg;
-  const float* bg = b + (long)g * Cg;
+  const WT* wg = w + (long)g * Cg;
+  const WT* bg = b + (long)g * Cg;
   if ((HW & 3) == 0) {
     for (int i = tid * 4; i < L; i += nthr * 4) {
       int c = i / HW;  // uniform across the 4 elements since HW % 4 == 0
-      float wc = wg[c], bc = bg[c];
+      float wc = to_f32<WT>(wg[c]), bc = to_f32<WT>(bg[c]);
       f32x4 v = load4<T>(x + base + i);
       f32x4 o;
       o.x = (v.x - m) * rs * wc + bc;
@@ -67,7 +67,7 @@ __global__ void gn_fwd_kernel(const T* __restrict__ x, const float* __restrict__
   } else {
     for (int i = tid; i < L; i += nthr) {
       int c = i / HW;
-      float z = (to_f32<T>(x[base + i]) - m) * rs * wg[c] + bg[c];
+      float z = (to_f32<T>(x[base + i]) - m) * rs * to_f32<WT>(wg[c]) + to_f32<WT>(bg[c]);
       if (SILU) z = silu(z);
       y[base + i] = from_f32<T>(z);
     }
@@ -81,9 +81,9 @@ __global__ void gn_fwd_kernel(const T* __restrict__ x, const float* __restrict__
 //   dx = rs * (w[c]*dz - mean(w*dz) - yhat * mean(w*dz*yhat))
 //   dw[c] = sum dz*yhat ; db[c] = sum dz   (fp32 atomics, LDS-staged)
 // ===========================================================================
-template <typename T, bool SILU>
+template <typename T, typename WT, bool SILU>
 __global__ void gn_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x,
-                              const float* __restrict__ w, const float* __restrict__ b,
+                              const WT* __restrict__ w, const WT* __restrict__ b,
                               const float* __restrict__ mean, const float* __restrict__ rstd,
                               T* __restrict__ dx, float* __restrict__ dw,
                               float* __restrict__ db, int G, int Cg, int HW) {
@@ -99,8 +99,8 @@ __global__ void gn_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x,
   const int nthr = blockDim.x;
   const float m = mean[blockIdx.x];
   const float rs = rstd[blockIdx.x];
-  const float* wg = w + (long)g * Cg;
-  const float* bg = b + (long)g * Cg;
+  const WT* wg = w + (long)g * Cg;
+  const WT* bg = b + (long)g * Cg;
 
   for (int c = tid; c < Cg; c += nthr) { dw_l[c] = 0.f; db_l[c] = 0.f; }
   __syncthreads();
@@ -112,7 +112,7 @@ __global__ void gn_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x,
   if ((HW & 3) == 0) {
     for (int i = tid * 4; i < L; i += nthr * 4) {
       int c = i / HW;
-      float wc = wg[c], bc = bg[c];
+      float wc = to_f32<WT>(wg[c]), bc = to_f32<WT>(bg[c]);
       f32x4 xv = load4<T>(x + base + i);
       f32x4 gv = load4<T>(dy + base + i);
       float dwp = 0.f, dbp = 0.f;
@@ -139,7 +139,7 @@ __global__ void gn_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x,
   } else {
     for (int i = tid; i < L; i += nthr) {
       int c = i / HW;
-      float wc = wg[c], bc = bg[c];
+      float wc = to_f32<WT>(wg[c]), bc = to_f32<WT>(bg[c]);
       float xe = to_f32<T>(x[base + i]);
       float ge = to_f32<T>(dy[base + i]);
       float yh = (xe - m) * rs;
@@ -160,7 +160,7 @@ __global__ void gn_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x,
   if ((HW & 3) == 0) {
     for (int i = tid * 4; i < L; i += nthr * 4) {
       int c = i / HW;
-      float wc = wg[c], bc = bg[c];
+      float wc = to_f32<WT>(wg[c]), bc = to_f32<WT>(bg[c]);
       f32x4 xv = load4<T>(x + base + i);
       f32x4 gv = load4<T>(dy + base + i);
       f32x4 o;
@@ -178,7 +178,7 @@ __global__ void gn_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x,
   } else {
     for (int i = tid; i < L; i += nthr) {
       int c = i / HW;
-      float wc = wg[c], bc = bg[c];
+      float wc = to_f32<WT>(wg[c]), bc = to_f32<WT>(bg[c]);
       float xe = to_f32<T>(x[base + i]);
       float ge = to_f32<T>(dy[base + i]);
       float yh = (xe - m) * rs;
@@ -198,9 +198,9 @@ __global__ void gn_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x,
 // ===========================================================================
 // LayerNorm forward: one wave per row
 // ===========================================================================
-template <typename T>
-__global__ void ln_fwd_kernel(const T* __restrict__ x, const float* __restrict__ w,
-                              const float* __restrict__ b, T* __restrict__ y,
+template <typename T, typename WT>
+__global__ void ln_fwd_kernel(const T* __restrict__ x, const WT* __restrict__ w,
+                              const WT* __restrict__ b, T* __restrict__ y,
                               float* __restrict__ mean_out, float* __restrict__ rstd_out,
                               long M, int N, float eps) {
   const int wid = threadIdx.x / DCR_WAVE;
@@ -232,8 +232,8 @@ __global__ void ln_fwd_kernel(const T* __restrict__ x, const float* __restrict__
   if ((N & 3) == 0) {
     for (int j = lane * 4; j < N; j += DCR_WAVE * 4) {
       f32x4 v = load4<T>(xr + j);
-      f32x4 wv = load4<float>(w + j);
-      f32x4 bv = load4<float>(b + j);
+      f32x4 wv = load4<WT>(w + j);
+      f32x4 bv = load4<WT>(b + j);
       f32x4 o;
       o.x = (v.x - m) * rs * wv.x + bv.x;
       o.y = (v.y - m) * rs * wv.y + bv.y;
@@ -243,7 +243,8 @@ __global__ void ln_fwd_kernel(const T* __restrict__ x, const float* __restrict__
     }
   } else {
     for (int j = lane; j < N; j += DCR_WAVE) {
-      yr[j] = from_f32<T>((to_f32<T>(xr[j]) - m) * rs * w[j] + b[j]);
+      yr[j] = from_f32<T>((to_f32<T>(xr[j]) - m) * rs * to_f32<WT>(w[j])
+                          + to_f32<WT>(b[j]));
     }
   }
 }
@@ -251,9 +252,9 @@ __global__ void ln_fwd_kernel(const T* __restrict__ x, const float* __restrict__
 // ===========================================================================
 // LayerNorm backward: one wave per row; dw/db via LDS[2N] then global atomics
 // ===========================================================================
-template <typename T>
+template <typename T, typename WT>
 __global__ void ln_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x,
-                              const float* __restrict__ w,
+                              const WT* __restrict__ w,
                               const float* __restrict__ mean, const float* __restrict__ rstd,
                               T* __restrict__ dx, float* __restrict__ dw,
                               float* __restrict__ db, long M, int N) {
@@ -279,7 +280,7 @@ __global__ void ln_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x,
     for (int j = lane; j < N; j += DCR_WAVE) {
       float g = to_f32<T>(gr[j]);
       float yh = (to_f32<T>(xr[j]) - m) * rs;
-      float gw = g * w[j];
+      float gw = g * to_f32<WT>(w[j]);
       s1 += gw;
       s2 += gw * yh;
       atomicAdd(&dw_l[j], g * yh);
@@ -291,7 +292,7 @@ __global__ void ln_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x,
     for (int j = lane; j < N; j += DCR_WAVE) {
       float g = to_f32<T>(gr[j]);
       float yh = (to_f32<T>(xr[j]) - m) * rs;
-      float gw = g * w[j];
+      float gw = g * to_f32<WT>(w[j]);
       dr[j] = from_f32<T>(rs * (gw - m1 - yh * m2));
     }
   }
@@ -305,28 +306,30 @@ __global__ void ln_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x,
 // ===========================================================================
 // C++ launchers (instantiated per dtype; called from bindings.cpp)
 // ===========================================================================
-#define DCR_INST_T(T)                                                                   \
-  template __global__ void gn_fwd_kernel<T, true>(const T*, const float*, const float*, \
+#define DCR_INST_T(T, WT)                                                              \
+  template __global__ void gn_fwd_kernel<T, WT, true>(const T*, const WT*, const WT*,   \
       T*, float*, float*, int, int, int, float);                                        \
-  template __global__ void gn_fwd_kernel<T, false>(const T*, const float*, const float*,\
+  template __global__ void gn_fwd_kernel<T, WT, false>(const T*, const WT*, const WT*,  \
       T*, float*, float*, int, int, int, float);                                        \
-  template __global__ void gn_bwd_kernel<T, true>(const T*, const T*, const float*,     \
-      const float*, const float*, const float*, T*, float*, float*, int, int, int);     \
-  template __global__ void gn_bwd_kernel<T, false>(const T*, const T*, const float*,    \
-      const float*, const float*, const float*, T*, float*, float*, int, int, int);     \
-  template __global__ void ln_fwd_kernel<T>(const T*, const float*, const float*, T*,   \
+  template __global__ void gn_bwd_kernel<T, WT, true>(const T*, const T*, const WT*,    \
+      const WT*, const float*, const float*, T*, float*, float*, int, int, int);        \
+  template __global__ void gn_bwd_kernel<T, WT, false>(const T*, const T*, const WT*,   \
+      const WT*, const float*, const float*, T*, float*, float*, int, int, int);        \
+  template __global__ void ln_fwd_kernel<T, WT>(const T*, const WT*, const WT*, T*,     \
       float*, float*, long, int, float);                                                \
-  template __global__ void ln_bwd_kernel<T>(const T*, const T*, const float*,           \
+  template __global__ void ln_bwd_kernel<T, WT>(const T*, const T*, const WT*,          \
       const float*, const float*, T*, float*, float*, long, int);
 
-DCR_INST_T(float)
-DCR_INST_T(__hip_bfloat16)
-DCR_INST_T(__half)
+DCR_INST_T(float, float)
+DCR_INST_T(__hip_bfloat16, float)
+DCR_INST_T(__hip_bfloat16, __hip_bfloat16)
+DCR_INST_T(__half, float)
+DCR_INST_T(__half, __half)
 
 // forward decl (definition at end of file)
-template <typename T>
+template <typename T, typename WT>
 __global__ void ln_bwd_v2_kernel(const T* __restrict__ dy, const T* __restrict__ x,
-                                 const float* __restrict__ w,
+                                 const WT* __restrict__ w,
                                  const float* __restrict__ mean,
                                  const float* __restrict__ rstd,
                                  T* __restrict__ dx, float* __restrict__ dw,
@@ -340,77 +343,95 @@ __global__ void ln_bwd_v2_kernel(const T* __restrict__ dy, const T* __restrict__
 
 namespace dcr {
 
-template <typename T>
-static void gn_fwd_t(const void* x, const float* w, const float* b, void* y,
+template <typename T, typename WT>
+static void gn_fwd_t(const void* x, const void* w, const void* b, void* y,
                      float* mean, float* rstd, int NG, int G, int Cg, int HW,
                      float eps, bool silu, hipStream_t s) {
   dim3 grid(NG), block(256);
   if (silu)
-    hipLaunchKernelGGL((gn_fwd_kernel<T, true>), grid, block, 0, s,
-                       (const T*)x, w, b, (T*)y, mean, rstd, G, Cg, HW, eps);
+    hipLaunchKernelGGL((gn_fwd_kernel<T, WT, true>), grid, block, 0, s,
+                       (const T*)x, (const WT*)w, (const WT*)b, (T*)y, mean, rstd, G, Cg, HW, eps);
   else
-    hipLaunchKernelGGL((gn_fwd_kernel<T, false>), grid, block, 0, s,
-                       (const T*)x, w, b, (T*)y, mean, rstd, G, Cg, HW, eps);
+    hipLaunchKernelGGL((gn_fwd_kernel<T, WT, false>), grid, block, 0, s,
+                       (const T*)x, (const WT*)w, (const WT*)b, (T*)y, mean, rstd, G, Cg, HW, eps);
 }
 
-void gn_fwd_launch(DType dt, const void* x, const float* w, const float* b,
-                   void* y, float* mean, float* rstd, int NG, int G, int Cg,
-                   int HW, float eps, bool silu, hipStream_t s) {
+void gn_fwd_launch(DType dt, const void* x, const void* w, const void* b,
+                   bool w_f32, void* y, float* mean, float* rstd, int NG,
+                   int G, int Cg, int HW, float eps, bool silu, hipStream_t s) {
   switch (dt) {
-    case DT_F32: gn_fwd_t<float>(x, w, b, y, mean, rstd, NG, G, Cg, HW, eps, silu, s); break;
-    case DT_F16: gn_fwd_t<__half>(x, w, b, y, mean, rstd, NG, G, Cg, HW, eps, silu, s); break;
-    case DT_BF16: gn_fwd_t<__hip_bfloat16>(x, w, b, y, mean, rstd, NG, G, Cg, HW, eps, silu, s); break;
+    case DT_F32: gn_fwd_t<float, float>(x, w, b, y, mean, rstd, NG, G, Cg, HW, eps, silu, s); break;
+    case DT_F16:
+      if (w_f32) gn_fwd_t<__half, float>(x, w, b, y, mean, rstd, NG, G, Cg, HW, eps, silu, s);
+      else gn_fwd_t<__half, __half>(x, w, b, y, mean, rstd, NG, G, Cg, HW, eps, silu, s);
+      break;
+    case DT_BF16:
+      if (w_f32) gn_fwd_t<__hip_bfloat16, float>(x, w, b, y, mean, rstd, NG, G, Cg, HW, eps, silu, s);
+      else gn_fwd_t<__hip_bfloat16, __hip_bfloat16>(x, w, b, y, mean, rstd, NG, G, Cg, HW, eps, silu, s);
+      break;
   }
 }
 
-template <typename T>
-static void gn_bwd_t(const void* dy, const void* x, const float* w, const float* b,
+template <typename T, typename WT>
+static void gn_bwd_t(const void* dy, const void* x, const void* w, const void* b,
                      const float* mean, const float* rstd, void* dx, float* dw,
                      float* db, int NG, int G, int Cg, int HW, bool silu,
                      hipStream_t s) {
   dim3 grid(NG), block(256);
   size_t lds = (2 * Cg + 32) * sizeof(float);
   if (silu)
-    hipLaunchKernelGGL((gn_bwd_kernel<T, true>), grid, block, lds, s,
-                       (const T*)dy, (const T*)x, w, b, mean, rstd, (T*)dx, dw, db, G, Cg, HW);
+    hipLaunchKernelGGL((gn_bwd_kernel<T, WT, true>), grid, block, lds, s,
+                       (const T*)dy, (const T*)x, (const WT*)w, (const WT*)b, mean, rstd, (T*)dx, dw, db, G, Cg, HW);
   else
-    hipLaunchKernelGGL((gn_bwd_kernel<T, false>), grid, block, lds, s,
-                       (const T*)dy, (const T*)x, w, b, mean, rstd, (T*)dx, dw, db, G, Cg, HW);
+    hipLaunchKernelGGL((gn_bwd_kernel<T, WT, false>), grid, block, lds, s,
+                       (const T*)dy, (const T*)x, (const WT*)w, (const WT*)b, mean, rstd, (T*)dx, dw, db, G, Cg, HW);
 }
 
-void gn_bwd_launch(DType dt, const void* dy, const void* x, const float* w,
-                   const float* b, const float* mean, const float* rstd,
+void gn_bwd_launch(DType dt, const void* dy, const void* x, const void* w,
+                   const void* b, bool w_f32, const float* mean, const float* rstd,
                    void* dx, float* dw, float* db, int NG, int G, int Cg,
                    int HW, bool silu, hipStream_t s) {
   switch (dt) {
-    case DT_F32: gn_bwd_t<float>(dy, x, w, b, mean, rstd, dx, dw, db, NG, G, Cg, HW, silu, s); break;
-    case DT_F16: gn_bwd_t<__half>(dy, x, w, b, mean, rstd, dx, dw, db, NG, G, Cg, HW, silu, s); break;
-    case DT_BF16: gn_bwd_t<__hip_bfloat16>(dy, x, w, b, mean, rstd, dx, dw, db, NG, G, Cg, HW, silu, s); break;
+    case DT_F32: gn_bwd_t<float, float>(dy, x, w, b, mean, rstd, dx, dw, db, NG, G, Cg, HW, silu, s); break;
+    case DT_F16:
+      if (w_f32) gn_bwd_t<__half, float>(dy, x, w, b, mean, rstd, dx, dw, db, NG, G, Cg, HW, silu, s);
+      else gn_bwd_t<__half, __half>(dy, x, w, b, mean, rstd, dx, dw, db, NG, G, Cg, HW, silu, s);
+      break;
+    case DT_BF16:
+      if (w_f32) gn_bwd_t<__hip_bfloat16, float>(dy, x, w, b, mean, rstd, dx, dw, db, NG, G, Cg, HW, silu, s);
+      else gn_bwd_t<__hip_bfloat16, __hip_bfloat16>(dy, x, w, b, mean, rstd, dx, dw, db, NG, G, Cg, HW, silu, s);
+      break;
   }
 }
 
-template <typename T>
-static void ln_fwd_t(const void* x, const float* w, const float* b, void* y,
+template <typename T, typename WT>
+static void ln_fwd_t(const void* x, const void* w, const void* b, void* y,
                      float* mean, float* rstd, long M, int N, float eps,
                      hipStream_t s) {
   const int waves = 4;
   dim3 grid((M + waves - 1) / waves), block(waves * DCR_WAVE);
-  hipLaunchKernelGGL((ln_fwd_kernel<T>), grid, block, 0, s, (const T*)x, w, b,
-                     (T*)y, mean, rstd, M, N, eps);
+  hipLaunchKernelGGL((ln_fwd_kernel<T, WT>), grid, block, 0, s, (const T*)x,
+                     (const WT*)w, (const WT*)b, (T*)y, mean, rstd, M, N, eps);
 }
 
-void ln_fwd_launch(DType dt, const void* x, const float* w, const float* b,
-                   void* y, float* mean, float* rstd, long M, int N, float eps,
-                   hipStream_t s) {
+void ln_fwd_launch(DType dt, const void* x, const void* w, const void* b,
+                   bool w_f32, void* y, float* mean, float* rstd, long M,
+                   int N, float eps, hipStream_t s) {
   switch (dt) {
-    case DT_F32: ln_fwd_t<float>(x, w, b, y, mean, rstd, M, N, eps, s); break;
-    case DT_F16: ln_fwd_t<__half>(x, w, b, y, mean, rstd, M, N, eps, s); break;
-    case DT_BF16: ln_fwd_t<__hip_bfloat16>(x, w, b, y, mean, rstd, M, N, eps, s); break;
+    case DT_F32: ln_fwd_t<float, float>(x, w, b, y, mean, rstd, M, N, eps, s); break;
+    case DT_F16:
+      if (w_f32) ln_fwd_t<__half, float>(x, w, b, y, mean, rstd, M, N, eps, s);
+      else ln_fwd_t<__half, __half>(x, w, b, y, mean, rstd, M, N, eps, s);
+      break;
+    case DT_BF16:
+      if (w_f32) ln_fwd_t<__hip_bfloat16, float>(x, w, b, y, mean, rstd, M, N, eps, s);
+      else ln_fwd_t<__hip_bfloat16, __hip_bfloat16>(x, w, b, y, mean, rstd, M, N, eps, s);
+      break;
   }
 }
 
-template <typename T>
-static void ln_bwd_t(const void* dy, const void* x, const float* w,
+template <typename T, typename WT>
+static void ln_bwd_t(const void* dy, const void* x, const void* w,
                      const float* mean, const float* rstd, void* dx, float* dw,
                      float* db, long M, int N, hipStream_t s) {
   size_t lds = 2 * (size_t)N * sizeof(float);
@@ -424,24 +445,30 @@ static void ln_bwd_t(const void* dy, const void* x, const float* w,
       blocks = (M + waves - 1) / waves;
     }
     dim3 grid((unsigned)blocks), block(waves * DCR_WAVE);
-    hipLaunchKernelGGL((ln_bwd_v2_kernel<T>), grid, block, lds, s, (const T*)dy,
-                       (const T*)x, w, mean, rstd, (T*)dx, dw, db, M, N,
+    hipLaunchKernelGGL((ln_bwd_v2_kernel<T, WT>), grid, block, lds, s, (const T*)dy,
+                       (const T*)x, (const WT*)w, mean, rstd, (T*)dx, dw, db, M, N,
                        rows_per_wave);
     return;
   }
   const int waves = 4;
   dim3 grid((M + waves - 1) / waves), block(waves * DCR_WAVE);
-  hipLaunchKernelGGL((ln_bwd_kernel<T>), grid, block, lds, s, (const T*)dy,
-                     (const T*)x, w, mean, rstd, (T*)dx, dw, db, M, N);
+  hipLaunchKernelGGL((ln_bwd_kernel<T, WT>), grid, block, lds, s, (const T*)dy,
+                     (const T*)x, (const WT*)w, mean, rstd, (T*)dx, dw, db, M, N);
 }
 
-void ln_bwd_launch(DType dt, const void* dy, const void* x, const float* w,
-                   const float* mean, const float* rstd, void* dx, float* dw,
-                   float* db, long M, int N, hipStream_t s) {
+void ln_bwd_launch(DType dt, const void* dy, const void* x, const void* w,
+                   bool w_f32, const float* mean, const float* rstd, void* dx,
+                   float* dw, float* db, long M, int N, hipStream_t s) {
   switch (dt) {
-    case DT_F32: ln_bwd_t<float>(dy, x, w, mean, rstd, dx, dw, db, M, N, s); break;
-    case DT_F16: ln_bwd_t<__half>(dy, x, w, mean, rstd, dx, dw, db, M, N, s); break;
-    case DT_BF16: ln_bwd_t<__hip_bfloat16>(dy, x, w, mean, rstd, dx, dw, db, M, N, s); break;
+    case DT_F32: ln_bwd_t<float, float>(dy, x, w, mean, rstd, dx, dw, db, M, N, s); break;
+    case DT_F16:
+      if (w_f32) ln_bwd_t<__half, float>(dy, x, w, mean, rstd, dx, dw, db, M, N, s);
+      else ln_bwd_t<__half, __half>(dy, x, w, mean, rstd, dx, dw, db, M, N, s);
+      break;
+    case DT_BF16:
+      if (w_f32) ln_bwd_t<__hip_bfloat16, float>(dy, x, w, mean, rstd, dx, dw, db, M, N, s);
+      else ln_bwd_t<__hip_bfloat16, __hip_bfloat16>(dy, x, w, mean, rstd, dx, dw, db, M, N, s);
+      break;
   }
 }
 
@@ -456,10 +483,10 @@ void ln_bwd_launch(DType dt, const void* dy, const void* x, const float* w,
 // ===========================================================================
 #define LNB2_MAXCHUNK 8  // N <= 8*256 = 2048
 
-template <typename T>
+template <typename T, typename WT>
 __global__ __launch_bounds__(256)
 void ln_bwd_v2_kernel(const T* __restrict__ dy, const T* __restrict__ x,
-                      const float* __restrict__ w,
+                      const WT* __restrict__ w,
                       const float* __restrict__ mean, const float* __restrict__ rstd,
                       T* __restrict__ dx, float* __restrict__ dw,
                       float* __restrict__ db, long M, int N, int rows_per_wave) {
@@ -495,7 +522,7 @@ void ln_bwd_v2_kernel(const T* __restrict__ dy, const T* __restrict__ x,
       if (c < nchunk && j < N) {
         f32x4 xx = load4<T>(xr + j);
         f32x4 gg = load4<T>(gr + j);
-        f32x4 ww = load4<float>(w + j);
+        f32x4 ww = load4<WT>(w + j);
 #pragma unroll
         for (int k = 0; k < 4; ++k) {
           xv[c][k] = (&xx.x)[k];
@@ -554,11 +581,13 @@ void ln_bwd_v2_wire() {}  // anchor
 
 }  // namespace dcr
 
-#define DCR_INST_LNB2(T)                                                       \
-  template __global__ void ln_bwd_v2_kernel<T>(const T*, const T*,             \
-      const float*, const float*, const float*, T*, float*, float*, long,      \
+#define DCR_INST_LNB2(T, WT)                                                   \
+  template __global__ void ln_bwd_v2_kernel<T, WT>(const T*, const T*,         \
+      const WT*, const float*, const float*, T*, float*, float*, long,         \
       int, int);
 
-DCR_INST_LNB2(float)
-DCR_INST_LNB2(__hip_bfloat16)
-DCR_INST_LNB2(__half)
+DCR_INST_LNB2(float, float)
+DCR_INST_LNB2(__hip_bfloat16, float)
+DCR_INST_LNB2(__hip_bfloat16, __hip_bfloat16)
+DCR_INST_LNB2(__half, float)
+DCR_INST_LNB2(__half, __half)

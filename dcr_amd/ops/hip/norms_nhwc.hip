@@ -66,9 +66,9 @@ __global__ void gn_finalize_kernel(const float* __restrict__ ws,
 }
 
 // --------------------------------------------------------------- apply fwd
-template <typename T, bool SILU>
-__global__ void gn_nhwc_apply_kernel(const T* __restrict__ x, const float* __restrict__ w,
-                                     const float* __restrict__ b,
+template <typename T, typename WT, bool SILU>
+__global__ void gn_nhwc_apply_kernel(const T* __restrict__ x, const WT* __restrict__ w,
+                                     const WT* __restrict__ b,
                                      const float* __restrict__ mean,
                                      const float* __restrict__ rstd,
                                      T* __restrict__ y, long total, int R, int C, int G) {
@@ -81,8 +81,8 @@ __global__ void gn_nhwc_apply_kernel(const T* __restrict__ x, const float* __res
     long n = e / RC;
     int c = (int)(e % C);                  // C % 4 == 0: 4 consecutive c
     f32x4 xv = load4<T>(x + e);
-    f32x4 wv = load4<float>(w + c);
-    f32x4 bv = load4<float>(b + c);
+    f32x4 wv = load4<WT>(w + c);
+    f32x4 bv = load4<WT>(b + c);
     f32x4 o;
 #pragma unroll
     for (int k = 0; k < 4; ++k) {
@@ -98,9 +98,9 @@ __global__ void gn_nhwc_apply_kernel(const T* __restrict__ x, const float* __res
 
 // --------------------------------------------------------------- stats bwd
 // per-channel dw/db; per-group S1 = sum(w*dz), S2 = sum(w*dz*yhat)
-template <typename T, bool SILU>
+template <typename T, typename WT, bool SILU>
 __global__ void gn_nhwc_bwd_stats_kernel(const T* __restrict__ dy, const T* __restrict__ x,
-                                         const float* __restrict__ w, const float* __restrict__ b_,
+                                         const WT* __restrict__ w, const WT* __restrict__ b_,
                                          const float* __restrict__ mean,
                                          const float* __restrict__ rstd,
                                          float* __restrict__ ws, float* __restrict__ dw,
@@ -124,8 +124,8 @@ __global__ void gn_nhwc_bwd_stats_kernel(const T* __restrict__ dy, const T* __re
       const int g = (c4 + k) / Cg;
       m[k] = mean[(long)n * G + g];
       rs[k] = rstd[(long)n * G + g];
-      wc[k] = w[c4 + k];
-      bc[k] = b_[c4 + k];
+      wc[k] = to_f32<WT>(w[c4 + k]);
+      bc[k] = to_f32<WT>(b_[c4 + k]);
     }
     for (int r = r0; r < r1; ++r) {
       long idx = base + (long)r * C + c4;
@@ -161,9 +161,9 @@ __global__ void gn_nhwc_bwd_stats_kernel(const T* __restrict__ dy, const T* __re
 }
 
 // --------------------------------------------------------------- apply bwd
-template <typename T, bool SILU>
+template <typename T, typename WT, bool SILU>
 __global__ void gn_nhwc_bwd_apply_kernel(const T* __restrict__ dy, const T* __restrict__ x,
-                                         const float* __restrict__ w, const float* __restrict__ b_,
+                                         const WT* __restrict__ w, const WT* __restrict__ b_,
                                          const float* __restrict__ mean,
                                          const float* __restrict__ rstd,
                                          const float* __restrict__ ws,
@@ -179,8 +179,8 @@ __global__ void gn_nhwc_bwd_apply_kernel(const T* __restrict__ dy, const T* __re
     int c = (int)(e % C);
     f32x4 xv = load4<T>(x + e);
     f32x4 gv = load4<T>(dy + e);
-    f32x4 wv = load4<float>(w + c);
-    f32x4 bv = load4<float>(b_ + c);
+    f32x4 wv = load4<WT>(w + c);
+    f32x4 bv = load4<WT>(b_ + c);
     f32x4 o;
 #pragma unroll
     for (int k = 0; k < 4; ++k) {
@@ -216,8 +216,8 @@ static inline int nhwc_row_chunks(int N, int R) {
   return chunks;
 }
 
-template <typename T>
-static void gn_nhwc_fwd_t(const void* x, const float* w, const float* b, void* y,
+template <typename T, typename WT>
+static void gn_nhwc_fwd_t(const void* x, const void* w, const void* b, void* y,
                           float* ws, float* mean, float* rstd, int N, int R,
                           int C, int G, float eps, bool silu, hipStream_t s) {
   int chunks = nhwc_row_chunks(N, R);
@@ -233,26 +233,33 @@ static void gn_nhwc_fwd_t(const void* x, const float* w, const float* b, void* y
   long total = (long)N * R * C;
   dim3 agrid((int)min((total / 4 + 255) / 256, (long)8192)), ablock(256);
   if (silu)
-    hipLaunchKernelGGL((dcr_nhwc::gn_nhwc_apply_kernel<T, true>), agrid, ablock, 0, s,
-                       (const T*)x, w, b, mean, rstd, (T*)y, total, R, C, G);
+    hipLaunchKernelGGL((dcr_nhwc::gn_nhwc_apply_kernel<T, WT, true>), agrid, ablock, 0, s,
+                       (const T*)x, (const WT*)w, (const WT*)b, mean, rstd, (T*)y, total, R, C, G);
   else
-    hipLaunchKernelGGL((dcr_nhwc::gn_nhwc_apply_kernel<T, false>), agrid, ablock, 0, s,
-                       (const T*)x, w, b, mean, rstd, (T*)y, total, R, C, G);
+    hipLaunchKernelGGL((dcr_nhwc::gn_nhwc_apply_kernel<T, WT, false>), agrid, ablock, 0, s,
+                       (const T*)x, (const WT*)w, (const WT*)b, mean, rstd, (T*)y, total, R, C, G);
 }
 
-void gn_nhwc_fwd_launch(DType dt, const void* x, const float* w, const float* b,
-                        void* y, float* ws, float* mean, float* rstd, int N,
-                        int R, int C, int G, float eps, bool silu, hipStream_t s) {
+void gn_nhwc_fwd_launch(DType dt, const void* x, const void* w, const void* b,
+                        bool w_f32, void* y, float* ws, float* mean, float* rstd,
+                        int N, int R, int C, int G, float eps, bool silu,
+                        hipStream_t s) {
   switch (dt) {
-    case DT_F32: gn_nhwc_fwd_t<float>(x, w, b, y, ws, mean, rstd, N, R, C, G, eps, silu, s); break;
-    case DT_F16: gn_nhwc_fwd_t<__half>(x, w, b, y, ws, mean, rstd, N, R, C, G, eps, silu, s); break;
-    case DT_BF16: gn_nhwc_fwd_t<__hip_bfloat16>(x, w, b, y, ws, mean, rstd, N, R, C, G, eps, silu, s); break;
+    case DT_F32: gn_nhwc_fwd_t<float, float>(x, w, b, y, ws, mean, rstd, N, R, C, G, eps, silu, s); break;
+    case DT_F16:
+      if (w_f32) gn_nhwc_fwd_t<__half, float>(x, w, b, y, ws, mean, rstd, N, R, C, G, eps, silu, s);
+      else gn_nhwc_fwd_t<__half, __half>(x, w, b, y, ws, mean, rstd, N, R, C, G, eps, silu, s);
+      break;
+    case DT_BF16:
+      if (w_f32) gn_nhwc_fwd_t<__hip_bfloat16, float>(x, w, b, y, ws, mean, rstd, N, R, C, G, eps, silu, s);
+      else gn_nhwc_fwd_t<__hip_bfloat16, __hip_bfloat16>(x, w, b, y, ws, mean, rstd, N, R, C, G, eps, silu, s);
+      break;
   }
 }
 
-template <typename T>
-static void gn_nhwc_bwd_t(const void* dy, const void* x, const float* w,
-                          const float* b, const float* mean, const float* rstd,
+template <typename T, typename WT>
+static void gn_nhwc_bwd_t(const void* dy, const void* x, const void* w,
+                          const void* b, const float* mean, const float* rstd,
                           float* ws, void* dx, float* dw, float* db, int N,
                           int R, int C, int G, bool silu, hipStream_t s) {
   int chunks = nhwc_row_chunks(N, R);
@@ -260,34 +267,41 @@ static void gn_nhwc_bwd_t(const void* dy, const void* x, const float* w,
   dim3 grid(chunks, N), block(256);
   size_t lds = 2 * (size_t)C * sizeof(float);
   if (silu)
-    hipLaunchKernelGGL((dcr_nhwc::gn_nhwc_bwd_stats_kernel<T, true>), grid, block, lds, s,
-                       (const T*)dy, (const T*)x, w, b, mean, rstd, ws, dw, db,
+    hipLaunchKernelGGL((dcr_nhwc::gn_nhwc_bwd_stats_kernel<T, WT, true>), grid, block, lds, s,
+                       (const T*)dy, (const T*)x, (const WT*)w, (const WT*)b, mean, rstd, ws, dw, db,
                        N, R, C, G, rows_per_blk);
   else
-    hipLaunchKernelGGL((dcr_nhwc::gn_nhwc_bwd_stats_kernel<T, false>), grid, block, lds, s,
-                       (const T*)dy, (const T*)x, w, b, mean, rstd, ws, dw, db,
+    hipLaunchKernelGGL((dcr_nhwc::gn_nhwc_bwd_stats_kernel<T, WT, false>), grid, block, lds, s,
+                       (const T*)dy, (const T*)x, (const WT*)w, (const WT*)b, mean, rstd, ws, dw, db,
                        N, R, C, G, rows_per_blk);
   long total = (long)N * R * C;
   float inv_L = 1.f / ((float)R * (C / G));
   dim3 agrid((int)min((total / 4 + 255) / 256, (long)8192)), ablock(256);
   if (silu)
-    hipLaunchKernelGGL((dcr_nhwc::gn_nhwc_bwd_apply_kernel<T, true>), agrid, ablock, 0, s,
-                       (const T*)dy, (const T*)x, w, b, mean, rstd, ws, (T*)dx,
+    hipLaunchKernelGGL((dcr_nhwc::gn_nhwc_bwd_apply_kernel<T, WT, true>), agrid, ablock, 0, s,
+                       (const T*)dy, (const T*)x, (const WT*)w, (const WT*)b, mean, rstd, ws, (T*)dx,
                        total, R, C, G, inv_L);
   else
-    hipLaunchKernelGGL((dcr_nhwc::gn_nhwc_bwd_apply_kernel<T, false>), agrid, ablock, 0, s,
-                       (const T*)dy, (const T*)x, w, b, mean, rstd, ws, (T*)dx,
+    hipLaunchKernelGGL((dcr_nhwc::gn_nhwc_bwd_apply_kernel<T, WT, false>), agrid, ablock, 0, s,
+                       (const T*)dy, (const T*)x, (const WT*)w, (const WT*)b, mean, rstd, ws, (T*)dx,
                        total, R, C, G, inv_L);
 }
 
-void gn_nhwc_bwd_launch(DType dt, const void* dy, const void* x, const float* w,
-                        const float* b, const float* mean, const float* rstd,
-                        float* ws, void* dx, float* dw, float* db, int N,
-                        int R, int C, int G, bool silu, hipStream_t s) {
+void gn_nhwc_bwd_launch(DType dt, const void* dy, const void* x, const void* w,
+                        const void* b, bool w_f32, const float* mean,
+                        const float* rstd, float* ws, void* dx, float* dw,
+                        float* db, int N, int R, int C, int G, bool silu,
+                        hipStream_t s) {
   switch (dt) {
-    case DT_F32: gn_nhwc_bwd_t<float>(dy, x, w, b, mean, rstd, ws, dx, dw, db, N, R, C, G, silu, s); break;
-    case DT_F16: gn_nhwc_bwd_t<__half>(dy, x, w, b, mean, rstd, ws, dx, dw, db, N, R, C, G, silu, s); break;
-    case DT_BF16: gn_nhwc_bwd_t<__hip_bfloat16>(dy, x, w, b, mean, rstd, ws, dx, dw, db, N, R, C, G, silu, s); break;
+    case DT_F32: gn_nhwc_bwd_t<float, float>(dy, x, w, b, mean, rstd, ws, dx, dw, db, N, R, C, G, silu, s); break;
+    case DT_F16:
+      if (w_f32) gn_nhwc_bwd_t<__half, float>(dy, x, w, b, mean, rstd, ws, dx, dw, db, N, R, C, G, silu, s);
+      else gn_nhwc_bwd_t<__half, __half>(dy, x, w, b, mean, rstd, ws, dx, dw, db, N, R, C, G, silu, s);
+      break;
+    case DT_BF16:
+      if (w_f32) gn_nhwc_bwd_t<__hip_bfloat16, float>(dy, x, w, b, mean, rstd, ws, dx, dw, db, N, R, C, G, silu, s);
+      else gn_nhwc_bwd_t<__hip_bfloat16, __hip_bfloat16>(dy, x, w, b, mean, rstd, ws, dx, dw, db, N, R, C, G, silu, s);
+      break;
   }
 }
 
