@@ -67,6 +67,9 @@ def parse_args():
     p.add_argument("--imsize", default=224, type=int)
     p.add_argument("--noeval", action="store_true")
     p.add_argument("--skip_fid", action="store_true")
+    p.add_argument("--ipr", action="store_true",
+                   help="Improved Precision & Recall (reference imports IPR at "
+                        "diff_retrieval.py:587 but left the call commented out)")
     p.add_argument("--project", default="imsimv2_retrieval", type=str)
     return p.parse_args()
 
@@ -219,6 +222,15 @@ def main():
                                         dims=2048)
         tracker.log({"fid": fid})
         print(f"FID: {fid:.3f}")
+
+    # Improved Precision & Recall (VGG16 manifold; reference :587,601-605)
+    if args.ipr and not args.noeval:
+        from dcr_amd.metrics import IPR
+        ipr = IPR(batch_size=16, k=3, device=str(device))
+        ipr.compute_manifold_ref(args.val_dir)
+        pr = ipr.precision_and_recall(args.query_dir)
+        tracker.log({"precision": pr.precision, "recall": pr.recall})
+        print(f"IPR precision={pr.precision:.4f} recall={pr.recall:.4f}")
 
     # gallery: top matches for the most-copied generations (reference :609-640)
     if not args.dontsave:

@@ -133,3 +133,24 @@ def test_retrieval_cli_end_to_end(tmp_path):
     log = out / "imsimv2_retrieval_log.jsonl"
     recs = [json.loads(l) for l in log.read_text().splitlines()]
     assert any("sim_mean" in r_ for r_ in recs)
+
+
+@pytest.mark.timeout(900)
+def test_train_cli_end_to_end(tmp_path):
+    """diff_train.py tiny synthetic run writes mangled output dir +
+    checkpoint layout + jsonl logs."""
+    r = subprocess.run(
+        [sys.executable, "diff_train.py", "--synthetic_data",
+         "--model_size", "tiny", "--resolution", "64",
+         "--train_batch_size", "2", "--max_train_steps", "2",
+         "--mixed_precision", "no", "--class_prompt", "classlevel",
+         "--num_workers", "0", "--seed", "0", "--save_steps", "1000",
+         "--modelsavesteps", "1000",
+         "--output_dir", str(tmp_path / "m")],
+        capture_output=True, text=True, cwd=str(Path(__file__).parent.parent),
+        timeout=870)
+    assert r.returncode == 0, r.stderr[-3000:]
+    out = tmp_path / "m_classlevel_nodup"
+    assert (out / "checkpoint" / "unet" / "config.json").exists(), \
+        list(tmp_path.iterdir())
+    assert (out / "checkpoint" / "state.pt").exists()
