@@ -154,3 +154,26 @@ def test_train_cli_end_to_end(tmp_path):
     assert (out / "checkpoint" / "unet" / "config.json").exists(), \
         list(tmp_path.iterdir())
     assert (out / "checkpoint" / "state.pt").exists()
+
+
+@pytest.mark.timeout(900)
+def test_inference_cli_end_to_end(tmp_path, monkeypatch):
+    """diff_inference.py from a tiny checkpoint: prompts.txt + numbered pngs."""
+    ckpt_root = tmp_path / "run_imagenette_classlevel_nodup"
+    pipe = tiny_pipe()
+    pipe.save_pretrained(ckpt_root / "checkpoint")
+    r = subprocess.run(
+        [sys.executable,
+         str(Path(__file__).parent.parent / "diff_inference.py"),
+         "--modelpath", str(ckpt_root),
+         "-nb", "2", "-imb", "2", "--resolution", "64", "--seed", "0"],
+        capture_output=True, text=True, cwd=str(tmp_path),
+        env={**__import__("os").environ,
+             "PYTHONPATH": str(Path(__file__).parent.parent)},
+        timeout=870)
+    assert r.returncode == 0, r.stderr[-3000:]
+    gen = tmp_path / "inferences" / "imagenette10_frozentext" / \
+        "run_imagenette_classlevel_nodup" / "classlevel"
+    assert (gen / "prompts.txt").exists(), r.stdout[-2000:]
+    pngs = sorted((gen / "generations").glob("*.png"))
+    assert [p.name for p in pngs] == ["0.png", "1.png", "2.png", "3.png"]

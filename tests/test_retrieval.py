@@ -177,3 +177,33 @@ def test_compute_map():
     assert aps[0] == 1.0
     assert aps[1] == 1.0  # junk(2) ranked above ok(3) is ignored
     assert m == 1.0
+
+
+def test_fid_stats_caching(tmp_path):
+    """save_fid_stats -> .npz consumed by calculate_fid_given_paths
+    (reference fid.py:226-228,258-275)."""
+    from dcr_amd.metrics import calculate_fid_given_paths, save_fid_stats
+    rng = np.random.default_rng(1)
+    d = tmp_path / "imgs"
+    d.mkdir()
+    for i in range(10):
+        Image.fromarray(rng.integers(0, 255, (48, 48, 3)).astype(np.uint8)) \
+            .save(d / f"{i}.png")
+    npz = tmp_path / "stats.npz"
+    save_fid_stats(str(d), str(npz), batch_size=5, dims=64)
+    assert npz.exists()
+    # identical statistics up to sqrtm noise on a low-rank covariance
+    fid = calculate_fid_given_paths([str(npz), str(d)], batch_size=5,
+                                    device="cpu", dims=64)
+    assert fid < 0.5, fid
+
+
+def test_ipr_realism_score():
+    from dcr_amd.metrics.ipr import IPR
+    torch.manual_seed(3)
+    ipr = IPR(batch_size=8, k=3)
+    ref = torch.rand(12, 3, 32, 32)
+    ipr.compute_manifold_ref(ref)
+    feats = ipr.extract_features(ref[:1])
+    r = ipr.realism(feats[0])
+    assert np.isfinite(r) and r > 0
