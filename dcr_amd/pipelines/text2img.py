@@ -68,11 +68,17 @@ class StableDiffusionPipeline:
         pipe = cls(unet, vae, te, tok, scheduler, embed_noise_lam)
         return pipe
 
-    def to(self, device):
+    def to(self, device, channels_last: bool = True):
         device = torch.device(device)
         self.unet.to(device)
         self.vae.to(device)
         self.text_encoder.to(device)
+        if device.type == "cuda":
+            torch.backends.cudnn.benchmark = True
+            if channels_last:
+                # NHWC convs + NHWC GroupNorm kernels
+                self.unet.to(memory_format=torch.channels_last)
+                self.vae.to(memory_format=torch.channels_last)
         self._device = device
         return self
 
@@ -130,6 +136,9 @@ class StableDiffusionPipeline:
         if latents is None:
             latents = torch.randn((n, lc, lh, lw), generator=generator,
                                   device=self.device, dtype=torch.float32).to(dtype)
+        if self.device.type == "cuda" and \
+                self.unet.conv_in.weight.is_contiguous(memory_format=torch.channels_last):
+            latents = latents.to(memory_format=torch.channels_last)
         latents = latents * self.scheduler.init_noise_sigma
 
         self.scheduler.set_timesteps(num_inference_steps, device=self.device)

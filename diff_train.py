@@ -57,7 +57,9 @@ def parse_args():
     p.add_argument("--hub_model_id", type=str, default=None)
     p.add_argument("--logging_dir", type=str, default="logs")
     p.add_argument("--mixed_precision", type=str, default="bf16",
-                   choices=["no", "fp16", "bf16"])
+                   choices=["no", "fp16", "bf16", "pure_bf16"])
+    p.add_argument("--channels_last", action="store_true",
+                   help="NHWC convs + NHWC GroupNorm kernels (fastest on MI355X)")
     p.add_argument("--local_rank", type=int, default=-1)
     p.add_argument("-j", "--num_workers", type=int, default=4)
     p.add_argument("--modelsavesteps", type=int, default=1000)
@@ -132,6 +134,7 @@ def main():
         adam_epsilon=args.adam_epsilon,
         max_grad_norm=args.max_grad_norm,
         mixed_precision=args.mixed_precision,
+        channels_last=args.channels_last,
         seed=args.seed,
         rand_noise_lam=args.rand_noise_lam,
         mixup_noise_lam=args.mixup_noise_lam,
