@@ -272,6 +272,13 @@ def get_velocity(
     return (sa * noise.float() - sb * x0.float()).to(x0.dtype)
 
 
+def _dense(t: torch.Tensor) -> bool:
+    """dense (no holes) in ANY layout — elementwise kernels walk raw memory."""
+    if t.is_contiguous():
+        return True
+    return t.dim() == 4 and t.is_contiguous(memory_format=torch.channels_last)
+
+
 def lincomb(x: torch.Tensor, y: torch.Tensor, a: float, b: float,
             z: Optional[torch.Tensor] = None, c: float = 0.0) -> torch.Tensor:
     """out = a*x + b*y (+ c*z) — one fused HIP kernel; the DDIM and
@@ -279,10 +286,10 @@ def lincomb(x: torch.Tensor, y: torch.Tensor, a: float, b: float,
     from . import ext
 
     m = ext()
-    if use_hip(x) and m is not None and x.numel() % 4 == 0:
-        return m.lincomb(x.contiguous(), y.contiguous(),
-                         z.contiguous() if z is not None else None,
-                         float(a), float(b), float(c))
+    if use_hip(x) and m is not None and x.numel() % 4 == 0 \
+            and _dense(x) and x.stride() == y.stride() \
+            and (z is None or z.stride() == x.stride()):
+        return m.lincomb(x, y, z, float(a), float(b), float(c))
     out = a * x.float() + b * y.float()
     if z is not None:
         out = out + c * z.float()
@@ -294,6 +301,8 @@ def cfg_combine(eps_uncond: torch.Tensor, eps_text: torch.Tensor, scale: float) 
     from . import ext
 
     m = ext()
-    if use_hip(eps_uncond) and m is not None:
-        return m.cfg_combine(eps_uncond.contiguous(), eps_text.contiguous(), float(scale))
+    if use_hip(eps_uncond) and m is not None \
+            and eps_uncond.stride() == eps_text.stride() \
+            and _dense(eps_uncond):
+        return m.cfg_combine(eps_uncond, eps_text, float(scale))
     return eps_uncond + scale * (eps_text - eps_uncond)

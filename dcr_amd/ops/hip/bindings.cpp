@@ -258,15 +258,24 @@ at::Tensor get_velocity(at::Tensor x0, at::Tensor noise, at::Tensor ac, at::Tens
   return sched_common(1, x0, noise, ac, t);
 }
 
+// elementwise over any DENSE layout (NCHW or channels_last) as long as
+// every operand shares the same strides — the kernel walks raw memory.
+static bool same_dense(const at::Tensor& a, const at::Tensor& b) {
+  return a.is_non_overlapping_and_dense() && b.is_non_overlapping_and_dense() &&
+         a.strides() == b.strides();
+}
+
 at::Tensor lincomb(at::Tensor X, at::Tensor Y,
                    c10::optional<at::Tensor> Z, double a, double b, double c) {
-  TORCH_CHECK(X.is_cuda() && X.is_contiguous() && Y.is_contiguous());
+  TORCH_CHECK(X.is_cuda());
+  TORCH_CHECK(same_dense(X, Y), "lincomb: X/Y layout mismatch");
   TORCH_CHECK(X.numel() % 4 == 0);
-  auto out = at::empty_like(X);
+  auto out = at::empty_like(X);  // preserves layout
   const void* zp = nullptr;
   at::Tensor Zc;
   if (Z.has_value()) {
-    Zc = Z->contiguous();
+    TORCH_CHECK(same_dense(X, *Z), "lincomb: X/Z layout mismatch");
+    Zc = *Z;
     zp = Zc.data_ptr();
   }
   lincomb_launch(dtype_of(X), X.data_ptr(), Y.data_ptr(), zp, out.data_ptr(),
@@ -275,7 +284,8 @@ at::Tensor lincomb(at::Tensor X, at::Tensor Y,
 }
 
 at::Tensor cfg_combine(at::Tensor eu, at::Tensor et, double scale) {
-  TORCH_CHECK(eu.is_cuda() && eu.is_contiguous() && et.is_contiguous());
+  TORCH_CHECK(eu.is_cuda());
+  TORCH_CHECK(same_dense(eu, et), "cfg: layout mismatch");
   TORCH_CHECK(eu.numel() % 4 == 0);
   auto out = at::empty_like(eu);
   cfg_launch(dtype_of(eu), eu.data_ptr(), et.data_ptr(), out.data_ptr(),
