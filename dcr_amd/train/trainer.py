@@ -168,6 +168,17 @@ class Trainer:
                 weight_pc=cfg.weight_pc, dup_weight=cfg.dup_weight, seed=cfg.seed)
             weights = getattr(self.dataset, "samplingweights", None)
 
+        # trainsubset: fraction of the dataset (reference diff_train.py:466-468)
+        if cfg.trainsubset is not None and cfg.trainsubset > 0:
+            n_keep = int(len(self.dataset) * float(cfg.trainsubset))
+            base = self.dataset
+            self.dataset = torch.utils.data.Subset(base, list(range(n_keep)))
+            # keep caption pool visible for sample prompts
+            if hasattr(base, "prompts"):
+                self.dataset.prompts = base.prompts
+            if weights is not None:
+                weights = weights[:n_keep]
+
         if weights is not None:
             # duplication sampling (reference: diff_train.py:470-479);
             # per-rank generator seed so ranks draw different samples

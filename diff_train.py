@@ -159,8 +159,16 @@ def main():
                                        tr.tokenizer, DDIMScheduler())
         was_training = tr.unet.training
         tr.unet.eval()
+        import numpy as np
         objects = tr.dataset.objects if hasattr(tr.dataset, "objects") else \
             ["church", "garbage truck", "tench"]
+        # instancelevel modes sample prompts from the caption pool
+        # (reference diff_train.py:582-591)
+        rand_prompts = None
+        prompts_map = getattr(tr.dataset, "prompts", None)
+        if cfg.class_prompt.startswith("instancelevel") and prompts_map:
+            choicelist = [v[0] for v in prompts_map.values()]
+            rand_prompts = list(np.random.choice(choicelist, 3))
         genseed = cfg.generation_seed
         for count, obj in enumerate(objects):
             if count > 2:
@@ -170,6 +178,8 @@ def main():
                 prompt = "An image"
             elif cfg.class_prompt == "classlevel":
                 prompt = f"An image of {obj}"
+            elif rand_prompts is not None:
+                prompt = str(rand_prompts[count])
             else:
                 prompt = f"An image of {obj}"
             gen = torch.Generator(tr.device.type).manual_seed(genseed)

@@ -152,3 +152,10 @@ def test_gradient_checkpointing(tmp_path):
     loss = tr.train_step(b)
     assert torch.isfinite(loss)
     assert tr.optimizer.flat_grad.abs().sum() == 0  # zeroed after step
+
+
+def test_trainsubset(tmp_path):
+    tr = Trainer(tiny_cfg(tmp_path, trainsubset=0.5, synthetic_size=8))
+    assert len(tr.dataset) == 4  # 50% of 8
+    b = next(iter(tr.dataloader))
+    assert torch.isfinite(tr.train_step(b))
