@@ -31,21 +31,49 @@ __global__ void gn_nhwc_stats_kernel(const T* __restrict__ x, float* __restrict_
   const long base = (long)n * R * C;
   const int Cg = C / G;
 
-  // vec4 over channels: thread t owns channels [4t, 4t+4) per stripe
-  for (int c4 = threadIdx.x * 4; c4 < C; c4 += blockDim.x * 4) {
-    f32x4 a = {0.f, 0.f, 0.f, 0.f}, b = {0.f, 0.f, 0.f, 0.f};
-    for (int r = r0; r < r1; ++r) {
-      f32x4 v = load4<T>(x + base + (long)r * C + c4);
+  if (C >= (int)blockDim.x * 4) {
+    // wide-C: thread t owns channels [4t, 4t+4) per stripe (full util)
+    for (int c4 = threadIdx.x * 4; c4 < C; c4 += blockDim.x * 4) {
+      f32x4 a = {0.f, 0.f, 0.f, 0.f}, b = {0.f, 0.f, 0.f, 0.f};
+      for (int r = r0; r < r1; ++r) {
+        f32x4 v = load4<T>(x + base + (long)r * C + c4);
+#pragma unroll
+        for (int k = 0; k < 4; ++k) {
+          (&a.x)[k] += (&v.x)[k];
+          (&b.x)[k] += (&v.x)[k] * (&v.x)[k];
+        }
+      }
+#pragma unroll
+      for (int k = 0; k < 4; ++k) { s1[c4 + k] = (&a.x)[k]; s2[c4 + k] = (&b.x)[k]; }
+    }
+    __syncthreads();
+  } else {
+    // narrow-C (C < 1024): split threads over (channel-vec, row group) so
+    // all 256 lanes stay busy; combine row groups via LDS atomics.
+    for (int c = threadIdx.x; c < 2 * C; c += blockDim.x) smem[c] = 0.f;
+    __syncthreads();
+    const int tpr = C / 4;
+    const int rpar = (int)blockDim.x / tpr;
+    const int cidx = (threadIdx.x % tpr) * 4;
+    const int rgrp = threadIdx.x / tpr;
+    if (rgrp < rpar) {
+      f32x4 a = {0.f, 0.f, 0.f, 0.f}, b = {0.f, 0.f, 0.f, 0.f};
+      for (int r = r0 + rgrp; r < r1; r += rpar) {
+        f32x4 v = load4<T>(x + base + (long)r * C + cidx);
+#pragma unroll
+        for (int k = 0; k < 4; ++k) {
+          (&a.x)[k] += (&v.x)[k];
+          (&b.x)[k] += (&v.x)[k] * (&v.x)[k];
+        }
+      }
 #pragma unroll
       for (int k = 0; k < 4; ++k) {
-        (&a.x)[k] += (&v.x)[k];
-        (&b.x)[k] += (&v.x)[k] * (&v.x)[k];
+        atomicAdd(&s1[cidx + k], (&a.x)[k]);
+        atomicAdd(&s2[cidx + k], (&b.x)[k]);
       }
     }
-#pragma unroll
-    for (int k = 0; k < 4; ++k) { s1[c4 + k] = (&a.x)[k]; s2[c4 + k] = (&b.x)[k]; }
+    __syncthreads();
   }
-  __syncthreads();
   for (int g = threadIdx.x; g < G; g += blockDim.x) {
     float a = 0.f, b = 0.f;
     for (int c = g * Cg; c < (g + 1) * Cg; ++c) { a += s1[c]; b += s2[c]; }
@@ -115,7 +143,18 @@ __global__ void gn_nhwc_bwd_stats_kernel(const T* __restrict__ dy, const T* __re
   const long base = (long)n * R * C;
   const int Cg = C / G;
 
-  for (int c4 = threadIdx.x * 4; c4 < C; c4 += blockDim.x * 4) {
+  // (same wide/narrow split as the forward stats kernel)
+  const bool wide = C >= (int)blockDim.x * 4;
+  int tpr = wide ? (int)blockDim.x : C / 4;
+  int rpar = wide ? 1 : (int)blockDim.x / tpr;
+  int rgrp = wide ? 0 : (int)threadIdx.x / tpr;
+  if (!wide) {
+    for (int c = threadIdx.x; c < 2 * C; c += blockDim.x) smem[c] = 0.f;
+    __syncthreads();
+  }
+  for (int c4 = (wide ? (int)threadIdx.x * 4 : ((int)threadIdx.x % tpr) * 4);
+       c4 < C; c4 += (wide ? (int)blockDim.x * 4 : C + 1)) {
+    if (rgrp >= rpar) break;
     float m[4], rs[4], wc[4], bc[4];
     float a[4] = {0, 0, 0, 0}, bb[4] = {0, 0, 0, 0};
     float dwc[4] = {0, 0, 0, 0}, dbc[4] = {0, 0, 0, 0};
@@ -127,7 +166,7 @@ __global__ void gn_nhwc_bwd_stats_kernel(const T* __restrict__ dy, const T* __re
       wc[k] = to_f32<WT>(w[c4 + k]);
       bc[k] = to_f32<WT>(b_[c4 + k]);
     }
-    for (int r = r0; r < r1; ++r) {
+    for (int r = r0 + rgrp; r < r1; r += rpar) {
       long idx = base + (long)r * C + c4;
       f32x4 xv = load4<T>(x + idx);
       f32x4 gv = load4<T>(dy + idx);
@@ -145,8 +184,13 @@ __global__ void gn_nhwc_bwd_stats_kernel(const T* __restrict__ dy, const T* __re
     }
 #pragma unroll
     for (int k = 0; k < 4; ++k) {
-      sa[c4 + k] = a[k];
-      sb[c4 + k] = bb[k];
+      if (wide) {
+        sa[c4 + k] = a[k];
+        sb[c4 + k] = bb[k];
+      } else {
+        atomicAdd(&sa[c4 + k], a[k]);
+        atomicAdd(&sb[c4 + k], bb[k]);
+      }
       atomicAdd(&dw[c4 + k], dwc[k]);
       atomicAdd(&db[c4 + k], dbc[k]);
     }
