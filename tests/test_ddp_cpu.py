@@ -91,3 +91,39 @@ def _trainer_worker(rank, world, port, tmpdir):
 def test_trainer_ddp_world2(tmp_path):
     port = 29713
     mp.spawn(_trainer_worker, args=(2, port, str(tmp_path)), nprocs=2, join=True)
+
+
+def _feat_worker(rank, world, port):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from torch.utils.data import DataLoader
+        from torch.utils.data.distributed import DistributedSampler
+        from dcr_amd.retrieval import extract_features
+        from dcr_amd.utils import MetricLogger
+
+        # deterministic dataset shared by both ranks
+        data = torch.arange(10, dtype=torch.float32).view(10, 1).repeat(1, 4)
+        ds = [(data[i].view(1, 2, 2), i) for i in range(10)]
+        sampler = DistributedSampler(ds, num_replicas=world, rank=rank,
+                                     shuffle=False)
+        loader = DataLoader(ds, batch_size=2, sampler=sampler)
+        feats = extract_features(torch.nn.Flatten(), loader, torch.device("cpu"))
+        # every rank must hold the complete, correctly-indexed matrix
+        # (reference all_gather N4/N5 semantics, utils_ret.py:763-786)
+        for i in range(10):
+            assert torch.equal(feats[i], data[i]), (rank, i, feats[i])
+
+        # SmoothedValue cross-rank sync (N6)
+        ml = MetricLogger()
+        ml.update(loss=float(rank + 1))  # rank0: 1.0, rank1: 2.0
+        ml.synchronize_between_processes()
+        assert abs(ml.meters["loss"].global_avg - 1.5) < 1e-6
+        dist.barrier()
+    finally:
+        dist.destroy_process_group()
+
+
+def test_extract_features_world2():
+    mp.spawn(_feat_worker, args=(2, 29719), nprocs=2, join=True)
