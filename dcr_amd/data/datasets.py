@@ -20,7 +20,7 @@ import json
 import pickle
 import random
 from pathlib import Path
-from typing import List, Optional, Sequence
+from typing import List, Optional
 
 import numpy as np
 import torch

@@ -9,7 +9,7 @@ matching diff_retrieval.py:597-600).
 from __future__ import annotations
 
 from pathlib import Path
-from typing import Iterable, List, Tuple, Union
+from typing import Iterable, List, Union
 
 import numpy as np
 import torch

@@ -9,7 +9,6 @@ with a causal mask.
 from __future__ import annotations
 
 import json
-import math
 from dataclasses import dataclass, asdict
 from pathlib import Path
 from typing import Optional

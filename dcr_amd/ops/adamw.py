@@ -18,7 +18,6 @@ buffer views.
 """
 from __future__ import annotations
 
-import math
 from typing import Iterable, List
 
 import torch

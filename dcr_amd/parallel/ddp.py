@@ -18,7 +18,6 @@ from __future__ import annotations
 
 from typing import List, Optional
 
-import torch
 import torch.distributed as dist
 
 from ..ops.adamw import FusedAdamW

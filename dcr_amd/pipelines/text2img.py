@@ -19,7 +19,7 @@ import torch
 from .. import ops
 from ..data.tokenizer import load_tokenizer
 from ..models import AutoencoderKL, CLIPTextModel, UNet2DConditionModel
-from ..schedulers import DDIMScheduler, DDPMScheduler, DPMSolverMultistepScheduler
+from ..schedulers import DDIMScheduler, DPMSolverMultistepScheduler
 from ..utils.image import tensor_to_pil
 
 

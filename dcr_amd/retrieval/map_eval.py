@@ -4,7 +4,7 @@ reference library exposes it, even though the DCR pipelines don't call
 it in their main paths)."""
 from __future__ import annotations
 
-from typing import Dict, List, Sequence
+from typing import Dict, Sequence
 
 import numpy as np
 

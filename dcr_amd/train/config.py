@@ -7,7 +7,7 @@ back out of the path, so the format is a compatibility contract).
 from __future__ import annotations
 
 import math
-from dataclasses import dataclass, field, asdict
+from dataclasses import dataclass, asdict
 from typing import Optional
 
 

@@ -19,7 +19,6 @@ MI355X design deltas vs the reference:
 """
 from __future__ import annotations
 
-import os
 import random
 import time
 from pathlib import Path

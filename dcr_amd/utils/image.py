@@ -4,7 +4,6 @@ from __future__ import annotations
 
 from typing import Sequence
 
-import numpy as np
 import torch
 from PIL import Image
 

@@ -135,7 +135,7 @@ def build_prompt_list(args, tokenizer):
 
 
 def main(args):
-    from dcr_amd.parallel import init_distributed_mode, get_rank, get_world_size
+    from dcr_amd.parallel import init_distributed_mode
     from dcr_amd.pipelines import StableDiffusionPipeline
     from dcr_amd.schedulers import DPMSolverMultistepScheduler
 

@@ -16,7 +16,6 @@ from __future__ import annotations
 
 import argparse
 import json
-import os
 from pathlib import Path
 
 import numpy as np
@@ -25,8 +24,7 @@ from torch.utils.data import DataLoader
 from torch.utils.data.distributed import DistributedSampler
 
 from dcr_amd.data import SynthDataset, EvalTransform
-from dcr_amd.parallel import (barrier, get_rank, get_world_size,
-                              init_distributed_mode, is_main_process)
+from dcr_amd.parallel import barrier, init_distributed_mode, is_main_process
 from dcr_amd.retrieval import (extract_features, gen_clipscore, glcm_entropy,
                                jpeg_size, l2_normalize, load_clip, load_dino,
                                load_sscd, pearson, sim_matrix,

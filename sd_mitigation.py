@@ -32,7 +32,7 @@ PROMPT_LIST = [
 def main(args):
     from dcr_amd.pipelines import StableDiffusionPipeline
     from dcr_amd.schedulers import DPMSolverMultistepScheduler
-    from dcr_amd.data.tokenizer import HashTokenizer, load_tokenizer
+    from dcr_amd.data.tokenizer import HashTokenizer
 
     device = torch.device("cuda" if torch.cuda.is_available() else "cpu")
 
