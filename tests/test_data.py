@@ -155,3 +155,24 @@ def test_bool_flag():
     assert not bool_flag("false") and not bool_flag("off")
     with _pytest.raises(Exception):
         bool_flag("maybe")
+
+
+def test_load_real_clip_tokenizer(tmp_path):
+    """load_tokenizer uses transformers CLIPTokenizer when vocab files
+    exist (reference tokenizer path, diff_train.py:371-383)."""
+    vocab = {"<|startoftext|>": 0, "<|endoftext|>": 1}
+    # minimal BPE vocab: byte-level symbols + a merged token
+    for i, ch in enumerate("abcdefghijklmnopqrstuvwxyz"):
+        vocab[ch] = 2 + i
+        vocab[ch + "</w>"] = 28 + i
+    vocab["ab</w>"] = 60
+    (tmp_path / "vocab.json").write_text(json.dumps(vocab))
+    (tmp_path / "merges.txt").write_text("#version: 0.2\na b</w>\n")
+    from dcr_amd.data.tokenizer import load_tokenizer
+    tok = load_tokenizer(tmp_path, model_max_length=77)
+    from transformers import CLIPTokenizer
+    assert isinstance(tok, CLIPTokenizer)
+    out = tok("ab", truncation=True, padding="max_length", max_length=16,
+              return_tensors="pt")
+    assert out.input_ids.shape == (1, 16)
+    assert out.input_ids[0, 0].item() == 0  # bos

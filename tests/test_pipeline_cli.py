@@ -177,3 +177,20 @@ def test_inference_cli_end_to_end(tmp_path, monkeypatch):
     assert (gen / "prompts.txt").exists(), r.stdout[-2000:]
     pngs = sorted((gen / "generations").glob("*.png"))
     assert [p.name for p in pngs] == ["0.png", "1.png", "2.png", "3.png"]
+
+
+def test_convert_diffusers_checkpoint_roundtrip(tmp_path):
+    """Our save_pretrained output IS diffusers-layout; the converter must
+    round-trip it losslessly (keys match by construction)."""
+    sys.path.insert(0, str(Path(__file__).parent.parent / "scripts"))
+    import convert_diffusers_checkpoint as conv
+    pipe = tiny_pipe()
+    src = tmp_path / "src"
+    pipe.save_pretrained(src)
+    dst = tmp_path / "dst"
+    conv.convert(src, dst, check=True)
+    pipe2 = StableDiffusionPipeline.from_pretrained(dst)
+    for k, v in pipe.unet.state_dict().items():
+        assert torch.equal(v, pipe2.unet.state_dict()[k]), k
+    for k, v in pipe.text_encoder.state_dict().items():
+        assert torch.equal(v, pipe2.text_encoder.state_dict()[k]), k
