@@ -354,12 +354,17 @@ class Trainer:
         self.tokenizer.save_pretrained(path / "tokenizer")
         self.noise_scheduler.save_pretrained(path / "scheduler")
         save_pipeline_index(path)
-        torch.save({
+        state = {
             "global_step": self.global_step,
             "optimizer": self.optimizer.state_dict(),
             "torch_rng": torch.get_rng_state(),
+            "numpy_rng": np.random.get_state(),
+            "python_rng": random.getstate(),
             "config": self.cfg.to_dict(),
-        }, path / "state.pt")
+        }
+        if torch.cuda.is_available():
+            state["cuda_rng"] = torch.cuda.get_rng_state_all()
+        torch.save(state, path / "state.pt")
 
     def load_checkpoint(self, path):
         path = Path(path)
@@ -369,6 +374,19 @@ class Trainer:
         load_module(self.text_encoder, path / "text_encoder")
         state_f = path / "state.pt"
         if state_f.exists():
-            st = torch.load(state_f, map_location=self.device, weights_only=False)
+            st = torch.load(state_f, map_location="cpu", weights_only=False)
             self.global_step = st["global_step"]
             self.optimizer.load_state_dict(st["optimizer"])
+            # restore RNG streams so resumed training is bit-faithful
+            if "torch_rng" in st:
+                torch.set_rng_state(st["torch_rng"].cpu().to(torch.uint8))
+            if "numpy_rng" in st:
+                np.random.set_state(st["numpy_rng"])
+            if "python_rng" in st:
+                random.setstate(st["python_rng"])
+            if "cuda_rng" in st and torch.cuda.is_available():
+                try:
+                    torch.cuda.set_rng_state_all(
+                        [t.cpu().to(torch.uint8) for t in st["cuda_rng"]])
+                except Exception as e:
+                    print(f"cuda rng restore skipped: {e}")

@@ -186,6 +186,21 @@ def main():
         if not args.dontsave:
             torch.save({"entropy": ent, "jpeg": jpg, "tv": tv},
                        out_dir / "complexity.pth")
+            try:
+                import matplotlib
+                matplotlib.use("Agg")
+                import matplotlib.pyplot as plt
+                for name, vals in (("entropy", ent), ("jpegsize", jpg),
+                                   ("tv", tv)):
+                    plt.figure(figsize=(4, 4))
+                    plt.scatter(vals, sims_np, s=8, alpha=0.6)
+                    plt.xlabel(name)
+                    plt.ylabel("top-1 similarity")
+                    plt.tight_layout()
+                    plt.savefig(out_dir / f"simplicityscatter_{name}.png", dpi=110)
+                    plt.close()
+            except Exception as e:
+                print(f"scatter plots skipped: {e}")
 
     # duplication analysis (reference :562-583): are the up-weighted
     # (duplicated) training images matched more often / more strongly?

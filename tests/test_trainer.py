@@ -159,3 +159,30 @@ def test_trainsubset(tmp_path):
     assert len(tr.dataset) == 4  # 50% of 8
     b = next(iter(tr.dataloader))
     assert torch.isfinite(tr.train_step(b))
+
+
+def test_resume_bit_faithful(tmp_path):
+    """4 straight steps == 2 steps -> checkpoint -> resume -> 2 steps
+    (same data, restored RNG streams)."""
+    cfg = tiny_cfg(tmp_path, seed=11)
+    tr_a = Trainer(cfg)
+    batch = next(iter(tr_a.dataloader))
+    for _ in range(4):
+        tr_a.train_step(batch)
+
+    tr_b = Trainer(tiny_cfg(tmp_path, seed=11))
+    b2 = next(iter(tr_b.dataloader))
+    assert torch.equal(batch["pixel_values"], b2["pixel_values"])
+    for _ in range(2):
+        tr_b.train_step(b2)
+    tr_b.save_checkpoint(tmp_path / "ck")
+
+    tr_c = Trainer(tiny_cfg(tmp_path, seed=99))  # different seed on purpose
+    tr_c.load_checkpoint(tmp_path / "ck")
+    assert tr_c.global_step == 2
+    for _ in range(2):
+        tr_c.train_step(b2)
+
+    assert torch.allclose(tr_a.optimizer.flat_param, tr_c.optimizer.flat_param,
+                          atol=1e-6), \
+        (tr_a.optimizer.flat_param - tr_c.optimizer.flat_param).abs().max()
