@@ -23,6 +23,7 @@ SOURCES = [
     HIP_DIR / "norms_nhwc.hip",
     HIP_DIR / "elementwise.hip",
     HIP_DIR / "attention.hip",
+    HIP_DIR / "conv_nhwc.hip",
 ]
 
 

@@ -1,0 +1,133 @@
+// Implicit-GEMM convolution forward (NHWC, bf16, MFMA) for MI355X.
+//
+// SURVEY.md §7 "hard parts" item 2: hand-written implicit-GEMM conv for
+// the SD UNet/VAE shapes. GEMM view of out[n,p,q,k] = sum_{r,s,c}
+// x[n, p*stride+r-pad, q*stride+s-pad, c] * w[k,r,s,c]:
+//   M = N*P*Q output pixels, N = K output channels, K-dim = R*S*C.
+// Same fragment discipline as attention.hip (v_mfma_f32_16x16x32_bf16,
+// "8 contiguous contraction elements at row lane&15"):
+//   * A (pixels)  : on-the-fly im2col — a 32-wide rsc slice with C%32==0
+//     never straddles an (r,s) tap, so each pixel's slice is 32
+//     CONTIGUOUS channels of one input row; staged to LDS (pitch 40).
+//   * B (weights) : [K,R,S,C] channels_last weight memory is contiguous
+//     in rsc for fixed k — fragments read DIRECTLY from global (the
+//     weight tile stays hot in L2 across the many pixel blocks).
+// Block = 4 waves = 64 pixels x 64 out-channels; grid (pixels/64, K/64, 1).
+//
+// Constraints (dispatcher falls back to MIOpen otherwise): bf16,
+// channels_last, C % 32 == 0, K % 64 == 0, kernel 3x3(pad 1) or
+// 1x1(pad 0), stride 1 or 2. Opt-in via DCR_NATIVE_CONV=1 (fwd only —
+// autograd uses MIOpen for backward).
+
+#include "dcr_common.h"
+
+namespace dcr_conv {
+
+typedef __attribute__((ext_vector_type(8))) short bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4_t;
+
+using bf16_t = __hip_bfloat16;
+
+#define CPITCH 40  // LDS row pitch (bf16) for the 32-wide A tile: 10 dwords
+                   // -> 16-lane ds_read_b128 groups land on 16 distinct banks
+
+__global__ __launch_bounds__(256)
+void conv_nhwc_fwd_kernel(const bf16_t* __restrict__ x, const bf16_t* __restrict__ w,
+                          const float* __restrict__ bias, bf16_t* __restrict__ y,
+                          int Nb, int Hin, int Win, int C, int K, int P, int Q,
+                          int R, int S, int stride, int pad) {
+  __shared__ short sA[64 * CPITCH];
+
+  const int m0 = blockIdx.x * 64;          // first output pixel of this block
+  const int k0 = blockIdx.y * 64;          // first output channel
+  const long NPQ = (long)Nb * P * Q;
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int l16 = lane & 15;
+  const int kgrp = lane >> 4;
+  const int wrow0 = wid * 16;              // wave's 16 pixels
+
+  f32x4_t acc[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) acc[i] = {0.f, 0.f, 0.f, 0.f};
+
+  // this thread's staging pixel: 4 threads per pixel row, 8 bf16 each
+  const int st_pix = threadIdx.x >> 2;     // 0..63
+  const int st_c8 = (threadIdx.x & 3) * 8; // 0..24 within the 32-slice
+  long st_m = m0 + st_pix;
+  int st_n = 0, st_p = 0, st_q = 0;
+  if (st_m < NPQ) {
+    st_n = (int)(st_m / (P * Q));
+    int pq = (int)(st_m % (P * Q));
+    st_p = pq / Q;
+    st_q = pq % Q;
+  }
+
+  const int rsc_total = R * S * C;
+  for (int rsc0 = 0; rsc0 < rsc_total; rsc0 += 32) {
+    // decode the tap for this 32-slice (C % 32 == 0: tap is slice-uniform)
+    const int tap = rsc0 / C;
+    const int r = tap / S;
+    const int s = tap % S;
+    const int c0 = rsc0 - tap * C;
+
+    __syncthreads();
+    {
+      // stage A: pixel st_pix, channels [c0+st_c8, +8) of tap (r, s)
+      uint4 v = make_uint4(0, 0, 0, 0);
+      const int hi = st_p * stride + r - pad;
+      const int wi = st_q * stride + s - pad;
+      if (st_m < NPQ && hi >= 0 && hi < Hin && wi >= 0 && wi < Win) {
+        const bf16_t* src = x + (((long)st_n * Hin + hi) * Win + wi) * C + c0 + st_c8;
+        v = *reinterpret_cast<const uint4*>(src);
+      }
+      *reinterpret_cast<uint4*>(sA + st_pix * CPITCH + st_c8) = v;
+    }
+    __syncthreads();
+
+    // A fragment: pixel row (wrow0 + l16), k-elems kgrp*8..+8 of the slice
+    bf16x8 af = *reinterpret_cast<const bf16x8*>(
+        sA + (wrow0 + l16) * CPITCH + kgrp * 8);
+    // B fragments straight from global: w[k0 + ns*16 + l16][rsc0 + kgrp*8]
+#pragma unroll
+    for (int ns = 0; ns < 4; ++ns) {
+      const bf16_t* wp = w + (long)(k0 + ns * 16 + l16) * rsc_total + rsc0 + kgrp * 8;
+      bf16x8 bf = *reinterpret_cast<const bf16x8*>(wp);
+      acc[ns] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf, acc[ns], 0, 0, 0);
+    }
+  }
+
+  // epilogue: += bias, store [pixel][k] (NHWC output: k contiguous)
+#pragma unroll
+  for (int rr = 0; rr < 4; ++rr) {
+    const long m = m0 + wrow0 + kgrp * 4 + rr;
+    if (m >= NPQ) continue;
+#pragma unroll
+    for (int ns = 0; ns < 4; ++ns) {
+      const int k = k0 + ns * 16 + l16;
+      float v = acc[ns][rr] + (bias ? bias[k] : 0.f);
+      y[m * K + k] = __float2bfloat16(v);
+    }
+  }
+}
+
+}  // namespace dcr_conv
+
+#include "dcr_launchers.h"
+
+namespace dcr {
+
+void conv_nhwc_fwd_launch(const void* x, const void* w, const float* bias,
+                          void* y, int Nb, int Hin, int Win, int C, int K,
+                          int P, int Q, int R, int S, int stride, int pad,
+                          hipStream_t st) {
+  long NPQ = (long)Nb * P * Q;
+  dim3 grid((unsigned)((NPQ + 63) / 64), (unsigned)(K / 64)), block(256);
+  hipLaunchKernelGGL(dcr_conv::conv_nhwc_fwd_kernel, grid, block, 0, st,
+                     (const dcr_conv::bf16_t*)x, (const dcr_conv::bf16_t*)w,
+                     bias, (dcr_conv::bf16_t*)y, Nb, Hin, Win, C, K, P, Q, R,
+                     S, stride, pad);
+}
+
+}  // namespace dcr

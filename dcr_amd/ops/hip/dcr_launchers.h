@@ -66,4 +66,10 @@ void attn_bwd_launch(const void* q, const void* k, const void* v,
                      hipStream_t s);
 void mfma_probe_launch(const void* A, const void* B, float* C, hipStream_t s);
 
+// conv_nhwc.hip (implicit-GEMM conv fwd, opt-in)
+void conv_nhwc_fwd_launch(const void* x, const void* w, const float* bias,
+                          void* y, int Nb, int Hin, int Win, int C, int K,
+                          int P, int Q, int R, int S, int stride, int pad,
+                          hipStream_t st);
+
 }  // namespace dcr

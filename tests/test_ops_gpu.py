@@ -4,6 +4,8 @@ Every kernel is compared against the fp32 reference on random tensors, in
 fp32 (tight tolerance) and bf16 (bf16 tolerance), across the SD-2.1
 launch shapes (incl. odd sizes: 77-token rows, HW=64, non-wave-multiple
 channel counts)."""
+import os
+
 import pytest
 import torch
 import torch.nn.functional as F
@@ -358,3 +360,29 @@ def test_dpm_step_gpu_matches_cpu(ext):
     err = (x_g.float().cpu() - x_c.float()).abs().max()
     scale = x_c.abs().max() + 1e-6
     assert err / scale < 0.08, (err, scale)  # bf16 accumulation over 8 steps
+
+
+# ------------------------------------------------- implicit-GEMM conv (opt-in)
+@pytest.mark.skipif(os.environ.get("DCR_NATIVE_CONV") != "1",
+                    reason="native conv is opt-in (DCR_NATIVE_CONV=1)")
+@pytest.mark.parametrize("shape", [
+    (2, 320, 32, 32, 320, 3, 1),    # ResNet conv, stride 1
+    (2, 320, 32, 32, 640, 1, 1),    # 1x1 shortcut
+    (2, 640, 16, 16, 640, 3, 2),    # downsample
+    (1, 128, 64, 64, 128, 3, 1),    # VAE early
+    (2, 1280, 8, 8, 1280, 3, 1),
+])
+def test_conv_nhwc_fwd(ext, shape):
+    N, C, H, W, K, R, stride = shape
+    pad = 1 if R == 3 else 0
+    torch.manual_seed(0)
+    x = torch.randn(N, C, H, W, device="cuda").to(torch.bfloat16) \
+        .to(memory_format=torch.channels_last)
+    w = (torch.randn(K, C, R, R, device="cuda") * 0.05).to(torch.bfloat16) \
+        .to(memory_format=torch.channels_last)
+    b = torch.randn(K, device="cuda")
+    y = ext.conv2d_nhwc_fwd(x, w, b, stride, pad)
+    ref = torch.nn.functional.conv2d(x.float(), w.float(), b, stride=stride,
+                                     padding=pad)
+    assert y.is_contiguous(memory_format=torch.channels_last)
+    _close(y, ref, 2e-2)
