@@ -207,3 +207,18 @@ def test_ipr_realism_score():
     feats = ipr.extract_features(ref[:1])
     r = ipr.realism(feats[0])
     assert np.isfinite(r) and r > 0
+
+
+def test_multiscale_features():
+    """multiscale eval: 1x + 1/sqrt(2) + 1/2 scales, summed and renormed
+    (reference utils_ret.py:676-698)."""
+    from torch.utils.data import DataLoader
+    torch.manual_seed(0)
+    model = torch.nn.Sequential(torch.nn.AdaptiveAvgPool2d(1), torch.nn.Flatten())
+    imgs = torch.rand(6, 3, 32, 32)
+    ds = [(imgs[i], i) for i in range(6)]
+    feats = extract_features(model, DataLoader(ds, batch_size=3),
+                             torch.device("cpu"), multiscale=True)
+    assert feats.shape == (6, 3)
+    norms = feats.norm(dim=-1)
+    assert torch.allclose(norms, torch.ones(6), atol=1e-4)  # renormalized
