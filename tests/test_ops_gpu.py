@@ -363,8 +363,6 @@ def test_dpm_step_gpu_matches_cpu(ext):
 
 
 # ------------------------------------------------- implicit-GEMM conv (opt-in)
-@pytest.mark.skipif(os.environ.get("DCR_NATIVE_CONV") != "1",
-                    reason="native conv is opt-in (DCR_NATIVE_CONV=1)")
 @pytest.mark.parametrize("shape", [
     (2, 320, 32, 32, 320, 3, 1),    # ResNet conv, stride 1
     (2, 320, 32, 32, 640, 1, 1),    # 1x1 shortcut
