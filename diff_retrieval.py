@@ -25,11 +25,11 @@ from torch.utils.data.distributed import DistributedSampler
 
 from dcr_amd.data import SynthDataset, EvalTransform
 from dcr_amd.parallel import barrier, init_distributed_mode, is_main_process
-from dcr_amd.retrieval import (extract_features, gen_clipscore, glcm_entropy,
-                               jpeg_size, l2_normalize, load_clip, load_dino,
-                               load_sscd, pearson, sim_matrix,
-                               similarity_histogram, top_matches, topk_stats,
-                               tv_loss)
+from dcr_amd.retrieval import (einsum_in_chunks, extract_features,
+                               gen_clipscore, glcm_entropy, jpeg_size,
+                               l2_normalize, load_clip, load_dino, load_sscd,
+                               pearson, sim_matrix, similarity_histogram,
+                               top_matches, topk_stats, tv_loss)
 from dcr_amd.utils import Tracker
 
 
@@ -83,6 +83,26 @@ def build_backbone(args, device):
     return load_sscd(args.pt_style, device=device)  # sscd variants
 
 
+@torch.no_grad()
+def _patch_features(model, loader, device, args):
+    """[N, D, P] L2-normalized per-patch descriptors: ViT token features
+    (get_intermediate_layers) or the CNN's final spatial map."""
+    import torch.nn.functional as F_
+    feats = []
+    for batch in loader:
+        imgs = batch[0].to(device)
+        if hasattr(model, "get_intermediate_layers"):
+            tok = model.get_intermediate_layers(imgs, n=args.layer)[0][:, 1:]
+            f = tok.transpose(1, 2)                     # [N, D, P]
+        elif hasattr(model, "backbone"):                # SSCDModel
+            fm = model.backbone.forward_features(imgs)  # [N, D, h, w]
+            f = fm.flatten(2)
+        else:
+            f = model(imgs)[..., None]
+        feats.append(F_.normalize(f, dim=1).cpu())
+    return torch.cat(feats)
+
+
 def main():
     args = parse_args()
     rank, world, local = init_distributed_mode()
@@ -122,9 +142,19 @@ def main():
     query_f = l2_normalize(query_f)
     val_f = l2_normalize(val_f)
 
-    # HOT LOOP 2: dense GEMM similarity (rocBLAS)
-    sim = sim_matrix(val_f, query_f).t()        # [n_gen, n_train]
-    sim_tt = sim_matrix(val_f, val_f)           # [n_train, n_train]
+    # HOT LOOP 2: similarity
+    if args.similarity_metric == "splitloss":
+        # patch-wise max similarity (reference :393-400, einsum :643-662):
+        # features here are [N, D, P] per-patch descriptors
+        qp_ = _patch_features(model, loader(query_ds), device, args)
+        vp_ = _patch_features(model, loader(val_ds), device, args)
+        sim = einsum_in_chunks(qp_, vp_, chunk=max(1, len(query_ds) //
+                                                   max(1, args.einsum_chunks)))
+        sim_tt = einsum_in_chunks(vp_, vp_, chunk=max(1, len(val_ds) //
+                                                      max(1, args.einsum_chunks)))
+    else:
+        sim = sim_matrix(val_f, query_f).t()    # [n_gen, n_train] (rocBLAS)
+        sim_tt = sim_matrix(val_f, val_f)       # [n_train, n_train]
 
     if not args.dontsave:
         torch.save(sim.cpu(), out_dir / "similarity.pth")

@@ -194,3 +194,28 @@ def test_convert_diffusers_checkpoint_roundtrip(tmp_path):
         assert torch.equal(v, pipe2.unet.state_dict()[k]), k
     for k, v in pipe.text_encoder.state_dict().items():
         assert torch.equal(v, pipe2.text_encoder.state_dict()[k]), k
+
+
+@pytest.mark.timeout(600)
+def test_retrieval_cli_splitloss(tmp_path):
+    """--similarity_metric splitloss (patch-wise einsum path)."""
+    rng = np.random.default_rng(0)
+    qdir, vdir = tmp_path / "gens", tmp_path / "train"
+    qdir.mkdir(); vdir.mkdir()
+    for i in range(3):
+        Image.fromarray(rng.integers(0, 255, (64, 64, 3)).astype(np.uint8)) \
+            .save(qdir / f"{i}.png")
+    for i in range(4):
+        Image.fromarray(rng.integers(0, 255, (64, 64, 3)).astype(np.uint8)) \
+            .save(vdir / f"{i}.png")
+    out = tmp_path / "s"
+    r = subprocess.run(
+        [sys.executable, "diff_retrieval.py", "--query_dir", str(qdir),
+         "--val_dir", str(vdir), "--pt_style", "sscd", "-b", "4", "-j", "0",
+         "--imsize", "64", "-ssp", str(out), "--skip_fid", "--noeval",
+         "--similarity_metric", "splitloss", "--einsum_chunks", "2"],
+        capture_output=True, text=True, cwd=str(Path(__file__).parent.parent),
+        timeout=570)
+    assert r.returncode == 0, r.stderr[-2000:]
+    sim = torch.load(out / "similarity.pth")
+    assert sim.shape == (3, 4)
