@@ -378,8 +378,35 @@ at::Tensor conv2d_nhwc_fwd(at::Tensor x, at::Tensor w,
   return y;
 }
 
+at::Tensor conv2d_nhwc_fwd_v2(at::Tensor x, at::Tensor w,
+                              c10::optional<at::Tensor> bias, int64_t stride,
+                              int64_t pad) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast) &&
+              w.is_contiguous(at::MemoryFormat::ChannelsLast));
+  const int64_t Nb = x.size(0), C = x.size(1), Hin = x.size(2), Win = x.size(3);
+  const int64_t K = w.size(0), R = w.size(2), S = w.size(3);
+  TORCH_CHECK(w.size(1) == C && C % 32 == 0 && K % 64 == 0);
+  TORCH_CHECK((R == 3 && S == 3) || (R == 1 && S == 1));
+  const int64_t P = (Hin + 2 * pad - R) / stride + 1;
+  const int64_t Q = (Win + 2 * pad - S) / stride + 1;
+  auto y = at::empty({Nb, K, P, Q},
+                     x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  at::Tensor bf;
+  const float* bp = nullptr;
+  if (bias.has_value()) {
+    bf = bias->to(at::kFloat).contiguous();
+    bp = bf.data_ptr<float>();
+  }
+  conv_nhwc_fwd_v2_launch(x.data_ptr(), w.data_ptr(), bp, y.data_ptr(), (int)Nb,
+                          (int)Hin, (int)Win, (int)C, (int)K, (int)P, (int)Q,
+                          (int)R, (int)S, (int)stride, (int)pad, cur_stream());
+  return y;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("conv2d_nhwc_fwd", &conv2d_nhwc_fwd);
+  mod.def("conv2d_nhwc_fwd_v2", &conv2d_nhwc_fwd_v2);
   mod.def("attn_fwd", &attn_fwd);
   mod.def("attn_bwd", &attn_bwd);
   mod.def("mfma_probe", &mfma_probe);

@@ -131,3 +131,136 @@ void conv_nhwc_fwd_launch(const void* x, const void* w, const float* bias,
 }
 
 }  // namespace dcr
+
+// ===========================================================================
+// v2: 128x128 tile, LDS-staged A (im2col gather) and B (weights), BK=32.
+// 4 waves as 2x2, each computing a 64x64 sub-tile (acc[4][4] f32x4).
+// Still single-buffered (2 barriers per K-step) — the guide's "step-2"
+// structure; double-buffering + split-K are the round-2 follow-ups.
+// ===========================================================================
+namespace dcr_conv {
+
+__global__ __launch_bounds__(256)
+void conv_nhwc_fwd_v2_kernel(const bf16_t* __restrict__ x, const bf16_t* __restrict__ w,
+                             const float* __restrict__ bias, bf16_t* __restrict__ y,
+                             int Nb, int Hin, int Win, int C, int K, int P, int Q,
+                             int R, int S, int stride, int pad) {
+  __shared__ short sA[128 * CPITCH];
+  __shared__ short sB[128 * CPITCH];
+
+  const long m0 = (long)blockIdx.x * 128;
+  const int k0 = blockIdx.y * 128;
+  const long NPQ = (long)Nb * P * Q;
+  const int rsc_total = R * S * C;
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int l16 = lane & 15;
+  const int kgrp = lane >> 4;
+  const int wr = (wid >> 1) * 64;          // wave pixel base within tile
+  const int wc = (wid & 1) * 64;           // wave k base within tile
+
+  f32x4_t acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  // staging assignments: 2 threads per row, 16 bf16 (32 B) each
+  const int st_row = threadIdx.x >> 1;     // 0..127
+  const int st_c16 = (threadIdx.x & 1) * 16;
+  long st_m = m0 + st_row;
+  int st_n = 0, st_p = 0, st_q = 0;
+  if (st_m < NPQ) {
+    st_n = (int)(st_m / (P * Q));
+    int pq = (int)(st_m % (P * Q));
+    st_p = pq / Q;
+    st_q = pq % Q;
+  }
+  const long wrow = (long)(k0 + st_row) * rsc_total;  // B source row base
+
+  for (int rsc0 = 0; rsc0 < rsc_total; rsc0 += 32) {
+    const int tap = rsc0 / C;
+    const int r = tap / S;
+    const int s = tap % S;
+    const int c0 = rsc0 - tap * C;
+
+    __syncthreads();
+    {
+      uint4 a0 = make_uint4(0, 0, 0, 0), a1 = a0;
+      const int hi = st_p * stride + r - pad;
+      const int wi = st_q * stride + s - pad;
+      if (st_m < NPQ && hi >= 0 && hi < Hin && wi >= 0 && wi < Win) {
+        const bf16_t* src = x + (((long)st_n * Hin + hi) * Win + wi) * C + c0 + st_c16;
+        const uint4* p4 = reinterpret_cast<const uint4*>(src);
+        a0 = p4[0];
+        a1 = p4[1];
+      }
+      uint4* d = reinterpret_cast<uint4*>(sA + st_row * CPITCH + st_c16);
+      d[0] = a0;
+      d[1] = a1;
+
+      uint4 b0 = make_uint4(0, 0, 0, 0), b1 = b0;
+      if (k0 + st_row < K) {
+        const uint4* p4 = reinterpret_cast<const uint4*>(w + wrow + rsc0 + st_c16);
+        b0 = p4[0];
+        b1 = p4[1];
+      }
+      uint4* db = reinterpret_cast<uint4*>(sB + st_row * CPITCH + st_c16);
+      db[0] = b0;
+      db[1] = b1;
+    }
+    __syncthreads();
+
+    bf16x8 af[4], bf[4];
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+      af[i] = *reinterpret_cast<const bf16x8*>(
+          sA + (wr + i * 16 + l16) * CPITCH + kgrp * 8);
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+      bf[j] = *reinterpret_cast<const bf16x8*>(
+          sB + (wc + j * 16 + l16) * CPITCH + kgrp * 8);
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[i], bf[j],
+                                                            acc[i][j], 0, 0, 0);
+  }
+
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+#pragma unroll
+    for (int rr = 0; rr < 4; ++rr) {
+      const long m = m0 + wr + i * 16 + kgrp * 4 + rr;
+      if (m >= NPQ) continue;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const int k = k0 + wc + j * 16 + l16;
+        if (k >= K) continue;
+        float v = acc[i][j][rr] + (bias ? bias[k] : 0.f);
+        y[m * K + k] = __float2bfloat16(v);
+      }
+    }
+  }
+}
+
+}  // namespace dcr_conv
+
+namespace dcr {
+
+void conv_nhwc_fwd_v2_launch(const void* x, const void* w, const float* bias,
+                             void* y, int Nb, int Hin, int Win, int C, int K,
+                             int P, int Q, int R, int S, int stride, int pad,
+                             hipStream_t st) {
+  long NPQ = (long)Nb * P * Q;
+  dim3 grid((unsigned)((NPQ + 127) / 128), (unsigned)((K + 127) / 128)),
+      block(256);
+  hipLaunchKernelGGL(dcr_conv::conv_nhwc_fwd_v2_kernel, grid, block, 0, st,
+                     (const dcr_conv::bf16_t*)x, (const dcr_conv::bf16_t*)w,
+                     bias, (dcr_conv::bf16_t*)y, Nb, Hin, Win, C, K, P, Q, R,
+                     S, stride, pad);
+}
+
+}  // namespace dcr

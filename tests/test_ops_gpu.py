@@ -370,7 +370,8 @@ def test_dpm_step_gpu_matches_cpu(ext):
     (1, 128, 64, 64, 128, 3, 1),    # VAE early
     (2, 1280, 8, 8, 1280, 3, 1),
 ])
-def test_conv_nhwc_fwd(ext, shape):
+@pytest.mark.parametrize("version", ["v1", "v2"])
+def test_conv_nhwc_fwd(ext, shape, version):
     N, C, H, W, K, R, stride = shape
     pad = 1 if R == 3 else 0
     torch.manual_seed(0)
@@ -379,7 +380,8 @@ def test_conv_nhwc_fwd(ext, shape):
     w = (torch.randn(K, C, R, R, device="cuda") * 0.05).to(torch.bfloat16) \
         .to(memory_format=torch.channels_last)
     b = torch.randn(K, device="cuda")
-    y = ext.conv2d_nhwc_fwd(x, w, b, stride, pad)
+    fn = ext.conv2d_nhwc_fwd if version == "v1" else ext.conv2d_nhwc_fwd_v2
+    y = fn(x, w, b, stride, pad)
     ref = torch.nn.functional.conv2d(x.float(), w.float(), b, stride=stride,
                                      padding=pad)
     assert y.is_contiguous(memory_format=torch.channels_last)
