@@ -398,9 +398,25 @@ at::Tensor conv2d_nhwc_fwd_v2(at::Tensor x, at::Tensor w,
     bf = bias->to(at::kFloat).contiguous();
     bp = bf.data_ptr<float>();
   }
-  conv_nhwc_fwd_v2_launch(x.data_ptr(), w.data_ptr(), bp, y.data_ptr(), (int)Nb,
-                          (int)Hin, (int)Win, (int)C, (int)K, (int)P, (int)Q,
-                          (int)R, (int)S, (int)stride, (int)pad, cur_stream());
+  // split-K when the (pixels/128) x (K/128) grid starves the 256 CUs
+  const int64_t NPQ = Nb * P * Q;
+  int64_t blocks = ((NPQ + 127) / 128) * ((K + 127) / 128);
+  int64_t nsteps = C * R * S / 32;
+  int splitz = 1;
+  if (blocks < 384) {
+    splitz = (int)std::min<int64_t>({(384 + blocks - 1) / blocks, nsteps, 16});
+    if (splitz < 1) splitz = 1;
+  }
+  at::Tensor ws;
+  float* wsp = nullptr;
+  if (splitz > 1) {
+    ws = at::zeros({NPQ * K}, x.options().dtype(at::kFloat));
+    wsp = ws.data_ptr<float>();
+  }
+  conv_nhwc_fwd_v2_launch(x.data_ptr(), w.data_ptr(), bp, y.data_ptr(), wsp,
+                          splitz, (int)Nb, (int)Hin, (int)Win, (int)C, (int)K,
+                          (int)P, (int)Q, (int)R, (int)S, (int)stride,
+                          (int)pad, cur_stream());
   return y;
 }
 

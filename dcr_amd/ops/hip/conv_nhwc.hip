@@ -140,9 +140,13 @@ void conv_nhwc_fwd_launch(const void* x, const void* w, const float* bias,
 // ===========================================================================
 namespace dcr_conv {
 
+// splitz > 1: blockIdx.z covers a slice of the rsc steps; fp32 partials
+// are atomically accumulated into ws[NPQ*K] and a finalize kernel adds
+// bias + casts (grid starvation fix for the 8x8/16x16 shapes).
 __global__ __launch_bounds__(256)
 void conv_nhwc_fwd_v2_kernel(const bf16_t* __restrict__ x, const bf16_t* __restrict__ w,
                              const float* __restrict__ bias, bf16_t* __restrict__ y,
+                             float* __restrict__ ws, int splitz,
                              int Nb, int Hin, int Win, int C, int K, int P, int Q,
                              int R, int S, int stride, int pad) {
   __shared__ short sA[128 * CPITCH];
@@ -179,7 +183,13 @@ void conv_nhwc_fwd_v2_kernel(const bf16_t* __restrict__ x, const bf16_t* __restr
   }
   const long wrow = (long)(k0 + st_row) * rsc_total;  // B source row base
 
-  for (int rsc0 = 0; rsc0 < rsc_total; rsc0 += 32) {
+  const int nsteps = rsc_total / 32;
+  const int spz = (nsteps + splitz - 1) / splitz;
+  const int step0 = blockIdx.z * spz;
+  const int step1 = min(nsteps, step0 + spz);
+
+  for (int step = step0; step < step1; ++step) {
+    const int rsc0 = step * 32;
     const int tap = rsc0 / C;
     const int r = tap / S;
     const int s = tap % S;
@@ -239,10 +249,25 @@ void conv_nhwc_fwd_v2_kernel(const bf16_t* __restrict__ x, const bf16_t* __restr
       for (int j = 0; j < 4; ++j) {
         const int k = k0 + wc + j * 16 + l16;
         if (k >= K) continue;
-        float v = acc[i][j][rr] + (bias ? bias[k] : 0.f);
-        y[m * K + k] = __float2bfloat16(v);
+        if (splitz > 1) {
+          atomicAdd(&ws[m * K + k], acc[i][j][rr]);
+        } else {
+          float v = acc[i][j][rr] + (bias ? bias[k] : 0.f);
+          y[m * K + k] = __float2bfloat16(v);
+        }
       }
     }
+  }
+}
+
+__global__ void conv_splitk_finalize_kernel(const float* __restrict__ ws,
+                                            const float* __restrict__ bias,
+                                            bf16_t* __restrict__ y, long total,
+                                            int K) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    float v = ws[i] + (bias ? bias[i % K] : 0.f);
+    y[i] = __float2bfloat16(v);
   }
 }
 
@@ -251,16 +276,25 @@ void conv_nhwc_fwd_v2_kernel(const bf16_t* __restrict__ x, const bf16_t* __restr
 namespace dcr {
 
 void conv_nhwc_fwd_v2_launch(const void* x, const void* w, const float* bias,
-                             void* y, int Nb, int Hin, int Win, int C, int K,
-                             int P, int Q, int R, int S, int stride, int pad,
-                             hipStream_t st) {
+                             void* y, float* ws, int splitz, int Nb, int Hin,
+                             int Win, int C, int K, int P, int Q, int R, int S,
+                             int stride, int pad, hipStream_t st) {
   long NPQ = (long)Nb * P * Q;
-  dim3 grid((unsigned)((NPQ + 127) / 128), (unsigned)((K + 127) / 128)),
+  dim3 grid((unsigned)((NPQ + 127) / 128), (unsigned)((K + 127) / 128),
+            (unsigned)splitz),
       block(256);
   hipLaunchKernelGGL(dcr_conv::conv_nhwc_fwd_v2_kernel, grid, block, 0, st,
                      (const dcr_conv::bf16_t*)x, (const dcr_conv::bf16_t*)w,
-                     bias, (dcr_conv::bf16_t*)y, Nb, Hin, Win, C, K, P, Q, R,
-                     S, stride, pad);
+                     bias, (dcr_conv::bf16_t*)y, ws, splitz, Nb, Hin, Win, C,
+                     K, P, Q, R, S, stride, pad);
+  if (splitz > 1) {
+    long total = NPQ * K;
+    long b = (total / 4 + 255) / 256;
+    if (b > 8192) b = 8192;
+    hipLaunchKernelGGL(dcr_conv::conv_splitk_finalize_kernel,
+                       dim3((unsigned)b), dim3(256), 0, st, ws, bias,
+                       (dcr_conv::bf16_t*)y, total, K);
+  }
 }
 
 }  // namespace dcr

@@ -72,8 +72,8 @@ void conv_nhwc_fwd_launch(const void* x, const void* w, const float* bias,
                           int P, int Q, int R, int S, int stride, int pad,
                           hipStream_t st);
 void conv_nhwc_fwd_v2_launch(const void* x, const void* w, const float* bias,
-                             void* y, int Nb, int Hin, int Win, int C, int K,
-                             int P, int Q, int R, int S, int stride, int pad,
-                             hipStream_t st);
+                             void* y, float* ws, int splitz, int Nb, int Hin,
+                             int Win, int C, int K, int P, int Q, int R, int S,
+                             int stride, int pad, hipStream_t st);
 
 }  // namespace dcr
