@@ -401,7 +401,7 @@ at::Tensor conv2d_nhwc_fwd_v2(at::Tensor x, at::Tensor w,
   // split-K when the (pixels/128) x (K/128) grid starves the 256 CUs
   const int64_t NPQ = Nb * P * Q;
   int64_t blocks = ((NPQ + 127) / 128) * ((K + 127) / 128);
-  int64_t nsteps = C * R * S / 32;
+  int64_t nsteps = C * R * S / ((C % 64 == 0) ? 64 : 32);
   int splitz = 1;
   if (blocks < 384) {
     splitz = (int)std::min<int64_t>({(384 + blocks - 1) / blocks, nsteps, 16});

@@ -143,14 +143,16 @@ namespace dcr_conv {
 // splitz > 1: blockIdx.z covers a slice of the rsc steps; fp32 partials
 // are atomically accumulated into ws[NPQ*K] and a finalize kernel adds
 // bias + casts (grid starvation fix for the 8x8/16x16 shapes).
+template <int BK>
 __global__ __launch_bounds__(256)
 void conv_nhwc_fwd_v2_kernel(const bf16_t* __restrict__ x, const bf16_t* __restrict__ w,
                              const float* __restrict__ bias, bf16_t* __restrict__ y,
                              float* __restrict__ ws, int splitz,
                              int Nb, int Hin, int Win, int C, int K, int P, int Q,
                              int R, int S, int stride, int pad) {
-  __shared__ short sA[128 * CPITCH];
-  __shared__ short sB[128 * CPITCH];
+  constexpr int PITCH2 = BK + 8;           // 16-lane b128 groups: 16 banks
+  __shared__ short sA[128 * PITCH2];
+  __shared__ short sB[128 * PITCH2];
 
   const long m0 = (long)blockIdx.x * 128;
   const int k0 = blockIdx.y * 128;
@@ -170,9 +172,9 @@ void conv_nhwc_fwd_v2_kernel(const bf16_t* __restrict__ x, const bf16_t* __restr
 #pragma unroll
     for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
-  // staging assignments: 2 threads per row, 16 bf16 (32 B) each
+  // staging: 2 threads per row, each covers BK/2 contiguous bf16
   const int st_row = threadIdx.x >> 1;     // 0..127
-  const int st_c16 = (threadIdx.x & 1) * 16;
+  const int st_c16 = (threadIdx.x & 1) * (BK / 2);
   long st_m = m0 + st_row;
   int st_n = 0, st_p = 0, st_q = 0;
   if (st_m < NPQ) {
@@ -183,60 +185,67 @@ void conv_nhwc_fwd_v2_kernel(const bf16_t* __restrict__ x, const bf16_t* __restr
   }
   const long wrow = (long)(k0 + st_row) * rsc_total;  // B source row base
 
-  const int nsteps = rsc_total / 32;
+  const int nsteps = rsc_total / BK;
   const int spz = (nsteps + splitz - 1) / splitz;
   const int step0 = blockIdx.z * spz;
   const int step1 = min(nsteps, step0 + spz);
+  constexpr int NCH = BK / 32;             // 16-elem uint4 chunks per half-row
+  constexpr int HALF = BK / 2;
 
   for (int step = step0; step < step1; ++step) {
-    const int rsc0 = step * 32;
-    const int tap = rsc0 / C;
+    const int rsc0 = step * BK;
+    const int tap = rsc0 / C;              // C % BK == 0: tap slice-uniform
     const int r = tap / S;
     const int s = tap % S;
     const int c0 = rsc0 - tap * C;
 
     __syncthreads();
     {
-      uint4 a0 = make_uint4(0, 0, 0, 0), a1 = a0;
+      uint4 av[NCH], bv[NCH];
+#pragma unroll
+      for (int t = 0; t < NCH; ++t) av[t] = make_uint4(0, 0, 0, 0);
+#pragma unroll
+      for (int t = 0; t < NCH; ++t) bv[t] = make_uint4(0, 0, 0, 0);
       const int hi = st_p * stride + r - pad;
       const int wi = st_q * stride + s - pad;
       if (st_m < NPQ && hi >= 0 && hi < Hin && wi >= 0 && wi < Win) {
-        const bf16_t* src = x + (((long)st_n * Hin + hi) * Win + wi) * C + c0 + st_c16;
-        const uint4* p4 = reinterpret_cast<const uint4*>(src);
-        a0 = p4[0];
-        a1 = p4[1];
+        const uint4* p4 = reinterpret_cast<const uint4*>(
+            x + (((long)st_n * Hin + hi) * Win + wi) * C + c0 + st_c16);
+#pragma unroll
+        for (int t = 0; t < NCH; ++t) av[t] = p4[t];
       }
-      uint4* d = reinterpret_cast<uint4*>(sA + st_row * CPITCH + st_c16);
-      d[0] = a0;
-      d[1] = a1;
-
-      uint4 b0 = make_uint4(0, 0, 0, 0), b1 = b0;
       if (k0 + st_row < K) {
         const uint4* p4 = reinterpret_cast<const uint4*>(w + wrow + rsc0 + st_c16);
-        b0 = p4[0];
-        b1 = p4[1];
+#pragma unroll
+        for (int t = 0; t < NCH; ++t) bv[t] = p4[t];
       }
-      uint4* db = reinterpret_cast<uint4*>(sB + st_row * CPITCH + st_c16);
-      db[0] = b0;
-      db[1] = b1;
+      uint4* d = reinterpret_cast<uint4*>(sA + st_row * PITCH2 + st_c16);
+      uint4* db = reinterpret_cast<uint4*>(sB + st_row * PITCH2 + st_c16);
+#pragma unroll
+      for (int t = 0; t < NCH; ++t) d[t] = av[t];
+#pragma unroll
+      for (int t = 0; t < NCH; ++t) db[t] = bv[t];
     }
     __syncthreads();
 
-    bf16x8 af[4], bf[4];
 #pragma unroll
-    for (int i = 0; i < 4; ++i)
-      af[i] = *reinterpret_cast<const bf16x8*>(
-          sA + (wr + i * 16 + l16) * CPITCH + kgrp * 8);
+    for (int kk = 0; kk < BK / 32; ++kk) {
+      bf16x8 af[4], bf[4];
 #pragma unroll
-    for (int j = 0; j < 4; ++j)
-      bf[j] = *reinterpret_cast<const bf16x8*>(
-          sB + (wc + j * 16 + l16) * CPITCH + kgrp * 8);
-#pragma unroll
-    for (int i = 0; i < 4; ++i)
+      for (int i = 0; i < 4; ++i)
+        af[i] = *reinterpret_cast<const bf16x8*>(
+            sA + (wr + i * 16 + l16) * PITCH2 + kk * 32 + kgrp * 8);
 #pragma unroll
       for (int j = 0; j < 4; ++j)
-        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[i], bf[j],
-                                                            acc[i][j], 0, 0, 0);
+        bf[j] = *reinterpret_cast<const bf16x8*>(
+            sB + (wc + j * 16 + l16) * PITCH2 + kk * 32 + kgrp * 8);
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[i], bf[j],
+                                                              acc[i][j], 0, 0, 0);
+    }
   }
 
 #pragma unroll
@@ -283,10 +292,16 @@ void conv_nhwc_fwd_v2_launch(const void* x, const void* w, const float* bias,
   dim3 grid((unsigned)((NPQ + 127) / 128), (unsigned)((K + 127) / 128),
             (unsigned)splitz),
       block(256);
-  hipLaunchKernelGGL(dcr_conv::conv_nhwc_fwd_v2_kernel, grid, block, 0, st,
-                     (const dcr_conv::bf16_t*)x, (const dcr_conv::bf16_t*)w,
-                     bias, (dcr_conv::bf16_t*)y, ws, splitz, Nb, Hin, Win, C,
-                     K, P, Q, R, S, stride, pad);
+  if (C % 64 == 0)
+    hipLaunchKernelGGL((dcr_conv::conv_nhwc_fwd_v2_kernel<64>), grid, block, 0,
+                       st, (const dcr_conv::bf16_t*)x,
+                       (const dcr_conv::bf16_t*)w, bias, (dcr_conv::bf16_t*)y,
+                       ws, splitz, Nb, Hin, Win, C, K, P, Q, R, S, stride, pad);
+  else
+    hipLaunchKernelGGL((dcr_conv::conv_nhwc_fwd_v2_kernel<32>), grid, block, 0,
+                       st, (const dcr_conv::bf16_t*)x,
+                       (const dcr_conv::bf16_t*)w, bias, (dcr_conv::bf16_t*)y,
+                       ws, splitz, Nb, Hin, Win, C, K, P, Q, R, S, stride, pad);
   if (splitz > 1) {
     long total = NPQ * K;
     long b = (total / 4 + 255) / 256;

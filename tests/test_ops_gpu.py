@@ -369,6 +369,7 @@ def test_dpm_step_gpu_matches_cpu(ext):
     (2, 640, 16, 16, 640, 3, 2),    # downsample
     (1, 128, 64, 64, 128, 3, 1),    # VAE early
     (2, 1280, 8, 8, 1280, 3, 1),
+    (2, 96, 16, 16, 64, 3, 1),      # C % 64 != 0 -> BK=32 path
 ])
 @pytest.mark.parametrize("version", ["v1", "v2"])
 def test_conv_nhwc_fwd(ext, shape, version):
