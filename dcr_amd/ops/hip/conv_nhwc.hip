@@ -189,7 +189,7 @@ void conv_nhwc_fwd_v2_kernel(const bf16_t* __restrict__ x, const bf16_t* __restr
   const int spz = (nsteps + splitz - 1) / splitz;
   const int step0 = blockIdx.z * spz;
   const int step1 = min(nsteps, step0 + spz);
-  constexpr int NCH = BK / 32;             // 16-elem uint4 chunks per half-row
+  constexpr int NCH = BK / 16;             // 8-elem uint4 chunks per half-row
   constexpr int HALF = BK / 2;
 
   for (int step = step0; step < step1; ++step) {
