@@ -13,6 +13,7 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
+from ..ops.conv import Conv2d
 from .layers import GroupNormOp
 
 
@@ -34,16 +35,16 @@ class ResnetBlock2D(nn.Module):
         self.output_scale_factor = output_scale_factor
 
         self.norm1 = GroupNormOp(groups, in_channels, eps=eps, fused_silu=True)
-        self.conv1 = nn.Conv2d(in_channels, out_channels, 3, padding=1)
+        self.conv1 = Conv2d(in_channels, out_channels, 3, padding=1)
         if temb_channels is not None:
             self.time_emb_proj = nn.Linear(temb_channels, out_channels)
         else:
             self.time_emb_proj = None
         self.norm2 = GroupNormOp(groups, out_channels, eps=eps, fused_silu=True)
         self.dropout = nn.Dropout(dropout)
-        self.conv2 = nn.Conv2d(out_channels, out_channels, 3, padding=1)
+        self.conv2 = Conv2d(out_channels, out_channels, 3, padding=1)
         if in_channels != out_channels:
-            self.conv_shortcut = nn.Conv2d(in_channels, out_channels, 1)
+            self.conv_shortcut = Conv2d(in_channels, out_channels, 1)
         else:
             self.conv_shortcut = None
 
@@ -64,7 +65,7 @@ class Downsample2D(nn.Module):
     def __init__(self, channels: int, out_channels: Optional[int] = None):
         super().__init__()
         out_channels = out_channels or channels
-        self.conv = nn.Conv2d(channels, out_channels, 3, stride=2, padding=1)
+        self.conv = Conv2d(channels, out_channels, 3, stride=2, padding=1)
 
     def forward(self, x):
         return self.conv(x)
@@ -74,7 +75,7 @@ class Upsample2D(nn.Module):
     def __init__(self, channels: int, out_channels: Optional[int] = None):
         super().__init__()
         out_channels = out_channels or channels
-        self.conv = nn.Conv2d(channels, out_channels, 3, padding=1)
+        self.conv = Conv2d(channels, out_channels, 3, padding=1)
 
     def forward(self, x):
         x = F.interpolate(x, scale_factor=2.0, mode="nearest")

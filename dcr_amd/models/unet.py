@@ -23,6 +23,7 @@ import torch.nn as nn
 from .attention import Transformer2DModel
 from .embeddings import TimestepEmbedding, timestep_embedding
 from .layers import GroupNormOp
+from ..ops.conv import Conv2d
 from .resnet import Downsample2D, ResnetBlock2D, Upsample2D
 
 
@@ -222,7 +223,7 @@ class UNet2DConditionModel(nn.Module):
         self.config = cfg
         ch = cfg.block_out_channels
         temb_ch = ch[0] * 4
-        self.conv_in = nn.Conv2d(cfg.in_channels, ch[0], 3, padding=1)
+        self.conv_in = Conv2d(cfg.in_channels, ch[0], 3, padding=1)
         self.time_embedding = TimestepEmbedding(ch[0], temb_ch)
 
         self.down_blocks = nn.ModuleList()
@@ -269,7 +270,7 @@ class UNet2DConditionModel(nn.Module):
 
         self.conv_norm_out = GroupNormOp(cfg.norm_num_groups, ch[0], eps=cfg.norm_eps,
                                          fused_silu=True)
-        self.conv_out = nn.Conv2d(ch[0], cfg.out_channels, 3, padding=1)
+        self.conv_out = Conv2d(ch[0], cfg.out_channels, 3, padding=1)
 
     @property
     def dtype(self):

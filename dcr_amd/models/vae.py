@@ -14,6 +14,7 @@ from typing import Optional, Tuple
 import torch
 import torch.nn as nn
 
+from ..ops.conv import Conv2d
 from .attention import Attention
 from .layers import GroupNormOp
 from .resnet import Downsample2D, ResnetBlock2D, Upsample2D
@@ -130,7 +131,7 @@ class Encoder(nn.Module):
     def __init__(self, cfg: VAEConfig):
         super().__init__()
         ch = cfg.block_out_channels
-        self.conv_in = nn.Conv2d(cfg.in_channels, ch[0], 3, padding=1)
+        self.conv_in = Conv2d(cfg.in_channels, ch[0], 3, padding=1)
         self.down_blocks = nn.ModuleList()
         out_c = ch[0]
         for i in range(len(ch)):
@@ -141,7 +142,7 @@ class Encoder(nn.Module):
         self.mid_block = MidBlock(ch[-1], cfg.norm_num_groups)
         self.conv_norm_out = GroupNormOp(cfg.norm_num_groups, ch[-1], eps=1e-6,
                                          fused_silu=True)
-        self.conv_out = nn.Conv2d(ch[-1], 2 * cfg.latent_channels, 3, padding=1)
+        self.conv_out = Conv2d(ch[-1], 2 * cfg.latent_channels, 3, padding=1)
 
     def forward(self, x):
         x = self.conv_in(x)
@@ -156,7 +157,7 @@ class Decoder(nn.Module):
     def __init__(self, cfg: VAEConfig):
         super().__init__()
         ch = list(reversed(cfg.block_out_channels))
-        self.conv_in = nn.Conv2d(cfg.latent_channels, ch[0], 3, padding=1)
+        self.conv_in = Conv2d(cfg.latent_channels, ch[0], 3, padding=1)
         self.mid_block = MidBlock(ch[0], cfg.norm_num_groups)
         self.up_blocks = nn.ModuleList()
         out_c = ch[0]
@@ -167,7 +168,7 @@ class Decoder(nn.Module):
                 add_upsample=i < len(ch) - 1))
         self.conv_norm_out = GroupNormOp(cfg.norm_num_groups, ch[-1], eps=1e-6,
                                          fused_silu=True)
-        self.conv_out = nn.Conv2d(ch[-1], cfg.out_channels, 3, padding=1)
+        self.conv_out = Conv2d(ch[-1], cfg.out_channels, 3, padding=1)
 
     def forward(self, z):
         z = self.conv_in(z)
@@ -215,8 +216,8 @@ class AutoencoderKL(nn.Module):
         self.config = cfg
         self.encoder = Encoder(cfg)
         self.decoder = Decoder(cfg)
-        self.quant_conv = nn.Conv2d(2 * cfg.latent_channels, 2 * cfg.latent_channels, 1)
-        self.post_quant_conv = nn.Conv2d(cfg.latent_channels, cfg.latent_channels, 1)
+        self.quant_conv = Conv2d(2 * cfg.latent_channels, 2 * cfg.latent_channels, 1)
+        self.post_quant_conv = Conv2d(cfg.latent_channels, cfg.latent_channels, 1)
 
     @property
     def dtype(self):

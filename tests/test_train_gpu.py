@@ -89,3 +89,36 @@ def test_channels_last_train_step_matches_nchw(tmp_path):
     # the correctness bar here (exact parity is covered per-kernel in
     # test_ops_gpu.py::test_groupnorm_nhwc_fwd_bwd)
     assert abs(a - b) / max(abs(a), 1e-6) < 0.25, (a, b)
+
+
+def test_native_conv_in_unet_shapes():
+    """the native conv fwd at the UNet's concatenated up-block widths."""
+    from dcr_amd import ops
+    m = ops.ext()
+    torch.manual_seed(2)
+    for (C, K, H) in [(1920, 1280, 16), (2560, 1280, 8), (960, 640, 32),
+                      (640, 320, 32)]:
+        x = torch.randn(4, C, H, H, device="cuda").to(torch.bfloat16) \
+            .to(memory_format=torch.channels_last)
+        w = (torch.randn(K, C, 3, 3, device="cuda") * 0.03).to(torch.bfloat16) \
+            .to(memory_format=torch.channels_last)
+        b = torch.randn(K, device="cuda")
+        y = m.conv2d_nhwc_fwd_v2(x, w, b, 1, 1)
+        ref = torch.nn.functional.conv2d(x.float(), w.float(), b, padding=1)
+        err = (y.float() - ref).abs().max().item()
+        assert err < 2e-2 * ref.abs().max().item(), (C, K, H, err)
+
+
+def test_native_conv_train_step(tmp_path):
+    """tiny channels_last train step with native conv dispatch active."""
+    from dcr_amd.train import TrainConfig, Trainer
+    cfg = TrainConfig(model_size="tiny", synthetic_data=True, synthetic_size=4,
+                      resolution=64, train_batch_size=2,
+                      mixed_precision="pure_bf16", channels_last=True,
+                      dataloader_num_workers=0, max_train_steps=2, seed=0,
+                      output_dir=str(tmp_path / "o"))
+    tr = Trainer(cfg, device=torch.device("cuda", 0))
+    b = next(iter(tr.dataloader))
+    l1 = tr.train_step(b)
+    l2 = tr.train_step(b)
+    assert torch.isfinite(l1) and torch.isfinite(l2)
