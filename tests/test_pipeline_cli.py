@@ -219,3 +219,57 @@ def test_retrieval_cli_splitloss(tmp_path):
     assert r.returncode == 0, r.stderr[-2000:]
     sim = torch.load(out / "similarity.pth")
     assert sim.shape == (3, 4)
+
+
+@pytest.mark.timeout(900)
+def test_full_product_pipeline(tmp_path):
+    """SURVEY §1 data flow end-to-end: diff_train writes checkpoint_{step}/
+    -> diff_inference reads it and writes generations/{i}.png + prompts.txt
+    -> diff_retrieval consumes (query=generations, val=train data)."""
+    root = Path(__file__).parent.parent
+    env = {**__import__("os").environ, "PYTHONPATH": str(root)}
+
+    # 1) train (tiny, synthetic, 2 steps) — note 'imagenette' in the name
+    #    so inference derives the imagenette10 savepath branch
+    out_root = tmp_path / "run_imagenette"
+    r = subprocess.run(
+        [sys.executable, str(root / "diff_train.py"), "--synthetic_data",
+         "--model_size", "tiny", "--resolution", "64", "--train_batch_size",
+         "2", "--max_train_steps", "2", "--mixed_precision", "no",
+         "--class_prompt", "classlevel", "--num_workers", "0", "--seed", "0",
+         "--save_steps", "1000", "--modelsavesteps", "1000",
+         "--output_dir", str(out_root)],
+        capture_output=True, text=True, cwd=str(tmp_path), env=env, timeout=400)
+    assert r.returncode == 0, r.stderr[-2000:]
+    model_dir = tmp_path / "run_imagenette_classlevel_nodup"
+    assert (model_dir / "checkpoint").is_dir()
+
+    # 2) inference from the checkpoint
+    r = subprocess.run(
+        [sys.executable, str(root / "diff_inference.py"), "--modelpath",
+         str(model_dir), "-nb", "3", "-imb", "1", "--resolution", "64",
+         "--seed", "0"],
+        capture_output=True, text=True, cwd=str(tmp_path), env=env, timeout=400)
+    assert r.returncode == 0, r.stderr[-2000:]
+    gen_dir = tmp_path / "inferences" / "imagenette10_frozentext" / \
+        "run_imagenette_classlevel_nodup" / "classlevel"
+    pngs = list((gen_dir / "generations").glob("*.png"))
+    assert len(pngs) == 3 and (gen_dir / "prompts.txt").exists()
+
+    # 3) retrieval: generations vs a synthetic "train" dir
+    rng = np.random.default_rng(0)
+    vdir = tmp_path / "train_imgs"
+    vdir.mkdir()
+    for i in range(5):
+        Image.fromarray(rng.integers(0, 255, (64, 64, 3)).astype(np.uint8)) \
+            .save(vdir / f"{i}.png")
+    out = tmp_path / "scores"
+    r = subprocess.run(
+        [sys.executable, str(root / "diff_retrieval.py"), "--query_dir",
+         str(gen_dir / "generations"), "--val_dir", str(vdir), "--pt_style",
+         "sscd", "-b", "4", "-j", "0", "--imsize", "64", "-ssp", str(out),
+         "--skip_fid", "--noeval"],
+        capture_output=True, text=True, cwd=str(tmp_path), env=env, timeout=400)
+    assert r.returncode == 0, r.stderr[-2000:]
+    sim = torch.load(out / "similarity.pth")
+    assert sim.shape == (3, 5)
