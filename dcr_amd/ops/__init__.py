@@ -84,6 +84,17 @@ def use_hip(*tensors: torch.Tensor) -> bool:
     return True
 
 
+# per-op HIP dispatch counters ("the native path is the one that runs"):
+# incremented by the functional wrappers on every HIP-kernel dispatch.
+from collections import Counter  # noqa: E402
+
+dispatch_counts: "Counter[str]" = Counter()
+
+
+def count_dispatch(name: str):
+    dispatch_counts[name] += 1
+
+
 from .functional import (  # noqa: E402,F401
     group_norm_silu,
     layer_norm,

@@ -22,7 +22,7 @@ from typing import Iterable, List
 
 import torch
 
-from . import use_hip, require_hip
+from . import use_hip, require_hip, count_dispatch
 
 
 class FusedAdamW:
@@ -87,6 +87,7 @@ class FusedAdamW:
         if use_hip(self.flat_param):
             m = require_hip("adamw")
             if m is not None:
+                count_dispatch('adamw')
                 if self.master is not None:
                     m.adamw_step_bf16(
                         self.flat_param, self.flat_grad, self.master,

@@ -19,7 +19,7 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
-from . import ext
+from . import count_dispatch, ext
 
 
 def _enabled() -> bool:
@@ -56,6 +56,7 @@ class _NativeConvFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, bias, stride, padding):
         m = ext()
+        count_dispatch('conv_nhwc')
         y = m.conv2d_nhwc_fwd_v2(x, weight, bias, stride, padding)
         ctx.save_for_backward(x, weight)
         ctx.conf = (stride, padding, bias is not None)

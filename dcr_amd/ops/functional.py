@@ -20,7 +20,7 @@ from typing import Optional
 import torch
 import torch.nn.functional as F
 
-from . import use_hip, require_hip
+from . import use_hip, require_hip, count_dispatch
 
 
 # --------------------------------------------------------------------------
@@ -33,6 +33,7 @@ class _GroupNormSiLUNHWC(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, bias, num_groups, eps, apply_silu):
         m = require_hip("group_norm_silu_nhwc")
+        count_dispatch('groupnorm_nhwc')
         y, mean, rstd = m.groupnorm_silu_nhwc_fwd(x, weight, bias, num_groups,
                                                   eps, apply_silu)
         ctx.save_for_backward(x, weight, bias, mean, rstd)
@@ -56,6 +57,7 @@ class _GroupNormSiLU(torch.autograd.Function):
         m = require_hip("group_norm_silu")
         if m is None:  # debug fallback on GPU
             return _gn_silu_ref(x, weight, bias, num_groups, eps, apply_silu)
+        count_dispatch('groupnorm')
         y, mean, rstd = m.groupnorm_silu_fwd(x, weight, bias, num_groups, eps, apply_silu)
         ctx.save_for_backward(x, weight, bias, mean, rstd)
         ctx.num_groups = num_groups
@@ -106,6 +108,7 @@ class _LayerNorm(torch.autograd.Function):
         m = require_hip("layer_norm")
         if m is None:
             return F.layer_norm(x, (x.shape[-1],), weight, bias, eps)
+        count_dispatch('layernorm')
         y, mean, rstd = m.layernorm_fwd(x, weight, bias, eps)
         ctx.save_for_backward(x, weight, mean, rstd)
         return y
@@ -139,6 +142,7 @@ class _GEGLU(torch.autograd.Function):
         if m is None:
             a, g = x.chunk(2, dim=-1)
             return a * F.gelu(g)
+        count_dispatch('geglu')
         y = m.geglu_fwd(x)
         ctx.save_for_backward(x)
         return y
@@ -167,6 +171,7 @@ class _FlashAttention(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v, scale, causal):
         m = require_hip("attn")
+        count_dispatch('attention')
         o, lse = m.attn_fwd(q, k, v, scale, causal)
         ctx.save_for_backward(q, k, v, o, lse)
         ctx.scale = scale
@@ -252,6 +257,7 @@ def add_noise(
 
     m = ext()
     if use_hip(x0) and m is not None:
+        count_dispatch('add_noise')
         return m.add_noise(x0.contiguous(), noise.contiguous(),
                            alphas_cumprod.to(x0.device, torch.float32), t.contiguous())
     sa, sb = _gather_sqrt(alphas_cumprod.to(x0.device), t, x0.dim())
@@ -289,6 +295,7 @@ def lincomb(x: torch.Tensor, y: torch.Tensor, a: float, b: float,
     if use_hip(x) and m is not None and x.numel() % 4 == 0 \
             and _dense(x) and x.stride() == y.stride() \
             and (z is None or z.stride() == x.stride()):
+        count_dispatch('lincomb')
         return m.lincomb(x, y, z, float(a), float(b), float(c))
     out = a * x.float() + b * y.float()
     if z is not None:

@@ -122,3 +122,22 @@ def test_native_conv_train_step(tmp_path):
     l1 = tr.train_step(b)
     l2 = tr.train_step(b)
     assert torch.isfinite(l1) and torch.isfinite(l2)
+
+
+def test_hip_kernels_are_the_executing_path(tmp_path):
+    """every hot op of a bench-config train step must dispatch to the
+    hand-written HIP kernels (fail-loud against silent eager fallback)."""
+    from dcr_amd import ops
+    from dcr_amd.train import TrainConfig, Trainer
+    ops.dispatch_counts.clear()
+    cfg = TrainConfig(model_size="tiny", synthetic_data=True, synthetic_size=4,
+                      resolution=64, train_batch_size=2,
+                      mixed_precision="pure_bf16", channels_last=True,
+                      dataloader_num_workers=0, max_train_steps=1, seed=0,
+                      output_dir=str(tmp_path / "o"))
+    tr = Trainer(cfg, device=torch.device("cuda", 0))
+    tr.train_step(next(iter(tr.dataloader)))
+    counts = dict(ops.dispatch_counts)
+    for op in ("groupnorm_nhwc", "layernorm", "geglu", "attention",
+               "add_noise", "adamw"):
+        assert counts.get(op, 0) > 0, (op, counts)
