@@ -130,8 +130,10 @@ def test_hip_kernels_are_the_executing_path(tmp_path):
     from dcr_amd import ops
     from dcr_amd.train import TrainConfig, Trainer
     ops.dispatch_counts.clear()
-    cfg = TrainConfig(model_size="tiny", synthetic_data=True, synthetic_size=4,
-                      resolution=64, train_batch_size=2,
+    # sd21: head_dim 64 + K%64 channels so EVERY native path is eligible
+    # (the tiny config's 32-wide heads legitimately use the math fallback)
+    cfg = TrainConfig(model_size="sd21", synthetic_data=True, synthetic_size=2,
+                      resolution=256, train_batch_size=1,
                       mixed_precision="pure_bf16", channels_last=True,
                       dataloader_num_workers=0, max_train_steps=1, seed=0,
                       output_dir=str(tmp_path / "o"))
@@ -139,5 +141,5 @@ def test_hip_kernels_are_the_executing_path(tmp_path):
     tr.train_step(next(iter(tr.dataloader)))
     counts = dict(ops.dispatch_counts)
     for op in ("groupnorm_nhwc", "layernorm", "geglu", "attention",
-               "add_noise", "adamw"):
+               "add_noise", "adamw", "conv_nhwc"):
         assert counts.get(op, 0) > 0, (op, counts)
