@@ -126,4 +126,5 @@ def test_ddim_full_denoise_recovers_clean_direction():
         ac = s.alphas_cumprod[int(t)]
         eps_hat = (x - ac.sqrt() * x0) / (1 - ac).sqrt()
         x = s.step(eps_hat, int(t), x).prev_sample
-    assert (x - x0).abs().max() < 0.05, (x - x0).abs().max()
+    # final step keeps sqrt(1-ac[0]) ~ 0.029 of eps (set_alpha_to_one=False)
+    assert (x - x0).abs().max() < 0.15, (x - x0).abs().max()
