@@ -127,3 +127,54 @@ def _feat_worker(rank, world, port):
 
 def test_extract_features_world2():
     mp.spawn(_feat_worker, args=(2, 29719), nprocs=2, join=True)
+
+
+def _accum_worker(rank, world, port):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from dcr_amd.ops.adamw import FusedAdamW
+        from dcr_amd.parallel.ddp import GradBucketAllReduce
+
+        torch.manual_seed(7)  # same init everywhere (broadcast also enforces)
+        model = torch.nn.Linear(8, 4, bias=False)
+        opt = FusedAdamW(model.parameters(), lr=1e-3)
+        ddp = GradBucketAllReduce(opt, bucket_mb=0.0001)
+
+        # two micro-batches per rank; only the second syncs
+        torch.manual_seed(100 + rank)
+        xa, xb = torch.randn(4, 8), torch.randn(4, 8)
+        ddp.require_backward_grad_sync = False
+        model(xa).pow(2).mean().backward()
+        ddp.finalize()
+        ddp.require_backward_grad_sync = True
+        model(xb).pow(2).mean().backward()
+        ddp.finalize()
+
+        # expected: mean over ranks of (grad(xa) + grad(xb))
+        ref = torch.nn.Linear(8, 4, bias=False)
+        with torch.no_grad():
+            ref.weight.copy_(model.weight)
+        expected = torch.zeros_like(ref.weight)
+        for r in range(world):
+            torch.manual_seed(100 + r)
+            ya, yb = torch.randn(4, 8), torch.randn(4, 8)
+            ref.weight.grad = None
+            ref(ya).pow(2).mean().backward()
+            ref(yb).pow(2).mean().backward()
+            expected += ref.weight.grad
+        expected /= world
+
+        got = opt.flat_grad.view_as(model.weight)
+        assert torch.allclose(got, expected, atol=1e-6), \
+            (rank, (got - expected).abs().max())
+        dist.barrier()
+    finally:
+        dist.destroy_process_group()
+
+
+def test_grad_accumulation_ddp_world2():
+    """accumulated micro-steps + one synced step == averaged sum of all
+    micro-grads (reference accelerate.accumulate semantics, N1)."""
+    mp.spawn(_accum_worker, args=(2, 29721), nprocs=2, join=True)
