@@ -117,3 +117,19 @@ def test_sd14_tiny_forward_shape():
     unet = U.UNet2DConditionModel(cfg)
     out = unet(torch.randn(1, 4, 8, 8), torch.tensor([5]), torch.randn(1, 7, 48))
     assert out.shape == (1, 4, 8, 8)
+
+
+def test_sd21_state_dict_fingerprint():
+    """key-set fingerprint (name:shape) of the SD-2.1 state dicts — the
+    diffusers-interop contract; a changed fingerprint means checkpoints
+    stop loading. Update ONLY with a deliberate layout change."""
+    import hashlib
+
+    def fp(module):
+        blob = "\n".join(f"{k}:{tuple(v.shape)}"
+                         for k, v in sorted(module.state_dict().items()))
+        return hashlib.sha256(blob.encode()).hexdigest()[:16]
+
+    assert fp(UNet2DConditionModel(UNetConfig.sd21())) == "d3d79e5d405aa4d8"
+    assert fp(AutoencoderKL(VAEConfig.sd())) == "40b5616ff8102279"
+    assert fp(CLIPTextModel(CLIPTextConfig.sd21())) == "5651574a79d93060"
