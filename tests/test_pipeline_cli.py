@@ -273,3 +273,41 @@ def test_full_product_pipeline(tmp_path):
     assert r.returncode == 0, r.stderr[-2000:]
     sim = torch.load(out / "similarity.pth")
     assert sim.shape == (3, 5)
+
+
+@pytest.mark.timeout(600)
+def test_retrieval_cli_torchrun_world2(tmp_path):
+    """distributed retrieval CLI (2 ranks, gloo): sharded extraction +
+    all-gather must produce the same similarity artifacts."""
+    import os as _os
+    rng = np.random.default_rng(3)
+    qdir, vdir = tmp_path / "gens", tmp_path / "train"
+    qdir.mkdir(); vdir.mkdir()
+    for i in range(5):
+        Image.fromarray(rng.integers(0, 255, (64, 64, 3)).astype(np.uint8)) \
+            .save(qdir / f"{i}.png")
+    for i in range(7):
+        Image.fromarray(rng.integers(0, 255, (64, 64, 3)).astype(np.uint8)) \
+            .save(vdir / f"{i}.png")
+    root = Path(__file__).parent.parent
+
+    def run(out, extra):
+        r = subprocess.run(
+            extra + [str(root / "diff_retrieval.py"), "--query_dir", str(qdir),
+                     "--val_dir", str(vdir), "--pt_style", "sscd", "-b", "4",
+                     "-j", "0", "--imsize", "64", "-ssp", str(out),
+                     "--skip_fid", "--noeval", "--dontsave"],
+            capture_output=True, text=True, cwd=str(root),
+            env={**_os.environ, "MASTER_ADDR": "127.0.0.1"}, timeout=560)
+        assert r.returncode == 0, r.stderr[-2000:]
+        log = out / "imsimv2_retrieval_log.jsonl"
+        recs = [json.loads(l) for l in log.read_text().splitlines()]
+        return next(r_ for r_ in recs if "sim_mean" in r_)
+
+    single = run(tmp_path / "o1", [sys.executable])
+    multi = run(tmp_path / "o2",
+                [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+                 "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+                 "--master-port", "29537"])
+    assert abs(single["sim_mean"] - multi["sim_mean"]) < 1e-4
+    assert abs(single["bg_mean"] - multi["bg_mean"]) < 1e-4
