@@ -60,3 +60,45 @@ def test_aten_convolution_backward_contract():
     assert torch.allclose(dx, x.grad, atol=1e-5)
     assert torch.allclose(dw, w.grad, atol=1e-4)
     assert torch.allclose(db, b.grad, atol=1e-5)
+
+
+def test_fused_adamw_cpu_matches_torch_adamw():
+    """the CPU fallback math (which anchors resume + GPU-kernel tests)
+    must match torch.optim.AdamW step-for-step."""
+    torch.manual_seed(0)
+    lin_a = torch.nn.Linear(16, 8)
+    lin_b = torch.nn.Linear(16, 8)
+    lin_b.load_state_dict(lin_a.state_dict())
+
+    from dcr_amd.ops.adamw import FusedAdamW
+    ours = FusedAdamW(lin_a.parameters(), lr=1e-3, betas=(0.9, 0.999),
+                      eps=1e-8, weight_decay=1e-2)
+    ref = torch.optim.AdamW(lin_b.parameters(), lr=1e-3, betas=(0.9, 0.999),
+                            eps=1e-8, weight_decay=1e-2)
+    for step in range(5):
+        torch.manual_seed(100 + step)
+        x = torch.randn(4, 16)
+        la = lin_a(x).pow(2).mean()
+        lb = lin_b(x).pow(2).mean()
+        la.backward()
+        lb.backward()
+        ours.step()
+        ours.zero_grad()
+        ref.step()
+        ref.zero_grad()
+    for pa, pb in zip(lin_a.parameters(), lin_b.parameters()):
+        assert torch.allclose(pa, pb, atol=1e-6), (pa - pb).abs().max()
+
+
+def test_clip_grad_norm_matches_torch():
+    from dcr_amd.ops.adamw import FusedAdamW
+    torch.manual_seed(1)
+    lin = torch.nn.Linear(32, 32)
+    opt = FusedAdamW(lin.parameters(), lr=1e-3)
+    lin(torch.randn(8, 32)).pow(2).sum().backward()
+    grads = [p.grad.clone() for p in lin.parameters()]
+    total = torch.sqrt(sum(g.pow(2).sum() for g in grads))
+    norm = opt.clip_grad_norm_(0.5)
+    assert torch.allclose(norm, total, atol=1e-5)
+    clipped = torch.linalg.vector_norm(opt.flat_grad)
+    assert clipped <= 0.5 * (1 + 1e-4)

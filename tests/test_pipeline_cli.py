@@ -311,3 +311,11 @@ def test_retrieval_cli_torchrun_world2(tmp_path):
                  "--master-port", "29537"])
     assert abs(single["sim_mean"] - multi["sim_mean"]) < 1e-4
     assert abs(single["bg_mean"] - multi["bg_mean"]) < 1e-4
+
+
+def test_pipeline_no_cfg_path():
+    """guidance_scale <= 1 skips the CFG double batch."""
+    pipe = tiny_pipe()
+    out = pipe("x", height=64, width=64, num_inference_steps=2,
+               guidance_scale=1.0, output_type="pt")
+    assert out.images.shape == (1, 3, 64, 64)
