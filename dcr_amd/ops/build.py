@@ -24,6 +24,7 @@ SOURCES = [
     HIP_DIR / "elementwise.hip",
     HIP_DIR / "attention.hip",
     HIP_DIR / "conv_nhwc.hip",
+    HIP_DIR / "conv_nhwc_bwd.hip",
 ]
 
 

@@ -420,7 +420,37 @@ at::Tensor conv2d_nhwc_fwd_v2(at::Tensor x, at::Tensor w,
   return y;
 }
 
+// conv backward drafts (round-2; validated before any dispatch)
+std::vector<at::Tensor> conv2d_nhwc_bwd(at::Tensor dy, at::Tensor x,
+                                        at::Tensor w, int64_t stride,
+                                        int64_t pad) {
+  TORCH_CHECK(dy.is_cuda() && dy.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(dy.is_contiguous(at::MemoryFormat::ChannelsLast) &&
+              x.is_contiguous(at::MemoryFormat::ChannelsLast) &&
+              w.is_contiguous(at::MemoryFormat::ChannelsLast));
+  const int64_t Nb = x.size(0), C = x.size(1), Hin = x.size(2), Win = x.size(3);
+  const int64_t K = w.size(0), R = w.size(2), S = w.size(3);
+  TORCH_CHECK(C % 64 == 0 && K % 64 == 0);
+  const int64_t P = dy.size(2), Q = dy.size(3);
+  const int64_t NPQ = Nb * P * Q;
+  int splits = (int)std::min<int64_t>(std::max<int64_t>(
+      384 / std::max<int64_t>((K / 64) * (R * S * C / 32), 1), 1), 16);
+  auto dw_ws = at::zeros({K * R * S * C}, x.options().dtype(at::kFloat));
+  auto dW = at::empty_like(w);
+  conv_bwd_weight_launch(dy.data_ptr(), x.data_ptr(), dw_ws.data_ptr<float>(),
+                         (int)Nb, (int)Hin, (int)Win, (int)C, (int)K, (int)P,
+                         (int)Q, (int)R, (int)S, (int)stride, (int)pad,
+                         splits, dW.data_ptr(), cur_stream());
+  auto dx = at::empty_like(x);
+  conv_bwd_data_launch(dy.data_ptr(), w.data_ptr(), dx.data_ptr(), (int)Nb,
+                       (int)Hin, (int)Win, (int)C, (int)K, (int)P, (int)Q,
+                       (int)R, (int)S, (int)stride, (int)pad, cur_stream());
+  (void)NPQ;
+  return {dx, dW};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("conv2d_nhwc_bwd", &conv2d_nhwc_bwd);
   mod.def("conv2d_nhwc_fwd", &conv2d_nhwc_fwd);
   mod.def("conv2d_nhwc_fwd_v2", &conv2d_nhwc_fwd_v2);
   mod.def("attn_fwd", &attn_fwd);

@@ -76,4 +76,13 @@ void conv_nhwc_fwd_v2_launch(const void* x, const void* w, const float* bias,
                              int Win, int C, int K, int P, int Q, int R, int S,
                              int stride, int pad, hipStream_t st);
 
+// conv_nhwc_bwd.hip (round-2 drafts; not dispatched)
+void conv_bwd_weight_launch(const void* dy, const void* x, float* dw_ws,
+                            int Nb, int Hin, int Win, int C, int K, int P,
+                            int Q, int R, int S, int stride, int pad,
+                            int splits, void* dW_out, hipStream_t st);
+void conv_bwd_data_launch(const void* dy, const void* w, void* dx, int Nb,
+                          int Hin, int Win, int C, int K, int P, int Q, int R,
+                          int S, int stride, int pad, hipStream_t st);
+
 }  // namespace dcr

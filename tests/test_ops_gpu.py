@@ -387,3 +387,31 @@ def test_conv_nhwc_fwd(ext, shape, version):
                                      padding=pad)
     assert y.is_contiguous(memory_format=torch.channels_last)
     _close(y, ref, 2e-2)
+
+
+@pytest.mark.skipif(os.environ.get("DCR_NATIVE_CONV_BWD") != "1",
+                    reason="conv bwd drafts: validate in round 2 "
+                           "(DCR_NATIVE_CONV_BWD=1)")
+@pytest.mark.parametrize("shape", [
+    (2, 320, 32, 32, 320, 3, 1),
+    (2, 640, 16, 16, 640, 3, 1),
+    (2, 128, 64, 64, 128, 1, 1),
+    (2, 640, 32, 32, 640, 3, 2),
+])
+def test_conv_nhwc_bwd(ext, shape):
+    N, C, H, W, K, R, stride = shape
+    pad = 1 if R == 3 else 0
+    torch.manual_seed(0)
+    x = torch.randn(N, C, H, W, device="cuda").to(torch.bfloat16) \
+        .to(memory_format=torch.channels_last)
+    w = (torch.randn(K, C, R, R, device="cuda") * 0.05).to(torch.bfloat16) \
+        .to(memory_format=torch.channels_last)
+    P = (H + 2 * pad - R) // stride + 1
+    dy = torch.randn(N, K, P, P, device="cuda").to(torch.bfloat16) \
+        .to(memory_format=torch.channels_last)
+    dx, dW = ext.conv2d_nhwc_bwd(dy, x, w, stride, pad)
+    rx, rw, _ = torch.ops.aten.convolution_backward(
+        dy.float(), x.float(), w.float(), None, [stride, stride], [pad, pad],
+        [1, 1], False, [0, 0], 1, [True, True, False])
+    _close(dx, rx, 3e-2)
+    _close(dW, rw, 3e-2)
