@@ -176,3 +176,29 @@ def test_load_real_clip_tokenizer(tmp_path):
               return_tensors="pt")
     assert out.input_ids.shape == (1, 16)
     assert out.input_ids[0, 0].item() == 0  # bos
+
+
+def test_transforms_shapes_and_range():
+    from dcr_amd.data.transforms import (TrainTransform, EvalTransform,
+                                         resize_shorter, center_crop)
+    img = Image.new("RGB", (100, 60), color=(200, 30, 80))
+    r = resize_shorter(img, 48)
+    assert min(r.size) == 48 and r.size[0] == 80
+    c = center_crop(r, 48)
+    assert c.size == (48, 48)
+    t = TrainTransform(32, center_crop=True)(img)
+    assert t.shape == (3, 32, 32) and -1.001 <= t.min() and t.max() <= 1.001
+    e = EvalTransform(32)(img)
+    assert e.shape == (3, 32, 32)
+
+
+def test_image_grid_utils():
+    from dcr_amd.utils import concat_h, image_grid, tensor_to_pil
+    imgs = [Image.new("RGB", (8, 8), color=(i * 40, 0, 0)) for i in range(4)]
+    row = concat_h(imgs)
+    assert row.size == (32, 8)
+    grid = image_grid(imgs, rows=2, cols=2)
+    assert grid.size == (16, 16)
+    t = torch.rand(3, 8, 8) * 2 - 1
+    p = tensor_to_pil(t)
+    assert p.size == (8, 8)
