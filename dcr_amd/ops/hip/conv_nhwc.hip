@@ -174,7 +174,7 @@ void conv_nhwc_fwd_v2_kernel(const bf16_t* __restrict__ x, const bf16_t* __restr
 
   // staging: 2 threads per row, each covers BK/2 contiguous bf16
   const int st_row = threadIdx.x >> 1;     // 0..127
-  const int st_c16 = (threadIdx.x & 1) * (BK / 2);
+  const int st_chalf = (threadIdx.x & 1) * (BK / 2);  // half-row column base
   long st_m = m0 + st_row;
   int st_n = 0, st_p = 0, st_q = 0;
   if (st_m < NPQ) {
@@ -210,17 +210,17 @@ void conv_nhwc_fwd_v2_kernel(const bf16_t* __restrict__ x, const bf16_t* __restr
       const int wi = st_q * stride + s - pad;
       if (st_m < NPQ && hi >= 0 && hi < Hin && wi >= 0 && wi < Win) {
         const uint4* p4 = reinterpret_cast<const uint4*>(
-            x + (((long)st_n * Hin + hi) * Win + wi) * C + c0 + st_c16);
+            x + (((long)st_n * Hin + hi) * Win + wi) * C + c0 + st_chalf);
 #pragma unroll
         for (int t = 0; t < NCH; ++t) av[t] = p4[t];
       }
       if (k0 + st_row < K) {
-        const uint4* p4 = reinterpret_cast<const uint4*>(w + wrow + rsc0 + st_c16);
+        const uint4* p4 = reinterpret_cast<const uint4*>(w + wrow + rsc0 + st_chalf);
 #pragma unroll
         for (int t = 0; t < NCH; ++t) bv[t] = p4[t];
       }
-      uint4* d = reinterpret_cast<uint4*>(sA + st_row * PITCH2 + st_c16);
-      uint4* db = reinterpret_cast<uint4*>(sB + st_row * PITCH2 + st_c16);
+      uint4* d = reinterpret_cast<uint4*>(sA + st_row * PITCH2 + st_chalf);
+      uint4* db = reinterpret_cast<uint4*>(sB + st_row * PITCH2 + st_chalf);
 #pragma unroll
       for (int t = 0; t < NCH; ++t) d[t] = av[t];
 #pragma unroll
