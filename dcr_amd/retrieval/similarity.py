@@ -126,6 +126,10 @@ def jpeg_size(pil_img, quality: int = 95) -> int:
 
 
 def pearson(x: np.ndarray, y: np.ndarray) -> float:
+    return pearson_with_p(x, y)[0]
+
+
+def pearson_with_p(x: np.ndarray, y: np.ndarray) -> Tuple[float, float]:
     from scipy.stats import pearsonr
-    r, _ = pearsonr(np.asarray(x, dtype=np.float64), np.asarray(y, dtype=np.float64))
-    return float(r)
+    r, p = pearsonr(np.asarray(x, dtype=np.float64), np.asarray(y, dtype=np.float64))
+    return float(r), float(p)

@@ -28,8 +28,9 @@ from dcr_amd.parallel import barrier, init_distributed_mode, is_main_process
 from dcr_amd.retrieval import (einsum_in_chunks, extract_features,
                                gen_clipscore, glcm_entropy, jpeg_size,
                                l2_normalize, load_clip, load_dino, load_sscd,
-                               pearson, sim_matrix, similarity_histogram,
-                               top_matches, topk_stats, tv_loss)
+                               pearson_with_p, sim_matrix,
+                               similarity_histogram, top_matches, topk_stats,
+                               tv_loss)
 from dcr_amd.utils import Tracker
 
 
@@ -195,37 +196,54 @@ def main():
             rec["clipscore_bg"] = csb.mean().item()
         tracker.log(rec)
 
-    # complexity correlations (reference :498-540)
+    # complexity correlations (reference :498-540): "are SIMPLE TRAINING
+    # images copied more?" — metrics are computed on each generation's
+    # top-1 MATCHED TRAIN image, correlated with the match similarity.
     if not args.noeval:
         from PIL import Image
-        ent, jpg, tv = [], [], []
-        for f in query_ds.files:
-            img = Image.open(f).convert("RGB")
+        match_idx = sim.argmax(dim=1).tolist()
+        dbsims = top1.numpy()
+        ent, crs, tv = [], [], []
+        for loc in match_idx:
+            img = Image.open(val_ds.files[loc]).convert("RGB")
             arr = np.asarray(img)
             ent.append(glcm_entropy(arr))
-            jpg.append(jpeg_size(img))
-            t = torch.from_numpy(arr.copy()).permute(2, 0, 1).float() / 255
+            crs.append(jpeg_size(img, quality=90) / 1024)   # KB, reference :524
+            t = torch.from_numpy(arr.copy()).permute(2, 0, 1).float()
             tv.append(tv_loss(t).item())
-        sims_np = top1.numpy()
-        comp = {
-            "entropy_sim_pearson": pearson(ent, sims_np),
-            "jpegsize_sim_pearson": pearson(jpg, sims_np),
-            "tv_sim_pearson": pearson(tv, sims_np),
-        }
-        tracker.log(comp)
+        ent = np.array(ent); crs = np.array(crs); tv = np.array(tv)
+        cc_ent, pval_ent = pearson_with_p(ent, dbsims)
+        cc_comp, pval_comp = pearson_with_p(crs, dbsims)
+        cc_tvl, pval_tvl = pearson_with_p(tv, dbsims)
+        cc_mixed, pval_mixed = pearson_with_p(ent * crs ** 0.5, dbsims)
+        tracker.log({
+            "cc_ent": cc_ent, "pval_ent": pval_ent,
+            "cc_comp": cc_comp, "pval_comp": pval_comp,
+            "cc_tvl": cc_tvl, "pval_tvl": pval_tvl,
+            "cc_mixed": cc_mixed, "pval_mixed": pval_mixed,
+        })
         if not args.dontsave:
-            torch.save({"entropy": ent, "jpeg": jpg, "tv": tv},
-                       out_dir / "complexity.pth")
+            # reference artifact names (:553-556)
+            torch.save(ent, out_dir / "entropies.pth")
+            torch.save(tv, out_dir / "totvar.pth")
+            torch.save(crs, out_dir / "compressions.pth")
+            torch.save(dbsims, out_dir / "dbsims.pth")
             try:
                 import matplotlib
                 matplotlib.use("Agg")
                 import matplotlib.pyplot as plt
-                for name, vals in (("entropy", ent), ("jpegsize", jpg),
-                                   ("tv", tv)):
+                # reference filenames (:544-559; the reference saves the
+                # 4th plot over the 3rd — SURVEY §2.6.7 — we keep all)
+                for name, vals, cc, pv in (
+                        ("entropies", ent, cc_ent, pval_ent),
+                        ("tvls", tv, cc_tvl, pval_tvl),
+                        ("crs", crs, cc_comp, pval_comp),
+                        ("mixed", ent * crs ** 0.5, cc_mixed, pval_mixed)):
                     plt.figure(figsize=(4, 4))
-                    plt.scatter(vals, sims_np, s=8, alpha=0.6)
-                    plt.xlabel(name)
-                    plt.ylabel("top-1 similarity")
+                    plt.scatter(vals, dbsims, s=8, alpha=0.6)
+                    plt.xlabel("simplicity")
+                    plt.ylabel("sims")
+                    plt.title(f"CC={cc:.4f}, pval={pv:.3g}")
                     plt.tight_layout()
                     plt.savefig(out_dir / f"simplicityscatter_{name}.png", dpi=110)
                     plt.close()

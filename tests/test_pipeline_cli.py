@@ -319,3 +319,36 @@ def test_pipeline_no_cfg_path():
     out = pipe("x", height=64, width=64, num_inference_steps=2,
                guidance_scale=1.0, output_type="pt")
     assert out.images.shape == (1, 3, 64, 64)
+
+
+@pytest.mark.timeout(600)
+def test_retrieval_complexity_on_matched_train_images(tmp_path):
+    """complexity metrics are computed on the matched TRAIN image and
+    logged with the reference's cc_*/pval_* keys + artifact names."""
+    rng = np.random.default_rng(5)
+    qdir, vdir = tmp_path / "gens", tmp_path / "train"
+    qdir.mkdir(); vdir.mkdir()
+    for i in range(4):
+        Image.fromarray(rng.integers(0, 255, (64, 64, 3)).astype(np.uint8)) \
+            .save(qdir / f"{i}.png")
+    for i in range(6):
+        Image.fromarray(rng.integers(0, 255, (64, 64, 3)).astype(np.uint8)) \
+            .save(vdir / f"{i}.png")
+    out = tmp_path / "s"
+    r = subprocess.run(
+        [sys.executable, "diff_retrieval.py", "--query_dir", str(qdir),
+         "--val_dir", str(vdir), "--pt_style", "sscd", "-b", "4", "-j", "0",
+         "--imsize", "64", "-ssp", str(out), "--skip_fid"],
+        capture_output=True, text=True, cwd=str(Path(__file__).parent.parent),
+        timeout=570)
+    assert r.returncode == 0, r.stderr[-2000:]
+    for name in ("entropies.pth", "totvar.pth", "compressions.pth",
+                 "dbsims.pth"):
+        assert (out / name).exists(), name
+    assert torch.load(out / "dbsims.pth", weights_only=False).shape == (4,)
+    recs = [json.loads(l) for l in
+            (out / "imsimv2_retrieval_log.jsonl").read_text().splitlines()]
+    comp = next(r_ for r_ in recs if "cc_ent" in r_)
+    for k in ("cc_ent", "pval_ent", "cc_comp", "pval_comp", "cc_tvl",
+              "pval_tvl", "cc_mixed", "pval_mixed"):
+        assert k in comp, k
