@@ -133,3 +133,18 @@ def test_sd21_state_dict_fingerprint():
     assert fp(UNet2DConditionModel(UNetConfig.sd21())) == "d3d79e5d405aa4d8"
     assert fp(AutoencoderKL(VAEConfig.sd())) == "40b5616ff8102279"
     assert fp(CLIPTextModel(CLIPTextConfig.sd21())) == "5651574a79d93060"
+
+
+def test_clip_quick_gelu_differs_from_gelu():
+    """SD-1.4's text tower uses quick_gelu (x*sigmoid(1.702x))."""
+    from dcr_amd.models.clip_text import CLIPMLP, CLIPTextConfig
+    torch.manual_seed(0)
+    cfg_q = CLIPTextConfig(hidden_size=16, intermediate_size=32,
+                           num_hidden_layers=1, num_attention_heads=2,
+                           hidden_act="quick_gelu")
+    m = CLIPMLP(cfg_q)
+    x = torch.randn(2, 4, 16)
+    yq = m(x)
+    m.act = "gelu"
+    yg = m(x)
+    assert not torch.allclose(yq, yg)

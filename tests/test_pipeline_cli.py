@@ -352,3 +352,12 @@ def test_retrieval_complexity_on_matched_train_images(tmp_path):
     for k in ("cc_ent", "pval_ent", "cc_comp", "pval_comp", "cc_tvl",
               "pval_tvl", "cc_mixed", "pval_mixed"):
         assert k in comp, k
+
+
+def test_sd_mitigation_prompt_list():
+    """the 12 known-replicating LAION prompts (reference sd_mitigation.py:81)."""
+    import sd_mitigation as sm
+    assert len(sm.PROMPT_LIST) == 12
+    assert sm.PROMPT_LIST[0] == "Wall View 002"
+    assert "The No Limits Business Woman Podcast" in sm.PROMPT_LIST
+    assert "Mothers influence on her young hippo" in sm.PROMPT_LIST
