@@ -361,3 +361,39 @@ def test_sd_mitigation_prompt_list():
     assert sm.PROMPT_LIST[0] == "Wall View 002"
     assert "The No Limits Business Woman Podcast" in sm.PROMPT_LIST
     assert "Mothers influence on her young hippo" in sm.PROMPT_LIST
+
+
+@pytest.mark.timeout(900)
+def test_inference_cli_mitigations(tmp_path):
+    """Newpipe embedding-noise + prompt augmentation via the CLI:
+    savepath suffixes (reference :77-81) and different outputs."""
+    ckpt_root = tmp_path / "run_imagenette_classlevel_nodup"
+    pipe = tiny_pipe()
+    pipe.save_pretrained(ckpt_root / "checkpoint")
+    root = Path(__file__).parent.parent
+    env = {**__import__("os").environ, "PYTHONPATH": str(root)}
+
+    r = subprocess.run(
+        [sys.executable, str(root / "diff_inference.py"), "--modelpath",
+         str(ckpt_root), "-nb", "1", "-imb", "1", "--resolution", "64",
+         "--seed", "0", "--rand_noise_lam", "0.5"],
+        capture_output=True, text=True, cwd=str(tmp_path), env=env, timeout=430)
+    assert r.returncode == 0, r.stderr[-2000:]
+    g1 = tmp_path / "inferences" / "imagenette10_frozentext" / \
+        "run_imagenette_classlevel_nodup" / "classlevel_ginfer0.5"
+    assert (g1 / "generations" / "0.png").exists()
+
+    r = subprocess.run(
+        [sys.executable, str(root / "diff_inference.py"), "--modelpath",
+         str(ckpt_root), "-nb", "1", "-imb", "1", "--resolution", "64",
+         "--seed", "0", "--capstyle", "instancelevel_blip",
+         "--rand_augs", "rand_word_add", "--rand_aug_repeats", "2"],
+        capture_output=True, text=True, cwd=str(tmp_path), env=env, timeout=430)
+    assert r.returncode == 0, r.stderr[-2000:]
+    g2 = tmp_path / "inferences" / "imagenette10_frozentext" / \
+        "run_imagenette_classlevel_nodup" / \
+        "instancelevel_blip_auginfer_rand_word_add_2"
+    assert (g2 / "prompts.txt").exists()
+    # augmented prompt gained 2 words vs the synthetic pool's base form
+    aug = (g2 / "prompts.txt").read_text().strip()
+    assert len(aug.split()) >= 7  # "An image of X variant N" + 2 words
