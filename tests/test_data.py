@@ -202,3 +202,22 @@ def test_image_grid_utils():
     t = torch.rand(3, 8, 8) * 2 - 1
     p = tensor_to_pil(t)
     assert p.size == (8, 8)
+
+
+def test_dataset_with_real_clip_tokenizer(tmp_path, image_folder):
+    """ObjectAttributeDataset works with a real transformers CLIPTokenizer
+    (same __call__/decode surface as HashTokenizer)."""
+    root, pj = image_folder
+    vocab = {"<|startoftext|>": 0, "<|endoftext|>": 1}
+    for i, ch in enumerate("abcdefghijklmnopqrstuvwxyz"):
+        vocab[ch] = 2 + i
+        vocab[ch + "</w>"] = 28 + i
+    (tmp_path / "tok" ).mkdir(exist_ok=True)
+    (tmp_path / "tok" / "vocab.json").write_text(json.dumps(vocab))
+    (tmp_path / "tok" / "merges.txt").write_text("#version: 0.2\n")
+    from dcr_amd.data.tokenizer import load_tokenizer
+    tok = load_tokenizer(tmp_path / "tok", model_max_length=77)
+    ds = ObjectAttributeDataset(str(root), tok, class_prompt="classlevel",
+                                size=64)
+    ex = ds[0]
+    assert ex["instance_prompt_ids"].shape == (1, 77)
