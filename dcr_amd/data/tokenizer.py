@@ -106,7 +106,10 @@ def load_tokenizer(path=None, model_max_length: int = 77):
         if (p / "vocab.json").exists() and (p / "merges.txt").exists():
             try:
                 from transformers import CLIPTokenizer
-                return CLIPTokenizer.from_pretrained(str(p))
+                tok = CLIPTokenizer.from_pretrained(str(p))
+                if tok.model_max_length > 10**6:  # no tokenizer_config.json
+                    tok.model_max_length = model_max_length
+                return tok
             except Exception:
                 pass
         cfgf = p / "tokenizer_config.json"
