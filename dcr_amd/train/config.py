@@ -107,8 +107,10 @@ def validate(cfg: TrainConfig):
 
 
 def get_lr(cfg: TrainConfig, step: int) -> float:
-    """LR schedules: constant / constant_with_warmup / linear / cosine
-    (reference choices: diff_train.py:178-189 via diffusers get_scheduler)."""
+    """LR schedules: all six reference choices (diff_train.py:178-189 via
+    diffusers get_scheduler, called with only warmup/total steps at
+    diff_train.py:504-509 — so cosine_with_restarts uses 1 cycle and
+    polynomial uses power=1, lr_end=1e-7, the library defaults)."""
     base = cfg.learning_rate
     total = cfg.max_train_steps or 1
     warm = cfg.lr_warmup_steps
@@ -126,4 +128,20 @@ def get_lr(cfg: TrainConfig, step: int) -> float:
             return base * (step + 1) / max(1, warm)
         prog = (step - warm) / max(1, total - warm)
         return base * 0.5 * (1 + math.cos(math.pi * min(1.0, prog)))
+    if s == "cosine_with_restarts":
+        if step < warm:
+            return base * (step + 1) / max(1, warm)
+        prog = (step - warm) / max(1, total - warm)
+        if prog >= 1.0:
+            return 0.0
+        num_cycles = 1
+        return base * max(0.0, 0.5 * (1 + math.cos(math.pi * ((num_cycles * prog) % 1.0))))
+    if s == "polynomial":
+        lr_end, power = 1e-7, 1.0
+        if step < warm:
+            return base * (step + 1) / max(1, warm)
+        if step > total:
+            return lr_end
+        remaining = 1 - (step - warm) / max(1, total - warm)
+        return (base - lr_end) * remaining ** power + lr_end
     raise ValueError(f"unknown lr_scheduler {s}")

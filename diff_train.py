@@ -45,7 +45,8 @@ def parse_args():
     p.add_argument("--learning_rate", type=float, default=5e-6)
     p.add_argument("--scale_lr", action="store_true")
     p.add_argument("--lr_scheduler", type=str, default="constant",
-                   choices=["linear", "cosine", "constant", "constant_with_warmup"])
+                   choices=["linear", "cosine", "cosine_with_restarts",
+                            "polynomial", "constant", "constant_with_warmup"])
     p.add_argument("--lr_warmup_steps", type=int, default=500)
     p.add_argument("--adam_beta1", type=float, default=0.9)
     p.add_argument("--adam_beta2", type=float, default=0.999)

@@ -104,6 +104,41 @@ def test_lr_schedule():
     assert get_lr(cfg, 50) == pytest.approx(1e-3)
 
 
+def test_lr_schedule_all_reference_choices():
+    """All six reference schedules (diff_train.py:178-189) are implemented
+    with diffusers get_scheduler semantics."""
+    import math
+    base, warm, total = 1e-3, 10, 110
+    mk = lambda s: TrainConfig(learning_rate=base, lr_warmup_steps=warm,
+                               lr_scheduler=s, max_train_steps=total)
+    # linear: warmup then linear decay to 0 at total
+    cfg = mk("linear")
+    assert get_lr(cfg, 4) == pytest.approx(base * 5 / warm)
+    assert get_lr(cfg, total) == pytest.approx(0.0)
+    # cosine: half-way through post-warmup it's at base/2
+    cfg = mk("cosine")
+    mid = warm + (total - warm) // 2
+    assert get_lr(cfg, mid) == pytest.approx(base * 0.5, rel=0.05)
+    assert get_lr(cfg, total) == pytest.approx(0.0, abs=1e-6)
+    # cosine_with_restarts (1 cycle == cosine), 0 after total
+    cfg = mk("cosine_with_restarts")
+    assert get_lr(cfg, mid) == pytest.approx(base * 0.5, rel=0.05)
+    assert get_lr(cfg, total + 5) == 0.0
+    # polynomial (power=1): linear decay to lr_end=1e-7
+    cfg = mk("polynomial")
+    assert get_lr(cfg, mid) == pytest.approx((base - 1e-7) * 0.5 + 1e-7, rel=0.05)
+    assert get_lr(cfg, total) == pytest.approx(1e-7)
+    assert get_lr(cfg, total + 50) == pytest.approx(1e-7)
+    # constant: flat everywhere
+    cfg = mk("constant")
+    assert get_lr(cfg, 0) == get_lr(cfg, total) == base
+    # all schedules monotone non-increasing after warmup
+    for s in ["linear", "cosine", "cosine_with_restarts", "polynomial"]:
+        cfg = mk(s)
+        lrs = [get_lr(cfg, t) for t in range(warm, total)]
+        assert all(a >= b - 1e-12 for a, b in zip(lrs, lrs[1:])), s
+
+
 def test_fit_writes_checkpoint_layout(tmp_path):
     cfg = tiny_cfg(tmp_path, max_train_steps=2, modelsavesteps=2, save_steps=1000,
                    log_every=1)
