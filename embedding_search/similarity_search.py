@@ -23,12 +23,18 @@ from dcr_amd.search import stream_top1, dump_matches
 
 def main():
     p = argparse.ArgumentParser()
-    p.add_argument("--generation_embedding", type=str, required=True,
+    p.add_argument("--generation_embedding", "--generation-embedding-path",
+                   dest="generation_embedding", type=str, required=True,
                    help="embedding.pkl of the generated images")
-    p.add_argument("--laion_embedding_folder", type=str, required=True,
+    p.add_argument("--laion_embedding_folder", "--laion-embedding-folder",
+                   dest="laion_embedding_folder", type=str, required=True,
                    help="root dir containing chunk subdirs/pickles")
-    p.add_argument("--dump_path", type=str, required=True)
-    p.add_argument("--query_chunks", type=int, default=1)
+    p.add_argument("--dump_path", "--dump-path", dest="dump_path",
+                   type=str, required=True)
+    p.add_argument("--query_chunks", "--num-chunks", dest="query_chunks",
+                   type=int, default=1,
+                   help="split the query matrix into this many chunks "
+                        "(reference --num-chunks)")
     args = p.parse_args()
 
     with open(args.generation_embedding, "rb") as fh:
