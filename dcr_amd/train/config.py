@@ -106,12 +106,16 @@ def validate(cfg: TrainConfig):
         raise Exception("Cant train special without blip captions")
 
 
-def get_lr(cfg: TrainConfig, step: int) -> float:
+def get_lr(cfg: TrainConfig, step: int, world_size: int = 1) -> float:
     """LR schedules: all six reference choices (diff_train.py:178-189 via
     diffusers get_scheduler, called with only warmup/total steps at
     diff_train.py:504-509 — so cosine_with_restarts uses 1 cycle and
-    polynomial uses power=1, lr_end=1e-7, the library defaults)."""
+    polynomial uses power=1, lr_end=1e-7, the library defaults).
+    scale_lr multiplies the base lr by accum*batch*world
+    (diff_train.py:419-422)."""
     base = cfg.learning_rate
+    if cfg.scale_lr:
+        base *= cfg.gradient_accumulation_steps * cfg.train_batch_size * world_size
     total = cfg.max_train_steps or 1
     warm = cfg.lr_warmup_steps
     s = cfg.lr_scheduler

@@ -273,7 +273,7 @@ class Trainer:
                     self.optimizer.flat_grad.mul_(inv)
                     self.scaler.update()
                 self.optimizer.clip_grad_norm_(cfg.max_grad_norm)
-                self.optimizer.step(lr=get_lr(cfg, self.global_step))
+                self.optimizer.step(lr=get_lr(cfg, self.global_step, self.world))
                 self.optimizer.zero_grad()
             self.global_step += 1
         return loss.detach()

@@ -132,6 +132,11 @@ def test_lr_schedule_all_reference_choices():
     # constant: flat everywhere
     cfg = mk("constant")
     assert get_lr(cfg, 0) == get_lr(cfg, total) == base
+    # scale_lr multiplies base by accum * batch * world (diff_train.py:419-422)
+    cfg = TrainConfig(learning_rate=base, lr_warmup_steps=0, lr_scheduler="constant",
+                      max_train_steps=total, scale_lr=True,
+                      gradient_accumulation_steps=2, train_batch_size=4)
+    assert get_lr(cfg, 5, world_size=3) == pytest.approx(base * 2 * 4 * 3)
     # all schedules monotone non-increasing after warmup
     for s in ["linear", "cosine", "cosine_with_restarts", "polynomial"]:
         cfg = mk(s)
