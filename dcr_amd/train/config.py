@@ -16,6 +16,7 @@ class TrainConfig:
     # model
     pretrained_model_name_or_path: str = "stabilityai/stable-diffusion-2-1"
     revision: Optional[str] = None
+    tokenizer_name: Optional[str] = None    # explicit tokenizer dir (diff_train.py:371)
     unet_from_scratch: str = "no"          # 'yes' => random-init UNet (BASELINE path)
     unet_config: Optional[str] = None       # ./unet_config.json when from scratch
     model_size: str = "sd21"                # sd21 | tiny (tiny = CPU tests)

@@ -112,6 +112,9 @@ class Trainer:
             self.vae = AutoencoderKL(vcfg)
             self.text_encoder = CLIPTextModel(tcfg)
             self.tokenizer = HashTokenizer()
+        if cfg.tokenizer_name:
+            # reference diff_train.py:371-374: explicit tokenizer dir wins
+            self.tokenizer = load_tokenizer(cfg.tokenizer_name)
 
         self.vae.requires_grad_(False)
         if not self.cfg.train_text_encoder:

@@ -101,6 +101,7 @@ def main():
 
     cfg = TrainConfig(
         pretrained_model_name_or_path=args.pretrained_model_name_or_path,
+        tokenizer_name=args.tokenizer_name,
         revision=args.revision,
         unet_from_scratch=args.unet_from_scratch,
         unet_config=args.unet_config if os.path.exists(args.unet_config) else None,

@@ -18,6 +18,16 @@ def tiny_cfg(tmp_path, **kw):
     return TrainConfig(**base)
 
 
+def test_tokenizer_name_flag_wins(tmp_path):
+    """--tokenizer_name loads the tokenizer from an explicit dir
+    (reference diff_train.py:371-374)."""
+    from dcr_amd.data import HashTokenizer
+    tokdir = tmp_path / "tok"
+    HashTokenizer(model_max_length=33).save_pretrained(tokdir)
+    tr = Trainer(tiny_cfg(tmp_path, tokenizer_name=str(tokdir)))
+    assert tr.tokenizer.model_max_length == 33
+
+
 def test_loss_decreases(tmp_path):
     tr = Trainer(tiny_cfg(tmp_path))
     batch = next(iter(tr.dataloader))
