@@ -34,15 +34,21 @@ def sim_matrix(a: torch.Tensor, b: torch.Tensor,
 
 
 def einsum_in_chunks(feat_a: torch.Tensor, feat_b: torch.Tensor,
-                     chunk: int = 64) -> torch.Tensor:
-    """Patch-wise 'splitloss' similarity (reference diff_retrieval.py:643-662):
-    feats are [N, C, P] per-patch descriptors; similarity is
-    max over patch pairs of the chunked einsum('ncp,mcq->nmpq') dot."""
+                     chunk: int = 64, stype: str = "cross") -> torch.Tensor:
+    """Patch-wise 'splitloss' similarity; feats are [N, C, P] per-patch
+    descriptors. stype='cross' takes the max over all patch PAIRS
+    (reference's einsum_in_chunks intent, diff_retrieval.py:643-662,
+    'ncp,mdp->nmcd' in its layout); stype='' matches the reference's live
+    splitloss path (diff_retrieval.py:397-400): same-patch dot, max over
+    patches."""
     sims = []
     for i in range(0, feat_a.shape[0], chunk):
         a = feat_a[i:i + chunk].float()
-        s = torch.einsum("ncp,mcq->nmpq", a, feat_b.float())
-        sims.append(s.amax(dim=(2, 3)))
+        if stype == "cross":
+            s = torch.einsum("ncp,mcq->nmpq", a, feat_b.float()).amax(dim=(2, 3))
+        else:
+            s = torch.einsum("ncp,mcp->nmp", a, feat_b.float()).amax(dim=2)
+        sims.append(s)
     return torch.cat(sims, dim=0)
 
 

@@ -73,15 +73,38 @@ def parse_args():
     return p.parse_args()
 
 
+# reference arch mappings (diff_retrieval.py:250-283)
+_DINO_ARCH = {"vit_base": "dino_vitb16", "vit_base8": "dino_vitb8",
+              "vit_small": "dino_vits16"}
+_CLIP_ARCH = {"vit_large": "ViT-L/14", "vit_base": "ViT-B/16",
+              "resnet50": "RN50x16"}
+_SSCD_ARCH = {"resnet50": "sscd", "resnet50_im": "sscd_im",
+              "resnet50_disc": "sscd_disc_large"}
+
+
 def build_backbone(args, device):
-    """Backbone per --pt_style (reference diff_retrieval.py:249-285)."""
+    """Backbone per (--pt_style, --arch), reference diff_retrieval.py:249-285.
+    Our folded spellings (--pt_style dino_vits8 / sscd_im / ...) also work."""
     if args.pt_style.startswith("dino"):
-        arch = args.pt_style if args.pt_style != "dino" else "dino_vitb16"
+        if args.pt_style != "dino":            # folded spelling
+            arch = args.pt_style
+        elif args.arch in _DINO_ARCH:
+            arch = _DINO_ARCH[args.arch]
+        elif args.arch.startswith("dino_"):
+            arch = args.arch
+        else:
+            raise NotImplementedError(
+                f"dino arch {args.arch!r}: supported are {sorted(_DINO_ARCH)} "
+                "(reference's dino_resnet50/cifar10 variants are not built)")
         return load_dino(arch, weights=args.pretrained or None, device=device)
     if args.pt_style == "clip":
-        model, _ = load_clip(device=device)
+        model, _ = load_clip(_CLIP_ARCH.get(args.arch, "ViT-B/16"),
+                             device=device)
         return lambda x: model.encode_image(x)
-    return load_sscd(args.pt_style, device=device)  # sscd variants
+    # sscd: reference picks the torchscript file by --arch
+    name = _SSCD_ARCH.get(args.arch, args.pt_style) \
+        if args.pt_style == "sscd" else args.pt_style
+    return load_sscd(name, device=device)
 
 
 @torch.no_grad()
@@ -106,6 +129,10 @@ def _patch_features(model, loader, device, args):
 
 def main():
     args = parse_args()
+    if args.similarity_metric == "splitlosscross":
+        # reference alias (diff_retrieval.py:188-190)
+        args.similarity_metric = "splitloss"
+        args.stype = "cross"
     rank, world, local = init_distributed_mode()
     device = torch.device("cuda", local) if torch.cuda.is_available() \
         else torch.device("cpu")
@@ -150,9 +177,11 @@ def main():
         qp_ = _patch_features(model, loader(query_ds), device, args)
         vp_ = _patch_features(model, loader(val_ds), device, args)
         sim = einsum_in_chunks(qp_, vp_, chunk=max(1, len(query_ds) //
-                                                   max(1, args.einsum_chunks)))
+                                                   max(1, args.einsum_chunks)),
+                               stype=args.stype)
         sim_tt = einsum_in_chunks(vp_, vp_, chunk=max(1, len(val_ds) //
-                                                      max(1, args.einsum_chunks)))
+                                                      max(1, args.einsum_chunks)),
+                                  stype=args.stype)
     else:
         sim = sim_matrix(val_f, query_f).t()    # [n_gen, n_train] (rocBLAS)
         sim_tt = sim_matrix(val_f, val_f)       # [n_train, n_train]

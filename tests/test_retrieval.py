@@ -63,6 +63,33 @@ def test_einsum_in_chunks_matches_dense():
     out = einsum_in_chunks(a, b, chunk=2)
     ref = torch.einsum("ncp,mcq->nmpq", a, b).amax(dim=(2, 3))
     assert torch.allclose(out, ref, atol=1e-5)
+    # same-patch variant (reference's live splitloss, diff_retrieval.py:397-400)
+    out2 = einsum_in_chunks(a, b, chunk=2, stype="")
+    ref2 = torch.einsum("ncp,mcp->nmp", a, b).amax(dim=2)
+    assert torch.allclose(out2, ref2, atol=1e-5)
+    # cross >= same-patch everywhere (max over a superset of pairs)
+    assert (out >= out2 - 1e-5).all()
+
+
+def test_backbone_arch_mapping():
+    """Reference (pt_style, arch) pairs select the right backbone
+    (diff_retrieval.py:249-285)."""
+    import argparse
+    from diff_retrieval import build_backbone
+    def mk(**kw):
+        d = dict(pt_style="sscd", arch="resnet50", pretrained="")
+        d.update(kw)
+        return argparse.Namespace(**d)
+    m = build_backbone(mk(arch="resnet50_disc"), "cpu")  # sscd_disc_large: 1024-d
+    x = torch.randn(2, 3, 224, 224)
+    assert m(x).shape[-1] == 1024
+    assert build_backbone(mk(), "cpu")(x).shape[-1] == 512
+    d = build_backbone(mk(pt_style="dino", arch="vit_small"), "cpu")
+    assert d.patch_embed.proj.kernel_size == (16, 16)
+    d8 = build_backbone(mk(pt_style="dino", arch="vit_base8"), "cpu")
+    assert d8.patch_embed.proj.kernel_size == (8, 8)
+    with pytest.raises(NotImplementedError):
+        build_backbone(mk(pt_style="dino", arch="resnet50"), "cpu")
 
 
 def test_complexity_metrics():
