@@ -79,24 +79,29 @@ def main(args):
 
     count = 0
     for prompt in prompt_list:
-        images = pipe(prompt, height=args.resolution, width=args.resolution,
-                      num_inference_steps=50, num_images_per_prompt=args.im_batch,
-                      generator=generator).images
-        for image in images:
-            if image.size[0] > args.resolution:
-                image = resize(args.resolution, args.resolution, image)
-            image.save(f"{savepath}/generations/{count}.png")
-            count += 1
+        for _ in range(args.nbatches):
+            images = pipe(prompt, height=args.resolution, width=args.resolution,
+                          num_inference_steps=50,
+                          num_images_per_prompt=args.im_batch,
+                          generator=generator).images
+            for image in images:
+                if image.size[0] > args.resolution:
+                    image = resize(args.resolution, args.resolution, image)
+                image.save(f"{savepath}/generations/{count}.png")
+                count += 1
     print(f"wrote {count} generations to {savepath}")
 
 
 if __name__ == "__main__":
     parser = argparse.ArgumentParser()
-    parser.add_argument("--synset_map", type=str, default=None)
-    parser.add_argument("-nb", "--nbatches", type=int, default=1)
+    parser.add_argument("--synset_map", type=str, default=None,
+                        help="(reference flag; unused)")
+    parser.add_argument("-nb", "--nbatches", type=int, default=1,
+                        help="generation rounds per prompt")
     parser.add_argument("-imb", "--im_batch", type=int, default=1)
     parser.add_argument("--resolution", type=int, default=512)
-    parser.add_argument("--iternum", default=None, type=int)
+    parser.add_argument("--iternum", default=None, type=int,
+                        help="(reference flag; unused)")
     parser.add_argument("--rand_noise_lam", type=float, default=None)
     parser.add_argument("--rand_augs", type=str, default=None)
     parser.add_argument("--rand_aug_repeats", type=int, default=4)
