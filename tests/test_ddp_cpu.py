@@ -93,6 +93,56 @@ def test_trainer_ddp_world2(tmp_path):
     mp.spawn(_trainer_worker, args=(2, port, str(tmp_path)), nprocs=2, join=True)
 
 
+def _dup_sampler_worker(rank, world, port, tmpdir):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["LOCAL_RANK"] = str(rank)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from dcr_amd.train import TrainConfig, Trainer
+
+        cfg = TrainConfig(model_size="tiny", instance_data_dir=os.path.join(
+                              tmpdir, "imagenette"),
+                          class_prompt="classlevel", duplication="dup_image",
+                          resolution=64, train_batch_size=2,
+                          mixed_precision="no", dataloader_num_workers=0,
+                          max_train_steps=2, seed=0,
+                          output_dir=os.path.join(tmpdir, f"out"))
+        tr = Trainer(cfg, device=torch.device("cpu"))
+        assert getattr(tr.dataset, "samplingweights", None) is not None or \
+            hasattr(tr.dataset, "dataset"), "dup_image should set weights"
+        # per-rank WeightedRandomSampler generators must differ
+        idx = torch.tensor(list(iter(tr.dataloader.sampler))[:8])
+        gathered = [torch.empty_like(idx) for _ in range(world)]
+        dist.all_gather(gathered, idx)
+        assert not torch.equal(gathered[0], gathered[1]), \
+            "ranks drew identical duplication samples"
+        # params still sync through a step
+        batch = next(iter(tr.dataloader))
+        tr.train_step(batch)
+        flat = tr.optimizer.flat_param.clone()
+        gflat = [torch.empty_like(flat) for _ in range(world)]
+        dist.all_gather(gflat, flat)
+        assert torch.allclose(gflat[0], gflat[1], atol=1e-6)
+        dist.barrier()
+    finally:
+        dist.destroy_process_group()
+
+
+def test_dup_weighted_sampler_ddp_world2(tmp_path):
+    from PIL import Image
+    root = tmp_path / "imagenette"
+    for cls in ["church", "tench"]:
+        d = root / cls
+        d.mkdir(parents=True)
+        for i in range(6):
+            Image.new("RGB", (80, 70), color=(i * 30, 100, 50)).save(d / f"{i}.png")
+    mp.spawn(_dup_sampler_worker, args=(2, 29723, str(tmp_path)), nprocs=2,
+             join=True)
+
+
 def _feat_worker(rank, world, port):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
