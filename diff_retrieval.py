@@ -164,7 +164,9 @@ def main():
 
     out_dir = Path(args.sim_save_path)
     out_dir.mkdir(parents=True, exist_ok=True)
-    tracker = Tracker(args.project, name=Path(args.query_dir).name,
+    # run name = last 3 path components + metric (reference :378-382)
+    dp = "/".join(Path(args.query_dir).parts[-3:])
+    tracker = Tracker(args.project, name=f"{dp}_{args.similarity_metric}",
                       config=vars(args), out_dir=out_dir)
 
     query_f = l2_normalize(query_f)
