@@ -378,9 +378,9 @@ at::Tensor conv2d_nhwc_fwd(at::Tensor x, at::Tensor w,
   return y;
 }
 
-at::Tensor conv2d_nhwc_fwd_v2(at::Tensor x, at::Tensor w,
-                              c10::optional<at::Tensor> bias, int64_t stride,
-                              int64_t pad) {
+static at::Tensor conv2d_nhwc_fwd_v2v3(at::Tensor x, at::Tensor w,
+                                       c10::optional<at::Tensor> bias,
+                                       int64_t stride, int64_t pad, int ver) {
   TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16);
   TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast) &&
               w.is_contiguous(at::MemoryFormat::ChannelsLast));
@@ -413,11 +413,26 @@ at::Tensor conv2d_nhwc_fwd_v2(at::Tensor x, at::Tensor w,
     ws = at::zeros({NPQ * K}, x.options().dtype(at::kFloat));
     wsp = ws.data_ptr<float>();
   }
-  conv_nhwc_fwd_v2_launch(x.data_ptr(), w.data_ptr(), bp, y.data_ptr(), wsp,
-                          splitz, (int)Nb, (int)Hin, (int)Win, (int)C, (int)K,
-                          (int)P, (int)Q, (int)R, (int)S, (int)stride,
-                          (int)pad, cur_stream());
+  auto launch = (ver == 3) ? conv_nhwc_fwd_v3_launch : conv_nhwc_fwd_v2_launch;
+  launch(x.data_ptr(), w.data_ptr(), bp, y.data_ptr(), wsp,
+         splitz, (int)Nb, (int)Hin, (int)Win, (int)C, (int)K,
+         (int)P, (int)Q, (int)R, (int)S, (int)stride,
+         (int)pad, cur_stream());
   return y;
+}
+
+at::Tensor conv2d_nhwc_fwd_v2(at::Tensor x, at::Tensor w,
+                              c10::optional<at::Tensor> bias, int64_t stride,
+                              int64_t pad) {
+  return conv2d_nhwc_fwd_v2v3(x, w, bias, stride, pad, 2);
+}
+
+// round-2 draft: double-buffered staging — validated on hardware before
+// any dispatch (DCR_NATIVE_CONV_V3=1 gates its tests)
+at::Tensor conv2d_nhwc_fwd_v3(at::Tensor x, at::Tensor w,
+                              c10::optional<at::Tensor> bias, int64_t stride,
+                              int64_t pad) {
+  return conv2d_nhwc_fwd_v2v3(x, w, bias, stride, pad, 3);
 }
 
 // conv backward drafts (round-2; validated before any dispatch)
@@ -453,6 +468,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("conv2d_nhwc_bwd", &conv2d_nhwc_bwd);
   mod.def("conv2d_nhwc_fwd", &conv2d_nhwc_fwd);
   mod.def("conv2d_nhwc_fwd_v2", &conv2d_nhwc_fwd_v2);
+  mod.def("conv2d_nhwc_fwd_v3", &conv2d_nhwc_fwd_v3);
   mod.def("attn_fwd", &attn_fwd);
   mod.def("attn_bwd", &attn_bwd);
   mod.def("mfma_probe", &mfma_probe);

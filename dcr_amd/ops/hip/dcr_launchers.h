@@ -75,6 +75,11 @@ void conv_nhwc_fwd_v2_launch(const void* x, const void* w, const float* bias,
                              void* y, float* ws, int splitz, int Nb, int Hin,
                              int Win, int C, int K, int P, int Q, int R, int S,
                              int stride, int pad, hipStream_t st);
+// v3: double-buffered staging (round-2 draft; not dispatched)
+void conv_nhwc_fwd_v3_launch(const void* x, const void* w, const float* bias,
+                             void* y, float* ws, int splitz, int Nb, int Hin,
+                             int Win, int C, int K, int P, int Q, int R, int S,
+                             int stride, int pad, hipStream_t st);
 
 // conv_nhwc_bwd.hip (round-2 drafts; not dispatched)
 void conv_bwd_weight_launch(const void* dy, const void* x, float* dw_ws,

@@ -45,14 +45,17 @@ def main():
         bb = b.to(torch.bfloat16)
         t_v1 = timeit(lambda: m.conv2d_nhwc_fwd(x, w, b, stride, pad))
         t_v2 = timeit(lambda: m.conv2d_nhwc_fwd_v2(x, w, b, stride, pad))
+        t_v3 = timeit(lambda: m.conv2d_nhwc_fwd_v3(x, w, b, stride, pad))
         t_miopen = timeit(lambda: F.conv2d(x, w, bb, stride=stride, padding=pad))
         P = (H + 2 * pad - R) // stride + 1
         flops = 2 * N * P * P * K * C * R * R
         print(f"N{N} C{C} H{H} K{K} R{R}s{stride}: "
               f"v1 {t_v1:.3f} ms ({flops / t_v1 / 1e9:.0f} TF) "
               f"v2 {t_v2:.3f} ms ({flops / t_v2 / 1e9:.0f} TF) "
+              f"v3 {t_v3:.3f} ms ({flops / t_v3 / 1e9:.0f} TF) "
               f"MIOpen {t_miopen:.3f} ms ({flops / t_miopen / 1e9:.0f} TF) "
-              f"-> v2/MIOpen {t_miopen / t_v2:.2f}x")
+              f"-> v2/MIOpen {t_miopen / t_v2:.2f}x "
+              f"v3/v2 {t_v2 / t_v3:.2f}x")
 
 
 if __name__ == "__main__":

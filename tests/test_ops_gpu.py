@@ -389,6 +389,37 @@ def test_conv_nhwc_fwd(ext, shape, version):
     _close(y, ref, 2e-2)
 
 
+@pytest.mark.skipif(os.environ.get("DCR_NATIVE_CONV_V3") != "1",
+                    reason="conv v3 double-buffered draft: validate in "
+                           "round 2 (DCR_NATIVE_CONV_V3=1)")
+@pytest.mark.parametrize("shape", [
+    (2, 320, 32, 32, 320, 3, 1),
+    (2, 320, 32, 32, 640, 1, 1),
+    (2, 640, 16, 16, 640, 3, 2),
+    (1, 128, 64, 64, 128, 3, 1),
+    (2, 1280, 8, 8, 1280, 3, 1),    # split-K path (grid-starved)
+    (2, 96, 16, 16, 64, 3, 1),      # BK=32 instance
+])
+def test_conv_nhwc_fwd_v3(ext, shape):
+    N, C, H, W, K, R, stride = shape
+    pad = 1 if R == 3 else 0
+    torch.manual_seed(0)
+    x = torch.randn(N, C, H, W, device="cuda").to(torch.bfloat16) \
+        .to(memory_format=torch.channels_last)
+    w = (torch.randn(K, C, R, R, device="cuda") * 0.05).to(torch.bfloat16) \
+        .to(memory_format=torch.channels_last)
+    b = torch.randn(K, device="cuda")
+    y = ext.conv2d_nhwc_fwd_v3(x, w, b, stride, pad)
+    # v3 must be bit-comparable to v2 (same tile walk, same accum order
+    # within a block) and close to the fp32 reference
+    y2 = ext.conv2d_nhwc_fwd_v2(x, w, b, stride, pad)
+    ref = torch.nn.functional.conv2d(x.float(), w.float(), b, stride=stride,
+                                     padding=pad)
+    _close(y, ref, 2e-2)
+    assert torch.equal(y.float(), y2.float()) or \
+        (y.float() - y2.float()).abs().max() < 1e-2  # atomics order may differ
+
+
 @pytest.mark.skipif(os.environ.get("DCR_NATIVE_CONV_BWD") != "1",
                     reason="conv bwd drafts: validate in round 2 "
                            "(DCR_NATIVE_CONV_BWD=1)")
