@@ -236,6 +236,32 @@ void adamw_step_bf16(at::Tensor p, at::Tensor g, at::Tensor master,
                     (float)wd, step, cur_stream());
 }
 
+// device-state AdamW (round-2 draft): hyper = float[8] cuda tensor
+// {lr, b1^t, b2^t, inv_bc1, inv_bc2, clip_coef, gnorm_sq, step};
+// init {lr, 1, 1, 1, 1, 1, 0, 0}. The 3-kernel sequence is stream-ordered
+// and hipGraph-capturable; update hyper[0] (lr) from the host between
+// replays. Gated validation: DCR_DEV_ADAMW=1 tests.
+void adamw_step_dev(at::Tensor p, at::Tensor g, c10::optional<at::Tensor> master,
+                    at::Tensor m, at::Tensor v, at::Tensor hyper, double beta1,
+                    double beta2, double eps, double wd, double max_norm) {
+  TORCH_CHECK(hyper.is_cuda() && hyper.scalar_type() == at::kFloat &&
+              hyper.numel() == 8 && hyper.is_contiguous());
+  TORCH_CHECK(p.is_contiguous() && g.is_contiguous());
+  const bool bf16 = p.scalar_type() == at::kBFloat16;
+  float* mp = nullptr;
+  if (bf16) {
+    TORCH_CHECK(master.has_value() && master->scalar_type() == at::kFloat,
+                "bf16 device-state adamw needs the fp32 master");
+    mp = master->data_ptr<float>();
+  } else {
+    TORCH_CHECK(p.scalar_type() == at::kFloat);
+  }
+  adamw_dev_launch(bf16 ? 1 : 0, p.data_ptr(), g.data_ptr(), mp,
+                   m.data_ptr<float>(), v.data_ptr<float>(), p.numel(),
+                   (float)beta1, (float)beta2, (float)eps, (float)wd,
+                   (float)max_norm, hyper.data_ptr<float>(), cur_stream());
+}
+
 // ------------------------------------------------------------- scheduler math
 static at::Tensor sched_common(int mode, at::Tensor a, at::Tensor b, at::Tensor ac,
                                at::Tensor t) {
@@ -482,6 +508,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("geglu_bwd", &geglu_bwd);
   mod.def("adamw_step", &adamw_step);
   mod.def("adamw_step_bf16", &adamw_step_bf16);
+  mod.def("adamw_step_dev", &adamw_step_dev);
   mod.def("add_noise", &add_noise);
   mod.def("get_velocity", &get_velocity);
   mod.def("cfg_combine", &cfg_combine);

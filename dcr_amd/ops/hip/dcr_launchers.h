@@ -46,6 +46,11 @@ void adamw_launch(float* p, const float* g, float* m, float* v, long n,
 void adamw_bf16_launch(void* p, const void* g, float* master, float* m,
                        float* v, long n, float lr, float b1, float b2,
                        float eps, float wd, long step, hipStream_t s);
+// device-state AdamW (round-2 draft; hipGraph-capturable step)
+void adamw_dev_launch(int bf16, void* p, const void* g, float* master,
+                      float* m, float* v, long n, float b1, float b2,
+                      float eps, float wd, float max_norm, float* hyper,
+                      hipStream_t s);
 void sched_launch(DType dt, int mode, const void* x0, const void* noise,
                   const float* ac, const long* t, void* out, long per_sample,
                   long total, hipStream_t s);

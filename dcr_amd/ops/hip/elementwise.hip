@@ -238,6 +238,142 @@ DCR_INST_EW(__half)
 // ===========================================================================
 #include "dcr_launchers.h"
 
+// ---------------------------------------------------------------------------
+// Device-state AdamW (round-2 draft): the whole optimizer step — grad-norm,
+// clip coefficient, bias-correction powers, update — reads its scalars from
+// device memory so the 3-kernel sequence can be captured ONCE into a
+// hipGraph and replayed with zero host work per step (the LR schedule
+// writes hyper[0] before replay; everything else evolves on device).
+// hyper layout (float[8]):
+//   0 lr | 1 beta1^t | 2 beta2^t | 3 inv_bc1 | 4 inv_bc2
+//   5 clip_coef | 6 grad_norm_sq accumulator | 7 step count
+// init: {lr, 1, 1, 1, 1, 1, 0, 0}.
+// NOT dispatched this round — FusedAdamW keeps the host-scalar path;
+// hardware validation + trainer wiring is round 2 (DCR_DEV_ADAMW=1 tests).
+// ---------------------------------------------------------------------------
+template <typename T>
+__global__ void gnormsq_kernel(const T* __restrict__ g, long n,
+                               float* __restrict__ accum) {
+  __shared__ float lds[2 * 256 / 64];
+  float s = 0.f;
+  const long nvec = n / 4;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
+       i += (long)gridDim.x * blockDim.x) {
+    f32x4 gv = load4<T>(g + i * 4);
+    s += gv.x * gv.x + gv.y * gv.y + gv.z * gv.z + gv.w * gv.w;
+  }
+  long tail = nvec * 4 + (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (tail < n) {
+    float x = to_f32<T>(g[tail]);
+    s += x * x;
+  }
+  float2 r = block_reduce_sum2(s, 0.f, lds);
+  if (threadIdx.x == 0) atomicAdd(accum, r.x);
+}
+
+__global__ void adamw_prologue_kernel(float* __restrict__ hyper, float beta1,
+                                      float beta2, float max_norm) {
+  // one thread: advance step state, turn the norm accumulator into the
+  // clip coefficient, reset the accumulator for the next step
+  float b1t = hyper[1] * beta1;
+  float b2t = hyper[2] * beta2;
+  hyper[1] = b1t;
+  hyper[2] = b2t;
+  hyper[3] = 1.f / (1.f - b1t);
+  hyper[4] = 1.f / (1.f - b2t);
+  float gn = sqrtf(hyper[6]);
+  hyper[5] = (max_norm > 0.f && gn > max_norm) ? max_norm / (gn + 1e-6f) : 1.f;
+  hyper[6] = 0.f;
+  hyper[7] += 1.f;
+}
+
+__global__ void adamw_dev_kernel(float* __restrict__ p, const float* __restrict__ g,
+                                 float* __restrict__ m, float* __restrict__ v,
+                                 long n, float beta1, float beta2, float eps,
+                                 float wd, const float* __restrict__ hyper) {
+  const float lr = hyper[0];
+  const float inv_bc1 = hyper[3];
+  const float inv_bc2 = hyper[4];
+  const float clip = hyper[5];
+  const long nvec = n / 4;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
+       i += (long)gridDim.x * blockDim.x) {
+    long e = i * 4;
+    f32x4 pv = load4<float>(p + e);
+    f32x4 gv = load4<float>(g + e);
+    f32x4 mv = load4<float>(m + e);
+    f32x4 vv = load4<float>(v + e);
+#pragma unroll
+    for (int k = 0; k < 4; ++k) {
+      float gg = (&gv.x)[k] * clip;
+      float mm = beta1 * (&mv.x)[k] + (1.f - beta1) * gg;
+      float vvk = beta2 * (&vv.x)[k] + (1.f - beta2) * gg * gg;
+      (&mv.x)[k] = mm;
+      (&vv.x)[k] = vvk;
+      float denom = sqrtf(vvk * inv_bc2) + eps;
+      (&pv.x)[k] -= lr * ((mm * inv_bc1) / denom + wd * (&pv.x)[k]);
+    }
+    store4<float>(p + e, pv);
+    store4<float>(m + e, mv);
+    store4<float>(v + e, vv);
+  }
+  long tail = nvec * 4 + (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (tail < n) {
+    float gg = g[tail] * clip;
+    float mm = beta1 * m[tail] + (1.f - beta1) * gg;
+    float vvk = beta2 * v[tail] + (1.f - beta2) * gg * gg;
+    m[tail] = mm; v[tail] = vvk;
+    float denom = sqrtf(vvk * inv_bc2) + eps;
+    p[tail] -= lr * ((mm * inv_bc1) / denom + wd * p[tail]);
+  }
+}
+
+__global__ void adamw_bf16_dev_kernel(__hip_bfloat16* __restrict__ p,
+                                      const __hip_bfloat16* __restrict__ g,
+                                      float* __restrict__ master,
+                                      float* __restrict__ m, float* __restrict__ v,
+                                      long n, float beta1, float beta2, float eps,
+                                      float wd, const float* __restrict__ hyper) {
+  const float lr = hyper[0];
+  const float inv_bc1 = hyper[3];
+  const float inv_bc2 = hyper[4];
+  const float clip = hyper[5];
+  const long nvec = n / 4;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
+       i += (long)gridDim.x * blockDim.x) {
+    long e = i * 4;
+    f32x4 gv = load4<__hip_bfloat16>(g + e);
+    f32x4 pv = load4<float>(master + e);
+    f32x4 mv = load4<float>(m + e);
+    f32x4 vv = load4<float>(v + e);
+#pragma unroll
+    for (int kk = 0; kk < 4; ++kk) {
+      float gg = (&gv.x)[kk] * clip;
+      float mm = beta1 * (&mv.x)[kk] + (1.f - beta1) * gg;
+      float vvk = beta2 * (&vv.x)[kk] + (1.f - beta2) * gg * gg;
+      (&mv.x)[kk] = mm;
+      (&vv.x)[kk] = vvk;
+      float denom = sqrtf(vvk * inv_bc2) + eps;
+      (&pv.x)[kk] -= lr * ((mm * inv_bc1) / denom + wd * (&pv.x)[kk]);
+    }
+    store4<float>(master + e, pv);
+    store4<float>(m + e, mv);
+    store4<float>(v + e, vv);
+    store4<__hip_bfloat16>(p + e, pv);
+  }
+  long tail = nvec * 4 + (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (tail < n) {
+    float gg = to_f32<__hip_bfloat16>(g[tail]) * clip;
+    float mm = beta1 * m[tail] + (1.f - beta1) * gg;
+    float vvk = beta2 * v[tail] + (1.f - beta2) * gg * gg;
+    m[tail] = mm; v[tail] = vvk;
+    float denom = sqrtf(vvk * inv_bc2) + eps;
+    float pv = master[tail] - lr * ((mm * inv_bc1) / denom + wd * master[tail]);
+    master[tail] = pv;
+    p[tail] = from_f32<__hip_bfloat16>(pv);
+  }
+}
+
 namespace dcr {
 
 static inline int ew_blocks(long nvec, int block = 256) {
@@ -286,6 +422,30 @@ void adamw_bf16_launch(void* p, const void* g, float* master, float* m,
   hipLaunchKernelGGL(adamw_bf16_kernel, grid, block, 0, s,
                      (__hip_bfloat16*)p, (const __hip_bfloat16*)g, master, m, v,
                      n, lr, b1, b2, eps, wd, 1.f / bc1, 1.f / bc2);
+}
+
+// device-state AdamW (round-2 draft): norm -> prologue -> update on one
+// stream; every scalar that changes per step lives in hyper[8] on device
+void adamw_dev_launch(int bf16, void* p, const void* g, float* master,
+                      float* m, float* v, long n, float b1, float b2,
+                      float eps, float wd, float max_norm, float* hyper,
+                      hipStream_t s) {
+  dim3 grid(ew_blocks(n / 4)), block(256);
+  if (bf16)
+    hipLaunchKernelGGL((gnormsq_kernel<__hip_bfloat16>), grid, block, 0, s,
+                       (const __hip_bfloat16*)g, n, hyper + 6);
+  else
+    hipLaunchKernelGGL((gnormsq_kernel<float>), grid, block, 0, s,
+                       (const float*)g, n, hyper + 6);
+  hipLaunchKernelGGL(adamw_prologue_kernel, dim3(1), dim3(1), 0, s, hyper,
+                     b1, b2, max_norm);
+  if (bf16)
+    hipLaunchKernelGGL(adamw_bf16_dev_kernel, grid, block, 0, s,
+                       (__hip_bfloat16*)p, (const __hip_bfloat16*)g, master,
+                       m, v, n, b1, b2, eps, wd, hyper);
+  else
+    hipLaunchKernelGGL(adamw_dev_kernel, grid, block, 0, s, (float*)p,
+                       (const float*)g, m, v, n, b1, b2, eps, wd, hyper);
 }
 
 void sched_launch(DType dt, int mode, const void* x0, const void* noise,

@@ -389,6 +389,78 @@ def test_conv_nhwc_fwd(ext, shape, version):
     _close(y, ref, 2e-2)
 
 
+@pytest.mark.skipif(os.environ.get("DCR_DEV_ADAMW") != "1",
+                    reason="device-state AdamW draft: validate in round 2 "
+                           "(DCR_DEV_ADAMW=1)")
+@pytest.mark.parametrize("bf16", [False, True])
+def test_adamw_dev_matches_host(ext, bf16):
+    """3-kernel device-state step == host-scalar clip+step over 6 steps,
+    including one step where the clip triggers."""
+    n = 10_001
+    torch.manual_seed(0)
+    lr, b1, b2, eps, wd, mx = 1e-3, 0.9, 0.999, 1e-8, 1e-2, 1.0
+    master1 = torch.randn(n, device="cuda")
+    master2 = master1.clone()
+    if bf16:
+        p1 = master1.bfloat16()
+        p2 = master2.bfloat16()
+    else:
+        p1, p2 = master1, master2
+    m1 = torch.zeros(n, device="cuda"); v1 = torch.zeros(n, device="cuda")
+    m2 = m1.clone(); v2 = v1.clone()
+    hyper = torch.tensor([lr, 1, 1, 1, 1, 1, 0, 0], device="cuda")
+    for step in range(1, 7):
+        g = torch.randn(n, device="cuda") * (10.0 if step == 3 else 0.01)
+        gg = g.bfloat16() if bf16 else g
+        # host path: explicit clip, then the host-scalar kernel
+        total = gg.float().norm()
+        coef = torch.clamp(mx / (total + 1e-6), max=1.0)
+        gh = (gg.float() * coef).to(gg.dtype)
+        if bf16:
+            ext.adamw_step_bf16(p1, gh, master1, m1, v1, lr, b1, b2, eps, wd, step)
+            ext.adamw_step_dev(p2, gg, master2, m2, v2, hyper, b1, b2, eps, wd, mx)
+        else:
+            ext.adamw_step(p1, gh, m1, v1, lr, b1, b2, eps, wd, step)
+            ext.adamw_step_dev(p2, gg, None, m2, v2, hyper, b1, b2, eps, wd, mx)
+    assert int(hyper[7].item()) == 6
+    # atomic-order norm + bf16 clip rounding leave tiny differences
+    tol = 2e-2 if bf16 else 1e-4
+    _close(master2, master1, tol)
+    _close(m2, m1, 2e-2 if bf16 else 1e-4)
+
+
+@pytest.mark.skipif(os.environ.get("DCR_DEV_ADAMW") != "1",
+                    reason="device-state AdamW draft: validate in round 2 "
+                           "(DCR_DEV_ADAMW=1)")
+def test_adamw_dev_graph_capture(ext):
+    """The whole optimizer step replays from a hipGraph with zero host work;
+    state (step counter, bias-correction powers) advances on device."""
+    n = 4096
+    torch.manual_seed(0)
+    p = torch.randn(n, device="cuda")
+    g = torch.randn(n, device="cuda") * 0.01
+    m = torch.zeros(n, device="cuda")
+    v = torch.zeros(n, device="cuda")
+    hyper = torch.tensor([1e-3, 1, 1, 1, 1, 1, 0, 0], device="cuda")
+    args = (p, g, None, m, v, hyper, 0.9, 0.999, 1e-8, 1e-2, 1.0)
+    # warmup on a side stream (capture requirement), then capture one step
+    s = torch.cuda.Stream()
+    s.wait_stream(torch.cuda.current_stream())
+    with torch.cuda.stream(s):
+        ext.adamw_step_dev(*args)
+    torch.cuda.current_stream().wait_stream(s)
+    steps_before = int(hyper[7].item())
+    graph = torch.cuda.CUDAGraph()
+    with torch.cuda.graph(graph):
+        ext.adamw_step_dev(*args)
+    for _ in range(3):
+        graph.replay()
+    torch.cuda.synchronize()
+    # warmup + capture + 3 replays all advanced the device step counter
+    assert int(hyper[7].item()) >= steps_before + 3
+    assert torch.isfinite(p).all()
+
+
 @pytest.mark.skipif(os.environ.get("DCR_NATIVE_CONV_V3") != "1",
                     reason="conv v3 double-buffered draft: validate in "
                            "round 2 (DCR_NATIVE_CONV_V3=1)")
