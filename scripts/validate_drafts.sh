@@ -1,0 +1,21 @@
+#!/bin/bash
+# One-gpurun-call hardware validation of every round-2 draft kernel.
+# Round-2 session starts with:
+#   /usr/local/graft/bin/gpurun --timeout 900 -- 'bash scripts/validate_drafts.sh 2>&1 | tee gpurun_out/drafts.log'
+# Then iterate on whatever fails, exactly like the conv-fwd kernel went
+# design -> validated -> beats-MIOpen in four short calls.
+set -ex
+cd "$(dirname "$0")/.."
+
+# conv fwd v3: double-buffered staging (expect >= v2; guide says +40% in
+# this occupancy regime if the pipeline holds)
+DCR_NATIVE_CONV_V3=1 python -m pytest tests/test_ops_gpu.py -k fwd_v3 -x -q
+
+# conv backward drafts (bwd-weight pixel-split, bwd-data tap loop)
+DCR_NATIVE_CONV_BWD=1 python -m pytest tests/test_ops_gpu.py -k nhwc_bwd -x -q
+
+# device-state AdamW + hipGraph capture
+DCR_DEV_ADAMW=1 python -m pytest tests/test_ops_gpu.py -k adamw_dev -x -q
+
+# perf A/B (v3/v2 ratio printed per shape)
+python scripts/bench_conv.py
