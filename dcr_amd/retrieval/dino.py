@@ -148,10 +148,31 @@ _DINO_FACTORY = {
 }
 
 
+class ResNetDINO(nn.Module):
+    """DINO ResNet-50 backbone (reference dino_vits.py:416-431: torchvision
+    resnet50 with fc=Identity): 2048-d global-average-pooled features.
+    Reuses this framework's ResNet50 trunk."""
+
+    def __init__(self):
+        super().__init__()
+        from .backbones import ResNet50
+        self.backbone = ResNet50()
+
+    def forward(self, x):
+        return self.backbone.forward_features(x).mean(dim=(2, 3))
+
+
 def load_dino(arch: str = "dino_vitb16", weights: Optional[str] = None,
               device="cpu"):
     """Factory mirroring dino_vits.py:340-487 hub loaders; local weights
     (state dict .pth) are loaded when provided/found, else random init."""
+    if arch == "dino_resnet50":
+        model = ResNetDINO()
+        cand = Path(weights) if weights else Path("./pretrainedmodels") / f"{arch}.pth"
+        if cand.exists():
+            sd = torch.load(str(cand), map_location="cpu", weights_only=True)
+            model.backbone.load_state_dict(sd, strict=False)
+        return model.to(device).eval()
     fac, ps = _DINO_FACTORY.get(arch, _DINO_FACTORY["dino_vitb16"])
     model = fac(patch_size=ps)
     cand = Path(weights) if weights else Path("./pretrainedmodels") / f"{arch}.pth"

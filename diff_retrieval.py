@@ -75,7 +75,7 @@ def parse_args():
 
 # reference arch mappings (diff_retrieval.py:250-283)
 _DINO_ARCH = {"vit_base": "dino_vitb16", "vit_base8": "dino_vitb8",
-              "vit_small": "dino_vits16"}
+              "vit_small": "dino_vits16", "resnet50": "dino_resnet50"}
 _CLIP_ARCH = {"vit_large": "ViT-L/14", "vit_base": "ViT-B/16",
               "resnet50": "RN50x16"}
 _SSCD_ARCH = {"resnet50": "sscd", "resnet50_im": "sscd_im",
@@ -95,7 +95,7 @@ def build_backbone(args, device):
         else:
             raise NotImplementedError(
                 f"dino arch {args.arch!r}: supported are {sorted(_DINO_ARCH)} "
-                "(reference's dino_resnet50/cifar10 variants are not built)")
+                "(reference's cifar10 variant is not built)")
         return load_dino(arch, weights=args.pretrained or None, device=device)
     if args.pt_style == "clip":
         model, _ = load_clip(_CLIP_ARCH.get(args.arch, "ViT-B/16"),

@@ -88,8 +88,10 @@ def test_backbone_arch_mapping():
     assert d.patch_embed.proj.kernel_size == (16, 16)
     d8 = build_backbone(mk(pt_style="dino", arch="vit_base8"), "cpu")
     assert d8.patch_embed.proj.kernel_size == (8, 8)
+    rn = build_backbone(mk(pt_style="dino", arch="resnet50"), "cpu")
+    assert rn(torch.randn(1, 3, 64, 64)).shape == (1, 2048)
     with pytest.raises(NotImplementedError):
-        build_backbone(mk(pt_style="dino", arch="resnet50"), "cpu")
+        build_backbone(mk(pt_style="dino", arch="vit_base_cifar10"), "cpu")
 
 
 def test_complexity_metrics():
