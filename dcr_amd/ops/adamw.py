@@ -33,12 +33,15 @@ class FusedAdamW:
         betas=(0.9, 0.999),
         eps: float = 1e-8,
         weight_decay: float = 1e-2,
+        device_state: bool = False,
+        max_grad_norm: float = 1.0,
     ):
         self.lr = lr
         self.beta1, self.beta2 = betas
         self.eps = eps
         self.weight_decay = weight_decay
         self.step_count = 0
+        self.max_grad_norm = max_grad_norm
 
         self.params: List[torch.nn.Parameter] = [p for p in params if p.requires_grad]
         if not self.params:
@@ -72,6 +75,15 @@ class FusedAdamW:
         self.numel = total
         if self.master is not None:
             self.master.copy_(self.flat_param.float())
+
+        # device-state mode (round-2 draft, DCR_DEV_ADAMW=1): all per-step
+        # scalars live in hyper[8] on device so the whole optimizer tail is
+        # hipGraph-capturable — see elementwise.hip adamw_dev section.
+        # hyper = {lr, b1^t, b2^t, inv_bc1, inv_bc2, clip_coef, gnorm_sq, t}
+        self.hyper = None
+        if device_state and device.type == "cuda":
+            self.hyper = torch.tensor([lr, 1, 1, 1, 1, 1, 0, 0],
+                                      dtype=torch.float32, device=device)
 
     # -- torch.optim-ish surface ------------------------------------------
     def zero_grad(self, set_to_none: bool = False):
@@ -117,6 +129,24 @@ class FusedAdamW:
             )
 
     @torch.no_grad()
+    def step_dev(self, lr: float | None = None):
+        """Device-state step (clip INCLUDED — do not call clip_grad_norm_
+        separately): 3 stream-ordered kernels reading every scalar from the
+        hyper buffer; capturable into a hipGraph. The only host work is the
+        4-byte lr write when the schedule moves."""
+        assert self.hyper is not None, "built without device_state=True"
+        if lr is not None and lr != self.lr:
+            self.lr = lr
+            self.hyper[0].fill_(lr)  # outside any captured graph
+        self.step_count += 1
+        m = require_hip("adamw")
+        count_dispatch('adamw_dev')
+        m.adamw_step_dev(self.flat_param, self.flat_grad, self.master,
+                         self.exp_avg, self.exp_avg_sq, self.hyper,
+                         self.beta1, self.beta2, self.eps, self.weight_decay,
+                         self.max_grad_norm)
+
+    @torch.no_grad()
     def clip_grad_norm_(self, max_norm: float) -> torch.Tensor:
         """Global L2 grad clip (reference: diff_train.py:657-663, max 1.0)."""
         norm = torch.linalg.vector_norm(self.flat_grad.float())
@@ -136,6 +166,8 @@ class FusedAdamW:
         }
         if self.master is not None:
             d["master"] = self.master
+        if self.hyper is not None:
+            d["hyper"] = self.hyper
         return d
 
     def load_state_dict(self, sd):
@@ -144,6 +176,8 @@ class FusedAdamW:
         self.exp_avg.copy_(sd["exp_avg"])
         self.exp_avg_sq.copy_(sd["exp_avg_sq"])
         self.flat_param.copy_(sd["flat_param"])
+        if self.hyper is not None and "hyper" in sd:
+            self.hyper.copy_(sd["hyper"])
         if self.master is not None:
             if "master" in sd:
                 self.master.copy_(sd["master"])

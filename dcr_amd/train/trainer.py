@@ -19,6 +19,7 @@ MI355X design deltas vs the reference:
 """
 from __future__ import annotations
 
+import os
 import random
 import time
 from pathlib import Path
@@ -209,7 +210,9 @@ class Trainer:
             params += list(self.text_encoder.parameters())
         self.optimizer = FusedAdamW(
             params, lr=lr, betas=(cfg.adam_beta1, cfg.adam_beta2),
-            eps=cfg.adam_epsilon, weight_decay=cfg.adam_weight_decay)
+            eps=cfg.adam_epsilon, weight_decay=cfg.adam_weight_decay,
+            device_state=os.environ.get("DCR_DEV_ADAMW") == "1",
+            max_grad_norm=cfg.max_grad_norm)
         self.ddp = GradBucketAllReduce(self.optimizer, bucket_mb=cfg.ddp_bucket_mb)
 
     # ------------------------------------------------------------------
@@ -275,8 +278,13 @@ class Trainer:
                     inv = 1.0 / self.scaler.get_scale()
                     self.optimizer.flat_grad.mul_(inv)
                     self.scaler.update()
-                self.optimizer.clip_grad_norm_(cfg.max_grad_norm)
-                self.optimizer.step(lr=get_lr(cfg, self.global_step, self.world))
+                lr_step = get_lr(cfg, self.global_step, self.world)
+                if self.optimizer.hyper is not None:
+                    # device-state path: clip fused into the kernel sequence
+                    self.optimizer.step_dev(lr=lr_step)
+                else:
+                    self.optimizer.clip_grad_norm_(cfg.max_grad_norm)
+                    self.optimizer.step(lr=lr_step)
                 self.optimizer.zero_grad()
             self.global_step += 1
         return loss.detach()
