@@ -14,8 +14,9 @@ DCR_NATIVE_CONV_V3=1 python -m pytest tests/test_ops_gpu.py -k fwd_v3 -x -q
 # conv backward drafts (bwd-weight pixel-split, bwd-data tap loop)
 DCR_NATIVE_CONV_BWD=1 python -m pytest tests/test_ops_gpu.py -k nhwc_bwd -x -q
 
-# device-state AdamW + hipGraph capture
+# device-state AdamW + hipGraph capture + whole-train-step wiring
 DCR_DEV_ADAMW=1 python -m pytest tests/test_ops_gpu.py -k adamw_dev -x -q
+DCR_DEV_ADAMW=1 python -m pytest tests/test_train_gpu.py -k device_state -x -q
 
 # perf A/B (v3/v2 ratio printed per shape)
 python scripts/bench_conv.py

@@ -19,6 +19,27 @@ def test_tiny_train_step_gpu(tmp_path):
     assert losses[-1] < losses[0], losses
 
 
+@pytest.mark.skipif(__import__("os").environ.get("DCR_DEV_ADAMW") != "1",
+                    reason="device-state AdamW draft: validate in round 2 "
+                           "(DCR_DEV_ADAMW=1)")
+def test_tiny_train_step_device_state_adamw(tmp_path, monkeypatch):
+    """Whole train step through the device-state optimizer path: hyper
+    advances on device and the loss still goes down."""
+    from dcr_amd.train import TrainConfig, Trainer
+    cfg = TrainConfig(model_size="tiny", synthetic_data=True, synthetic_size=8,
+                      resolution=64, train_batch_size=2, mixed_precision="bf16",
+                      dataloader_num_workers=0, max_train_steps=4, seed=0,
+                      learning_rate=1e-4, lr_warmup_steps=1,
+                      output_dir=str(tmp_path / "out"))
+    tr = Trainer(cfg, device=torch.device("cuda", 0))
+    assert tr.optimizer.hyper is not None
+    batch = next(iter(tr.dataloader))
+    losses = [tr.train_step(batch).item() for _ in range(6)]
+    torch.cuda.synchronize()
+    assert int(tr.optimizer.hyper[7].item()) == 6  # device step counter
+    assert losses[-1] < losses[0], losses
+
+
 def test_sd21_train_step_gpu(tmp_path):
     """One full-size flagship step must run and produce a finite loss."""
     from dcr_amd.train import TrainConfig, Trainer
