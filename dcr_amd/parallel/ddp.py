@@ -51,8 +51,13 @@ class GradBucketAllReduce:
         if not self._enabled:
             return
 
-        # broadcast initial flat params so every rank starts identical
+        # broadcast initial flat params so every rank starts identical;
+        # in pure-bf16 mode the fp32 master was cloned from the PRE-broadcast
+        # rank-local weights (per-rank init seeds) and is the source of truth
+        # for every update — refresh it or ranks diverge from step 1
         dist.broadcast(self.opt.flat_param, src=0, group=self.pg)
+        if self.opt.master is not None:
+            self.opt.master.copy_(self.opt.flat_param.float())
 
         bucket_elems = int(bucket_mb * 1024 * 1024 /
                            self.opt.flat_grad.element_size())

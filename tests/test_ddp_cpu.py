@@ -93,6 +93,46 @@ def test_trainer_ddp_world2(tmp_path):
     mp.spawn(_trainer_worker, args=(2, port, str(tmp_path)), nprocs=2, join=True)
 
 
+def _bf16_master_worker(rank, world, port, tmpdir):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["LOCAL_RANK"] = str(rank)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from dcr_amd.train import TrainConfig, Trainer
+
+        cfg = TrainConfig(model_size="tiny", synthetic_data=True, synthetic_size=4,
+                          resolution=64, train_batch_size=2,
+                          mixed_precision="pure_bf16",
+                          dataloader_num_workers=0, max_train_steps=2, seed=0,
+                          output_dir=os.path.join(tmpdir, "out"))
+        tr = Trainer(cfg, device=torch.device("cpu"))
+        assert tr.optimizer.master is not None
+        # per-rank init seeds diverge the raw init; the ddp broadcast must
+        # sync BOTH flat_param and the fp32 master (source of truth)
+        mst = tr.optimizer.master.clone()
+        gathered = [torch.empty_like(mst) for _ in range(world)]
+        dist.all_gather(gathered, mst)
+        assert torch.equal(gathered[0], gathered[1]), \
+            "fp32 master diverged across ranks after init broadcast"
+        batch = next(iter(tr.dataloader))
+        tr.train_step(batch)
+        mst = tr.optimizer.master.clone()
+        dist.all_gather(gathered, mst)
+        assert torch.allclose(gathered[0], gathered[1], atol=1e-7), \
+            "fp32 master diverged after one pure-bf16 DDP step"
+        dist.barrier()
+    finally:
+        dist.destroy_process_group()
+
+
+def test_pure_bf16_master_sync_world2(tmp_path):
+    mp.spawn(_bf16_master_worker, args=(2, 29725, str(tmp_path)), nprocs=2,
+             join=True)
+
+
 def _dup_sampler_worker(rank, world, port, tmpdir):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
