@@ -274,17 +274,29 @@ class Trainer:
         if sync_gradients:
             with self.prof.phase("optimizer"):
                 self.ddp.finalize()
+                grads_finite = True
                 if self.scaler is not None:
+                    # grads were all-reduced BEFORE unscaling, so an overflow
+                    # on any rank propagates to every rank -> consistent skip
                     inv = 1.0 / self.scaler.get_scale()
                     self.optimizer.flat_grad.mul_(inv)
-                    self.scaler.update()
-                lr_step = get_lr(cfg, self.global_step, self.world)
-                if self.optimizer.hyper is not None:
-                    # device-state path: clip fused into the kernel sequence
-                    self.optimizer.step_dev(lr=lr_step)
+                    grads_finite = bool(
+                        torch.isfinite(self.optimizer.flat_grad).all())
+                if grads_finite:
+                    lr_step = get_lr(cfg, self.global_step, self.world)
+                    if self.optimizer.hyper is not None:
+                        # device-state path: clip fused into the kernels
+                        self.optimizer.step_dev(lr=lr_step)
+                    else:
+                        self.optimizer.clip_grad_norm_(cfg.max_grad_norm)
+                        self.optimizer.step(lr=lr_step)
+                    if self.scaler is not None:
+                        self.scaler.update()  # success: growth tracking
                 else:
-                    self.optimizer.clip_grad_norm_(cfg.max_grad_norm)
-                    self.optimizer.step(lr=lr_step)
+                    # fp16 overflow: skip the update, back the scale off
+                    # (GradScaler semantics; reference utils_ret.py:834-860
+                    # relies on torch's internal version of this)
+                    self.scaler.update(self.scaler.get_scale() * 0.5)
                 self.optimizer.zero_grad()
             self.global_step += 1
         return loss.detach()

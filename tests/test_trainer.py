@@ -28,6 +28,43 @@ def test_tokenizer_name_flag_wins(tmp_path):
     assert tr.tokenizer.model_max_length == 33
 
 
+def test_fp16_overflow_skips_step(tmp_path):
+    """fp16 scaler path: non-finite unscaled grads must skip the optimizer
+    update and back the scale off; finite grads step and call update()."""
+    class _FakeScaler:
+        def __init__(self, scale):
+            self.scale_val = scale
+            self.updates = []
+
+        def scale(self, loss):
+            return loss * self.scale_val
+
+        def get_scale(self):
+            return self.scale_val
+
+        def update(self, new_scale=None):
+            self.updates.append(new_scale)
+            if new_scale is not None:
+                self.scale_val = new_scale
+
+    tr = Trainer(tiny_cfg(tmp_path), device=torch.device("cpu"))
+    batch = next(iter(tr.dataloader))
+    # finite path
+    tr.scaler = _FakeScaler(2.0)
+    p0 = tr.optimizer.flat_param.clone()
+    tr.train_step(batch)
+    assert not torch.equal(tr.optimizer.flat_param, p0)
+    assert tr.scaler.updates == [None]
+    # overflow path: tiny scale -> inv=1e300 -> inf grads -> skip + backoff
+    tr.scaler = _FakeScaler(1e-300)
+    p1 = tr.optimizer.flat_param.clone()
+    step_before = tr.optimizer.step_count
+    tr.train_step(batch)
+    assert torch.equal(tr.optimizer.flat_param, p1), "step not skipped"
+    assert tr.optimizer.step_count == step_before
+    assert tr.scaler.updates == [pytest.approx(5e-301)]  # 0.5x backoff
+
+
 def test_loss_decreases(tmp_path):
     tr = Trainer(tiny_cfg(tmp_path))
     batch = next(iter(tr.dataloader))
