@@ -313,6 +313,32 @@ def test_retrieval_cli_torchrun_world2(tmp_path):
     assert abs(single["bg_mean"] - multi["bg_mean"]) < 1e-4
 
 
+def test_inference_cli_torchrun_world2(tmp_path):
+    """rank-sharded generation (diff_inference under torchrun, 2 ranks):
+    disjoint batch slices, global numbering, complete output set."""
+    import os as _os
+    ckpt_root = tmp_path / "run_imagenette_classlevel_nodup"
+    pipe = tiny_pipe()
+    pipe.save_pretrained(ckpt_root / "checkpoint")
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29539",
+         str(Path(__file__).parent.parent / "diff_inference.py"),
+         "--modelpath", str(ckpt_root),
+         "-nb", "4", "-imb", "1", "--resolution", "64", "--seed", "0"],
+        capture_output=True, text=True, cwd=str(tmp_path),
+        env={**_os.environ, "PYTHONPATH": str(Path(__file__).parent.parent),
+             "MASTER_ADDR": "127.0.0.1"},
+        timeout=870)
+    assert r.returncode == 0, r.stderr[-3000:]
+    gen = tmp_path / "inferences" / "imagenette10_frozentext" / \
+        "run_imagenette_classlevel_nodup" / "classlevel"
+    assert (gen / "prompts.txt").exists(), r.stdout[-2000:]
+    pngs = sorted((gen / "generations").glob("*.png"))
+    assert [p.name for p in pngs] == ["0.png", "1.png", "2.png", "3.png"]
+
+
 def test_pipeline_no_cfg_path():
     """guidance_scale <= 1 skips the CFG double batch."""
     pipe = tiny_pipe()
