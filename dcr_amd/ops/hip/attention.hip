@@ -226,6 +226,147 @@ void attn_fwd_kernel(const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
 // ==========================================================================
 // Backward preprocess: delta[row] = sum_d dO[row,d]*O[row,d]
 // ==========================================================================
+// ==========================================================================
+// Forward v2 (round-2 draft): identical structure + masked-tail MFMA skip.
+// For the last key tile (and the whole pass at Lk<=64), every 16-key
+// ns-subtile beyond ceil(valid/16) is fully masked: its S values are
+// -inf before the exp, so its sP columns are zeros and its PV products
+// contribute nothing. Skipping those MFMAs is therefore BIT-EXACT vs v1
+// (fp32 accumulation of exact zeros) while saving 6 of 8 S-MFMAs and 4
+// of 8 PV-MFMAs on the Lk=77 cross-attention tail tile (13 valid keys).
+// NOT dispatched this round — gated tests assert torch.equal vs v1
+// (DCR_ATTN_V2=1), A/B in scripts/bench_attention.py.
+// ==========================================================================
+__global__ __launch_bounds__(256)
+void attn_fwd_v2_kernel(const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
+                        const bf16_t* __restrict__ v, bf16_t* __restrict__ o,
+                        float* __restrict__ lse, int Lq, int Lk, int H,
+                        float scale, int causal) {
+  __shared__ short sQ[TILE * PITCH];
+  __shared__ short sK[TILE * PITCH];
+  __shared__ short sVT[TILE * PITCH];
+  __shared__ short sP[TILE * PITCH];
+
+  const int bh = blockIdx.y;
+  const int b = bh / H, h = bh % H;
+  const int q0 = blockIdx.x * TILE;
+  const long qrs = (long)H * DHEAD;
+  const bf16_t* qp = q + (((long)b * Lq + q0) * H + h) * DHEAD;
+  const bf16_t* kp = k + ((long)b * Lk * H + h) * DHEAD;
+  const bf16_t* vp = v + ((long)b * Lk * H + h) * DHEAD;
+  bf16_t* op = o + ((long)b * Lq * H + h) * DHEAD;
+
+  load_tile(qp, qrs, Lq - q0, sQ);
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int l16 = lane & 15;
+  const int kgrp = lane >> 4;
+  const int wrow0 = wid * 16;
+
+  float row_max[4] = {-1e30f, -1e30f, -1e30f, -1e30f};
+  float row_sum[4] = {0.f, 0.f, 0.f, 0.f};
+  f32x4_t acc_o[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) acc_o[i] = {0.f, 0.f, 0.f, 0.f};
+
+  const int kv_end = causal ? min(Lk, q0 + TILE) : Lk;
+
+  for (int kv0 = 0; kv0 < kv_end; kv0 += TILE) {
+    __syncthreads();
+    load_tile(kp + (long)kv0 * qrs, qrs, Lk - kv0, sK);
+    load_tile_T(vp + (long)kv0 * qrs, qrs, Lk - kv0, sVT);
+    __syncthreads();
+
+    // number of 16-key subtiles with any valid key in this tile
+    const int kv_valid = min(TILE, Lk - kv0);
+    const int nv = (kv_valid + 15) >> 4;
+
+    bf16x8 qf0 = frag(sQ, wrow0 + l16, kgrp * 8);
+    bf16x8 qf1 = frag(sQ, wrow0 + l16, kgrp * 8 + 32);
+    f32x4_t s_frag[4];
+#pragma unroll
+    for (int ns = 0; ns < 4; ++ns) {
+      f32x4_t acc = {0.f, 0.f, 0.f, 0.f};
+      if (ns < nv) {
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            qf0, frag(sK, ns * 16 + l16, kgrp * 8), acc, 0, 0, 0);
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            qf1, frag(sK, ns * 16 + l16, kgrp * 8 + 32), acc, 0, 0, 0);
+      }
+      s_frag[ns] = acc;
+    }
+
+    float tile_max[4] = {-1e30f, -1e30f, -1e30f, -1e30f};
+#pragma unroll
+    for (int ns = 0; ns < 4; ++ns) {
+      const int key = kv0 + ns * 16 + l16;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float sv = s_frag[ns][r] * scale;
+        const int qrow = q0 + wrow0 + kgrp * 4 + r;
+        if (key >= Lk || (causal && key > qrow)) sv = -1e30f;
+        s_frag[ns][r] = sv;
+        tile_max[r] = fmaxf(tile_max[r], sv);
+      }
+    }
+    float alpha[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      float tm = qmax(tile_max[r]);
+      float mnew = fmaxf(row_max[r], tm);
+      alpha[r] = (mnew <= -1e29f) ? 1.f : __expf(row_max[r] - mnew);
+      row_max[r] = mnew;
+    }
+
+    float psum[4] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int ns = 0; ns < 4; ++ns) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float pv = (s_frag[ns][r] <= -1e29f)
+                       ? 0.f : __expf(s_frag[ns][r] - row_max[r]);
+        psum[r] += pv;
+        sP[(wrow0 + kgrp * 4 + r) * PITCH + ns * 16 + l16] = f2bf_rne(pv);
+      }
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) row_sum[r] = row_sum[r] * alpha[r] + qsum(psum[r]);
+
+#pragma unroll
+    for (int ds_ = 0; ds_ < 4; ++ds_)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) acc_o[ds_][r] *= alpha[r];
+
+    // PV: the high 32 keys of the tile are all invalid when nv <= 2 —
+    // their sP columns are zeros, skip the second MFMA pair entirely
+    bf16x8 pf0 = frag(sP, wrow0 + l16, kgrp * 8);
+#pragma unroll
+    for (int ds_ = 0; ds_ < 4; ++ds_)
+      acc_o[ds_] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          pf0, frag(sVT, ds_ * 16 + l16, kgrp * 8), acc_o[ds_], 0, 0, 0);
+    if (nv > 2) {
+      bf16x8 pf1 = frag(sP, wrow0 + l16, kgrp * 8 + 32);
+#pragma unroll
+      for (int ds_ = 0; ds_ < 4; ++ds_)
+        acc_o[ds_] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            pf1, frag(sVT, ds_ * 16 + l16, kgrp * 8 + 32), acc_o[ds_], 0, 0, 0);
+    }
+  }
+
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int qrow = q0 + wrow0 + kgrp * 4 + r;
+    if (qrow >= Lq) continue;
+    const float inv = (row_sum[r] > 0.f) ? 1.f / row_sum[r] : 0.f;
+#pragma unroll
+    for (int ds_ = 0; ds_ < 4; ++ds_)
+      op[(long)qrow * qrs + ds_ * 16 + l16] = __float2bfloat16(acc_o[ds_][r] * inv);
+    if (l16 == 0 && lse != nullptr)
+      lse[(long)bh * Lq + qrow] = row_max[r] + __logf(fmaxf(row_sum[r], 1e-30f));
+  }
+}
+
 __global__ void attn_bwd_delta_kernel(const bf16_t* __restrict__ dO,
                                       const bf16_t* __restrict__ O,
                                       float* __restrict__ delta, long total_rows,
@@ -534,6 +675,17 @@ void attn_fwd_launch(const void* q, const void* k, const void* v, void* o,
                      bool causal, hipStream_t s) {
   dim3 grid((Lq + TILE - 1) / TILE, BH), block(256);
   hipLaunchKernelGGL(dcr_attn::attn_fwd_kernel, grid, block, 0, s,
+                     (const dcr_attn::bf16_t*)q, (const dcr_attn::bf16_t*)k,
+                     (const dcr_attn::bf16_t*)v, (dcr_attn::bf16_t*)o, lse,
+                     Lq, Lk, H, scale, causal ? 1 : 0);
+}
+
+// v2: masked-tail MFMA skip (round-2 draft; not dispatched)
+void attn_fwd_v2_launch(const void* q, const void* k, const void* v, void* o,
+                        float* lse, int BH, int Lq, int Lk, int H, float scale,
+                        bool causal, hipStream_t s) {
+  dim3 grid((Lq + TILE - 1) / TILE, BH), block(256);
+  hipLaunchKernelGGL(dcr_attn::attn_fwd_v2_kernel, grid, block, 0, s,
                      (const dcr_attn::bf16_t*)q, (const dcr_attn::bf16_t*)k,
                      (const dcr_attn::bf16_t*)v, (dcr_attn::bf16_t*)o, lse,
                      Lq, Lk, H, scale, causal ? 1 : 0);

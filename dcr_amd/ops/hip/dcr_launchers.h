@@ -64,6 +64,10 @@ void lincomb_launch(DType dt, const void* X, const void* Y, const void* Z,
 void attn_fwd_launch(const void* q, const void* k, const void* v, void* o,
                      float* lse, int BH, int Lq, int Lk, int H, float scale,
                      bool causal, hipStream_t s);
+// v2: masked-tail MFMA skip (round-2 draft; not dispatched)
+void attn_fwd_v2_launch(const void* q, const void* k, const void* v, void* o,
+                        float* lse, int BH, int Lq, int Lk, int H, float scale,
+                        bool causal, hipStream_t s);
 void attn_bwd_launch(const void* q, const void* k, const void* v,
                      const void* o, const void* dO, const float* lse,
                      float* delta, void* dQ, void* dK, void* dV, int BH,

@@ -18,5 +18,9 @@ DCR_NATIVE_CONV_BWD=1 python -m pytest tests/test_ops_gpu.py -k nhwc_bwd -x -q
 DCR_DEV_ADAMW=1 python -m pytest tests/test_ops_gpu.py -k adamw_dev -x -q
 DCR_DEV_ADAMW=1 python -m pytest tests/test_train_gpu.py -k device_state -x -q
 
-# perf A/B (v3/v2 ratio printed per shape)
+# attention v2: bit-exact masked-tail MFMA skip
+DCR_ATTN_V2=1 python -m pytest tests/test_ops_gpu.py -k attn_fwd_v2 -x -q
+
+# perf A/B (v3/v2 conv ratio, v2/v1 attention ratio per shape)
 python scripts/bench_conv.py
+python scripts/bench_attention.py

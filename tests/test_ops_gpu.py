@@ -389,6 +389,32 @@ def test_conv_nhwc_fwd(ext, shape, version):
     _close(y, ref, 2e-2)
 
 
+@pytest.mark.skipif(os.environ.get("DCR_ATTN_V2") != "1",
+                    reason="attn v2 masked-tail-skip draft: validate in "
+                           "round 2 (DCR_ATTN_V2=1)")
+@pytest.mark.parametrize("shape", [
+    (2, 5, 1024, 77, False),    # SD cross-attn: tail tile has 13 valid keys
+    (2, 4, 256, 256, False),
+    (2, 16, 77, 77, True),      # CLIP causal
+    (1, 2, 100, 37, False),     # odd everything, Lk < 64
+    (2, 5, 4096, 4096, False),
+])
+def test_attn_fwd_v2_bitexact_vs_v1(ext, shape):
+    """The skipped MFMAs only ever accumulate exact zeros, so v2 must be
+    BIT-IDENTICAL to the production kernel."""
+    B, H, Lq, Lk, causal = shape
+    torch.manual_seed(0)
+    D = 64
+    q = torch.randn(B, Lq, H, D, device="cuda").to(torch.bfloat16)
+    k = torch.randn(B, Lk, H, D, device="cuda").to(torch.bfloat16)
+    v = torch.randn(B, Lk, H, D, device="cuda").to(torch.bfloat16)
+    scale = D ** -0.5
+    o1, l1 = ext.attn_fwd(q, k, v, scale, causal)
+    o2, l2 = ext.attn_fwd_v2(q, k, v, scale, causal)
+    assert torch.equal(o1, o2)
+    assert torch.equal(l1, l2)
+
+
 @pytest.mark.skipif(os.environ.get("DCR_DEV_ADAMW") != "1",
                     reason="device-state AdamW draft: validate in round 2 "
                            "(DCR_DEV_ADAMW=1)")
