@@ -76,35 +76,62 @@ def dump_matches(scores: np.ndarray, keys: List[str], out_pickle: str | Path):
 # ---------------------------------------------------------------------------
 @torch.no_grad()
 def sharded_topk(query: torch.Tensor, shard: torch.Tensor, k: int = 1,
-                 chunk: int = 1 << 20, global_offset: int = 0
-                 ) -> Tuple[torch.Tensor, torch.Tensor]:
+                 chunk: int = 1 << 20, global_offset: int = 0,
+                 compute_dtype: Optional[torch.dtype] = None,
+                 rerank_margin: int = 4) -> Tuple[torch.Tensor, torch.Tensor]:
     """Per-rank: chunked GEMM over this rank's shard, running top-k.
 
     query [Q, D], shard [Ns, D] both on device. Returns
     (scores [Q, k], global_idx [Q, k]) for this shard.
+
+    compute_dtype=torch.bfloat16 runs the big GEMM at the bf16 MFMA rate
+    (2x fp32, half the index-read HBM traffic) over a top-(k+margin)
+    candidate list, then RE-SCORES the surviving candidates in fp32 so the
+    returned scores (and their ranking) are exact fp32 values — the only
+    approximation left is a true top-k member falling below bf16's ~0.004
+    score resolution AND outside the margin.
     """
     Q = query.shape[0]
     device = query.device
-    best_v = torch.full((Q, k), -1e30, device=device)
-    best_i = torch.zeros((Q, k), dtype=torch.long, device=device)
+    lowp = compute_dtype is not None and compute_dtype != torch.float32
+    m = k + max(0, rerank_margin) if lowp else k
+    q_mat = query.to(compute_dtype) if lowp else query
+    best_v = torch.full((Q, m), -1e30, device=device)
+    best_i = torch.zeros((Q, m), dtype=torch.long, device=device)
     for s in range(0, shard.shape[0], chunk):
         block = shard[s:s + chunk]
-        sim = query @ block.t()                     # [Q, c] rocBLAS GEMM
-        kk = min(k, sim.shape[1])
+        if lowp:
+            sim = (q_mat @ block.t().to(compute_dtype)).float()
+        else:
+            sim = q_mat @ block.t()                 # [Q, c] rocBLAS GEMM
+        kk = min(m, sim.shape[1])
         v, i = sim.topk(kk, dim=1)
-        i = i + s + global_offset
+        i = i + s
         cat_v = torch.cat([best_v, v], dim=1)
         cat_i = torch.cat([best_i, i], dim=1)
-        sel_v, sel_pos = cat_v.topk(k, dim=1)
+        sel_v, sel_pos = cat_v.topk(m, dim=1)
         best_v = sel_v
         best_i = torch.gather(cat_i, 1, sel_pos)
-    return best_v, best_i
+    if lowp:
+        if shard.shape[0] > 0:
+            # exact fp32 re-rank of the m candidates per query
+            cand = shard[best_i.reshape(-1)].reshape(Q, m, -1).float()
+            exact = torch.einsum("qd,qmd->qm", query.float(), cand)
+            exact = torch.where(best_v > -1e29, exact,
+                                torch.full_like(exact, -1e30))
+            best_v, pos = exact.topk(k, dim=1)
+            best_i = torch.gather(best_i, 1, pos)
+        else:
+            best_v, best_i = best_v[:, :k], best_i[:, :k]
+    return best_v, best_i + global_offset
 
 
 @torch.no_grad()
 def distributed_knn(query: torch.Tensor, shard: torch.Tensor, k: int = 1,
                     shard_sizes: Optional[Sequence[int]] = None,
-                    chunk: int = 1 << 20) -> Tuple[torch.Tensor, torch.Tensor]:
+                    chunk: int = 1 << 20,
+                    compute_dtype: Optional[torch.dtype] = None
+                    ) -> Tuple[torch.Tensor, torch.Tensor]:
     """All ranks hold the same [Q, D] query and their own shard of the
     index. Per-rank top-k -> RCCL all_gather of [Q, k] (score, idx) ->
     global top-k on every rank. Message is k*(4+8) bytes/query instead of
@@ -116,7 +143,8 @@ def distributed_knn(query: torch.Tensor, shard: torch.Tensor, k: int = 1,
     else:
         sizes = list(shard_sizes)
     offset = sum(sizes[:rank])
-    v, i = sharded_topk(query, shard, k=k, chunk=chunk, global_offset=offset)
+    v, i = sharded_topk(query, shard, k=k, chunk=chunk, global_offset=offset,
+                        compute_dtype=compute_dtype)
     if world == 1 or not dist.is_initialized():
         return v, i
     vs = [torch.empty_like(v) for _ in range(world)]

@@ -94,6 +94,45 @@ def test_backbone_arch_mapping():
         build_backbone(mk(pt_style="dino", arch="vit_base_cifar10"), "cpu")
 
 
+def test_sharded_topk_bf16_rerank():
+    """bf16-GEMM candidate pass + fp32 re-rank returns exact fp32 scores
+    and recovers orderings that bf16 alone cannot resolve."""
+    from dcr_amd.search import sharded_topk
+    torch.manual_seed(0)
+    q = torch.nn.functional.normalize(torch.randn(10, 64), dim=1)
+    shard = torch.nn.functional.normalize(torch.randn(500, 64), dim=1)
+    v32, i32 = sharded_topk(q, shard, k=3, chunk=128)
+    vbf, ibf = sharded_topk(q, shard, k=3, chunk=128,
+                            compute_dtype=torch.bfloat16)
+    # re-ranked scores are exact fp32 values of the selected rows
+    sel = torch.einsum("qd,qkd->qk", q, shard[ibf])
+    assert torch.allclose(vbf, sel, atol=1e-6)
+    # random cosine scores are well separated at k=3: same result as fp32
+    assert torch.equal(i32, ibf)
+    assert torch.allclose(v32, vbf, atol=1e-6)
+
+    # crafted near-tie below bf16 resolution: two candidates score
+    # 0.900000 vs 0.900150 — bf16 rounds both to the same value; the fp32
+    # re-rank inside the margin must still order them correctly
+    base = torch.nn.functional.normalize(torch.randn(1, 64), dim=1)
+    a = torch.nn.functional.normalize(
+        0.9 * base + (1 - 0.9**2) ** 0.5 *
+        torch.nn.functional.normalize(torch.randn(1, 64) - base * (torch.randn(1, 64) @ base.t()), dim=1), dim=1)
+    # build candidates with controlled exact scores against `base`
+    d = 64
+    e1 = torch.zeros(1, d); e1[0, 0] = 1.0
+    e2 = torch.zeros(1, d); e2[0, 1] = 1.0
+    qq = e1  # query along axis 0
+    c_lo = 0.900000 * e1 + (1 - 0.900000**2) ** 0.5 * e2
+    c_hi = 0.900150 * e1 + (1 - 0.900150**2) ** 0.5 * e2
+    filler = torch.nn.functional.normalize(torch.randn(50, d), dim=1) * 0.5
+    shard2 = torch.cat([c_lo, filler, c_hi], dim=0)
+    v, i = sharded_topk(qq, shard2, k=1, chunk=16,
+                        compute_dtype=torch.bfloat16)
+    assert int(i[0, 0]) == shard2.shape[0] - 1, "re-rank missed the true top-1"
+    assert abs(float(v[0, 0]) - 0.900150) < 1e-5
+
+
 def test_complexity_metrics():
     rng = np.random.default_rng(0)
     noise = rng.integers(0, 255, (64, 64, 3)).astype(np.uint8)
