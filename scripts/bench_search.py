@@ -31,6 +31,9 @@ def main():
     ap.add_argument("--k", type=int, default=10)
     ap.add_argument("--chunk", type=int, default=1 << 20)
     ap.add_argument("--repeat", type=int, default=3)
+    ap.add_argument("--bf16", action="store_true",
+                    help="bf16 GEMM + fp32 re-rank (sharded_topk "
+                         "compute_dtype) — expect ~2x the fp32 rate")
     args = ap.parse_args()
 
     from dcr_amd.parallel import dist as dist_utils
@@ -47,12 +50,15 @@ def main():
     query = torch.randn(args.queries, args.dim, generator=gq).to(device)
     query = torch.nn.functional.normalize(query, dim=-1)
 
-    v, i = distributed_knn(query, shard, k=args.k, chunk=args.chunk)  # warmup
+    cd = torch.bfloat16 if args.bf16 else None
+    v, i = distributed_knn(query, shard, k=args.k, chunk=args.chunk,
+                           compute_dtype=cd)  # warmup
     if use_cuda:
         torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(args.repeat):
-        v, i = distributed_knn(query, shard, k=args.k, chunk=args.chunk)
+        v, i = distributed_knn(query, shard, k=args.k, chunk=args.chunk,
+                               compute_dtype=cd)
     if use_cuda:
         torch.cuda.synchronize()
     dt = (time.perf_counter() - t0) / args.repeat
@@ -69,6 +75,7 @@ def main():
             "k": args.k,
             "sec_per_search": round(dt, 4),
             "n_gpus": world,
+            "dtype": "bf16_rerank" if args.bf16 else "fp32",
         }))
 
 
