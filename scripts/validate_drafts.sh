@@ -24,3 +24,7 @@ DCR_ATTN_V2=1 python -m pytest tests/test_ops_gpu.py -k attn_fwd_v2 -x -q
 # perf A/B (v3/v2 conv ratio, v2/v1 attention ratio per shape)
 python scripts/bench_conv.py
 python scripts/bench_attention.py
+
+# bf16-GEMM kNN vs fp32 (expect ~2x)
+python scripts/bench_search.py --repeat 2
+python scripts/bench_search.py --repeat 2 --bf16
