@@ -65,6 +65,11 @@ def parse_args():
     p.add_argument("-j", "--num_workers", type=int, default=4)
     p.add_argument("--modelsavesteps", type=int, default=1000)
     p.add_argument("--output_dir", type=str, default="model_out")
+    p.add_argument("--resume", type=str, default=None,
+                   help="'auto' = latest checkpoint_*/ (or checkpoint/) in "
+                        "the mangled output dir; or an explicit checkpoint "
+                        "directory. Restores weights, optimizer state, step "
+                        "count and RNG streams (state.pt).")
     p.add_argument("--duplication", type=str, default="nodup",
                    choices=["nodup", "dup_both", "dup_image"])
     p.add_argument("--weight_pc", type=float, default=0.05)
@@ -151,6 +156,24 @@ def main():
         print(f"output_dir: {cfg.output_dir}")
 
     trainer = Trainer(cfg)
+
+    if args.resume:
+        from pathlib import Path as _P
+        if args.resume == "auto":
+            root = _P(cfg.output_dir)
+            cands = sorted(root.glob("checkpoint_*"),
+                           key=lambda p: int(p.name.split("_")[-1]))
+            ck = cands[-1] if cands else root / "checkpoint"
+        else:
+            ck = _P(args.resume)
+        if (ck / "state.pt").exists() or (ck / "unet").exists():
+            trainer.load_checkpoint(ck)
+            if is_main_process():
+                print(f"resumed from {ck} at step {trainer.global_step}")
+        elif args.resume != "auto":
+            raise SystemExit(f"--resume: no checkpoint at {ck}")
+        elif is_main_process():
+            print("--resume auto: no checkpoint found, starting fresh")
 
     def sample_fn(tr: Trainer, out_dir):
         """Periodic sample grids (reference diff_train.py:571-611,673-701)."""

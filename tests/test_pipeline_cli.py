@@ -156,6 +156,35 @@ def test_train_cli_end_to_end(tmp_path):
     assert (out / "checkpoint" / "state.pt").exists()
 
 
+def test_train_cli_resume(tmp_path):
+    """--resume auto picks the latest checkpoint_N and continues from its
+    step count."""
+    args = [sys.executable, "diff_train.py", "--synthetic_data",
+            "--model_size", "tiny", "--resolution", "64",
+            "--train_batch_size", "2", "--mixed_precision", "no",
+            "--class_prompt", "classlevel", "--num_workers", "0",
+            "--seed", "0", "--save_steps", "1000",
+            "--output_dir", str(tmp_path / "m")]
+    r = subprocess.run(args + ["--max_train_steps", "2",
+                               "--modelsavesteps", "2"],
+                       capture_output=True, text=True,
+                       cwd=str(Path(__file__).parent.parent), timeout=870)
+    assert r.returncode == 0, r.stderr[-3000:]
+    out = tmp_path / "m_classlevel_nodup"
+    assert (out / "checkpoint_2" / "state.pt").exists()
+    r2 = subprocess.run(args + ["--max_train_steps", "4",
+                                "--modelsavesteps", "1000",
+                                "--resume", "auto"],
+                        capture_output=True, text=True,
+                        cwd=str(Path(__file__).parent.parent), timeout=870)
+    assert r2.returncode == 0, r2.stderr[-3000:]
+    assert "resumed from" in r2.stdout and "step 2" in r2.stdout, r2.stdout
+    import torch as _t
+    st = _t.load(out / "checkpoint" / "state.pt", map_location="cpu",
+                 weights_only=False)
+    assert st["global_step"] == 4
+
+
 @pytest.mark.timeout(900)
 def test_inference_cli_end_to_end(tmp_path, monkeypatch):
     """diff_inference.py from a tiny checkpoint: prompts.txt + numbered pngs."""
