@@ -28,6 +28,26 @@ def test_tokenizer_name_flag_wins(tmp_path):
     assert tr.tokenizer.model_max_length == 33
 
 
+def test_train_text_encoder_flag(tmp_path):
+    """--train_text_encoder (reference diff_train.py:140): text-encoder
+    params join the optimizer and move; frozen otherwise."""
+    tr = Trainer(tiny_cfg(tmp_path, train_text_encoder=True))
+    p = next(tr.text_encoder.parameters())
+    before = p.detach().clone()
+    batch = next(iter(tr.dataloader))
+    loss = tr.train_step(batch)
+    assert torch.isfinite(loss)
+    assert not torch.equal(p.detach(), before), "text encoder did not train"
+
+    tr2 = Trainer(tiny_cfg(tmp_path, train_text_encoder=False,
+                           output_dir=str(tmp_path / "out2")))
+    p2 = next(tr2.text_encoder.parameters())
+    assert not p2.requires_grad
+    before2 = p2.detach().clone()
+    tr2.train_step(next(iter(tr2.dataloader)))
+    assert torch.equal(p2.detach(), before2)
+
+
 def test_fp16_overflow_skips_step(tmp_path):
     """fp16 scaler path: non-finite unscaled grads must skip the optimizer
     update and back the scale off; finite grads step and call update()."""
