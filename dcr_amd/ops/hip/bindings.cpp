@@ -503,8 +503,10 @@ std::vector<at::Tensor> conv2d_nhwc_bwd(at::Tensor dy, at::Tensor x,
   conv_bwd_data_launch(dy.data_ptr(), w.data_ptr(), dx.data_ptr(), (int)Nb,
                        (int)Hin, (int)Win, (int)C, (int)K, (int)P, (int)Q,
                        (int)R, (int)S, (int)stride, (int)pad, cur_stream());
-  (void)NPQ;
-  return {dx, dW};
+  auto db = at::zeros({K}, x.options().dtype(at::kFloat));
+  conv_bias_grad_launch(dy.data_ptr(), db.data_ptr<float>(), NPQ, (int)K,
+                        cur_stream());
+  return {dx, dW, db};
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {

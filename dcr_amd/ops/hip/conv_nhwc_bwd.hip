@@ -155,6 +155,22 @@ __global__ void bwdw_finalize_kernel(const float* __restrict__ ws,
     dW[i] = __float2bfloat16(ws[i]);
 }
 
+// db[k] = sum over output pixels of dy[m][k] (NHWC: k innermost, so
+// consecutive lanes read consecutive channels — coalesced). Replaces the
+// ~3 ms/step of aten bias-grad reductions once the native bwd dispatches.
+// grid (ceil(K/256), row_splits); fp32 atomics into a zeroed db.
+__global__ void conv_bias_grad_kernel(const bf16_t* __restrict__ dy,
+                                      float* __restrict__ db, long NPQ, int K,
+                                      long rows_per_block) {
+  const int k = blockIdx.x * blockDim.x + threadIdx.x;
+  if (k >= K) return;
+  const long r0 = (long)blockIdx.y * rows_per_block;
+  const long r1 = min(NPQ, r0 + rows_per_block);
+  float s = 0.f;
+  for (long m = r0; m < r1; ++m) s += to_f32<bf16_t>(dy[m * K + k]);
+  atomicAdd(&db[k], s);
+}
+
 // ==========================================================================
 // bwd-data
 // grid: (ceil(NHW/128), C/64); block 256 (4 waves as 2x2: 64 pix x 32 c
@@ -309,6 +325,20 @@ void conv_bwd_weight_launch(const void* dy, const void* x, float* dw_ws,
   hipLaunchKernelGGL(dcr_conv_bwd::bwdw_finalize_kernel, dim3((unsigned)b),
                      dim3(256), 0, st, dw_ws,
                      (dcr_conv_bwd::bf16_t*)dW_out, total);
+}
+
+void conv_bias_grad_launch(const void* dy, float* db, long NPQ, int K,
+                           hipStream_t st) {
+  long rows_per_block = 4096;
+  unsigned ysplit = (unsigned)((NPQ + rows_per_block - 1) / rows_per_block);
+  if (ysplit > 1024) {  // keep the grid bounded for huge NPQ
+    rows_per_block = (NPQ + 1023) / 1024;
+    ysplit = (unsigned)((NPQ + rows_per_block - 1) / rows_per_block);
+  }
+  dim3 grid((unsigned)((K + 255) / 256), ysplit), block(256);
+  hipLaunchKernelGGL(dcr_conv_bwd::conv_bias_grad_kernel, grid, block, 0, st,
+                     (const dcr_conv_bwd::bf16_t*)dy, db, NPQ, K,
+                     rows_per_block);
 }
 
 void conv_bwd_data_launch(const void* dy, const void* w, void* dx, int Nb,

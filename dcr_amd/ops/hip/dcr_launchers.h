@@ -98,5 +98,7 @@ void conv_bwd_weight_launch(const void* dy, const void* x, float* dw_ws,
 void conv_bwd_data_launch(const void* dy, const void* w, void* dx, int Nb,
                           int Hin, int Win, int C, int K, int P, int Q, int R,
                           int S, int stride, int pad, hipStream_t st);
+void conv_bias_grad_launch(const void* dy, float* db, long NPQ, int K,
+                           hipStream_t st);
 
 }  // namespace dcr

@@ -538,9 +538,10 @@ def test_conv_nhwc_bwd(ext, shape):
     P = (H + 2 * pad - R) // stride + 1
     dy = torch.randn(N, K, P, P, device="cuda").to(torch.bfloat16) \
         .to(memory_format=torch.channels_last)
-    dx, dW = ext.conv2d_nhwc_bwd(dy, x, w, stride, pad)
-    rx, rw, _ = torch.ops.aten.convolution_backward(
-        dy.float(), x.float(), w.float(), None, [stride, stride], [pad, pad],
-        [1, 1], False, [0, 0], 1, [True, True, False])
+    dx, dW, db = ext.conv2d_nhwc_bwd(dy, x, w, stride, pad)
+    rx, rw, rb = torch.ops.aten.convolution_backward(
+        dy.float(), x.float(), w.float(), [w.shape[0]], [stride, stride],
+        [pad, pad], [1, 1], False, [0, 0], 1, [True, True, True])
     _close(dx, rx, 3e-2)
     _close(dW, rw, 3e-2)
+    _close(db, rb, 1e-2)
