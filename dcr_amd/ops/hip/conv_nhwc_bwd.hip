@@ -167,7 +167,7 @@ __global__ void conv_bias_grad_kernel(const bf16_t* __restrict__ dy,
   const long r0 = (long)blockIdx.y * rows_per_block;
   const long r1 = min(NPQ, r0 + rows_per_block);
   float s = 0.f;
-  for (long m = r0; m < r1; ++m) s += to_f32<bf16_t>(dy[m * K + k]);
+  for (long m = r0; m < r1; ++m) s += dcr::to_f32<bf16_t>(dy[m * K + k]);
   atomicAdd(&db[k], s);
 }
 
