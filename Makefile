@@ -1,5 +1,5 @@
 # dcr_amd — common targets
-.PHONY: build test test-gpu bench sanitize clean
+.PHONY: build test test-gpu bench bench8 sanitize validate-drafts clean
 
 build:            ## compile the gfx950 HIP extension in-tree
 	python -c "import __graft_entry__ as g; g.build()"
@@ -19,6 +19,9 @@ bench8:           ## 8-GPU weak-scaling bench (one rank per GPU over RCCL)
 
 sanitize:         ## kernel tests with serialized launches (GPU box)
 	bash scripts/sanitize.sh
+
+validate-drafts:  ## hardware-validate the gated round-2 draft kernels (GPU box)
+	bash scripts/validate_drafts.sh
 
 clean:
 	rm -rf dcr_amd/ops/_build dcr_amd/ops/_dcr_hip.so
