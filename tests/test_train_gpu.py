@@ -54,6 +54,23 @@ def test_sd21_train_step_gpu(tmp_path):
     assert torch.isfinite(loss)
 
 
+def test_sd14_forward_gpu():
+    """SD-1.4 family (sd_mitigation's model): head_dim 40/80/160 takes the
+    composite attention path; norms/GEGLU still run the HIP kernels."""
+    from dcr_amd.models import (AutoencoderKL, CLIPTextModel, CLIPTextConfig,
+                                UNet2DConditionModel, UNetConfig, VAEConfig)
+    torch.manual_seed(0)
+    unet = UNet2DConditionModel(UNetConfig.sd14()).cuda().to(torch.bfloat16)
+    te = CLIPTextModel(CLIPTextConfig.sd14()).cuda().to(torch.bfloat16)
+    ids = torch.randint(0, 49408, (2, 77), device="cuda")
+    with torch.no_grad():
+        emb = te(ids)[0]
+        out = unet(torch.randn(2, 4, 32, 32, device="cuda").bfloat16(),
+                   torch.tensor([10, 500], device="cuda"), emb)
+    assert out.shape == (2, 4, 32, 32)
+    assert torch.isfinite(out.float()).all()
+
+
 def test_unet_fwd_hip_matches_cpu_reference():
     """Tiny UNet forward on GPU (HIP kernels) vs CPU (torch fallback)."""
     from dcr_amd.models import UNet2DConditionModel, UNetConfig
