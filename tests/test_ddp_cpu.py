@@ -26,7 +26,7 @@ def _worker(rank, world, port, tmpdir):
         flat = opt.flat_param.clone()
         gathered = [torch.empty_like(flat) for _ in range(world)]
         dist.all_gather(gathered, flat)
-        assert torch.equal(gathered[0], gathered[1]), "broadcast failed"
+        assert all(torch.equal(gathered[0], g) for g in gathered), "broadcast failed"
 
         # different data per rank
         torch.manual_seed(500 + rank)
@@ -39,13 +39,13 @@ def _worker(rank, world, port, tmpdir):
         gflat = opt.flat_grad.clone()
         ggath = [torch.empty_like(gflat) for _ in range(world)]
         dist.all_gather(ggath, gflat)
-        assert torch.allclose(ggath[0], ggath[1], atol=1e-7), "grad avg mismatch"
+        assert all(torch.allclose(ggath[0], g, atol=1e-7) for g in ggath), "grad avg mismatch"
 
         opt.step()
         pflat = opt.flat_param.clone()
         pgath = [torch.empty_like(pflat) for _ in range(world)]
         dist.all_gather(pgath, pflat)
-        assert torch.allclose(pgath[0], pgath[1], atol=1e-7), "params diverged"
+        assert all(torch.allclose(pgath[0], g, atol=1e-7) for g in pgath), "params diverged"
 
         # non-sync micro step launches no collectives and keeps buckets reset
         ddp.require_backward_grad_sync = False
@@ -59,6 +59,16 @@ def _worker(rank, world, port, tmpdir):
 def test_bucketed_allreduce_world2(tmp_path):
     port = 29711
     mp.spawn(_worker, args=(2, port, str(tmp_path)), nprocs=2, join=True)
+
+
+def test_bucketed_allreduce_world4(tmp_path):
+    """collective tests at node-scale world sizes (one MI355X
+    node is 8 GPUs; gloo stands in for RCCL here)."""
+    mp.spawn(_worker, args=(4, 29727, str(tmp_path)), nprocs=4, join=True)
+
+
+def test_bucketed_allreduce_world8(tmp_path):
+    mp.spawn(_worker, args=(8, 29729, str(tmp_path)), nprocs=8, join=True)
 
 
 def _trainer_worker(rank, world, port, tmpdir):
