@@ -29,3 +29,12 @@ def _worker(rank, world, port):
 
 def test_distributed_knn_world2():
     mp.spawn(_worker, args=(2, 29717), nprocs=2, join=True)
+
+
+def test_distributed_knn_world4():
+    """top-k all-gather at node-scale world sizes."""
+    mp.spawn(_worker, args=(4, 29731), nprocs=4, join=True)
+
+
+def test_distributed_knn_world8():
+    mp.spawn(_worker, args=(8, 29733), nprocs=8, join=True)
