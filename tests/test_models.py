@@ -87,6 +87,44 @@ def test_clip_text_causal():
     assert not torch.allclose(h1[0, 9], h2[0, 9], atol=1e-5)
 
 
+def test_clip_text_matches_transformers():
+    """Module parity vs the REAL library: our CLIP text
+    encoder's state dict loads into transformers' CLIPTextModel with zero
+    missing/unexpected keys (after the on-disk `text_model.` prefix that
+    transformers strips in memory) and the forward outputs are
+    bit-identical. This is the checkpoint-compatibility proof for the
+    text-encoder third of a diffusers-layout checkpoint."""
+    import torch
+    from transformers import CLIPTextConfig as HFCfg
+    from transformers import CLIPTextModel as HFModel
+
+    from dcr_amd.models import CLIPTextConfig, CLIPTextModel
+
+    for act in ("gelu", "quick_gelu"):
+        torch.manual_seed(0)
+        cfg = CLIPTextConfig(vocab_size=1000, hidden_size=32,
+                             intermediate_size=64, num_hidden_layers=2,
+                             num_attention_heads=2, hidden_act=act)
+        ours = CLIPTextModel(cfg).eval()
+        hf = HFModel(HFCfg(
+            vocab_size=cfg.vocab_size, hidden_size=cfg.hidden_size,
+            intermediate_size=cfg.intermediate_size,
+            num_hidden_layers=cfg.num_hidden_layers,
+            num_attention_heads=cfg.num_attention_heads,
+            max_position_embeddings=cfg.max_position_embeddings,
+            hidden_act=act, attention_dropout=0.0, bos_token_id=0,
+            eos_token_id=1)).eval()
+        sd = {k.removeprefix("text_model."): v
+              for k, v in ours.state_dict().items()}
+        missing, unexpected = hf.load_state_dict(sd, strict=False)
+        assert not missing and not unexpected, (missing, unexpected)
+        ids = torch.randint(0, cfg.vocab_size, (2, 77))
+        with torch.no_grad():
+            o1 = ours(ids)[0]
+            o2 = hf(ids).last_hidden_state
+        assert torch.equal(o1, o2), (act, (o1 - o2).abs().max())
+
+
 def test_checkpoint_roundtrip(tmp_path):
     unet = UNet2DConditionModel(UNetConfig.tiny())
     unet.save_pretrained(tmp_path / "unet")
