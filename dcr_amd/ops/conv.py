@@ -57,7 +57,10 @@ class _NativeConvFn(torch.autograd.Function):
     def forward(ctx, x, weight, bias, stride, padding):
         m = ext()
         count_dispatch('conv_nhwc')
-        y = m.conv2d_nhwc_fwd_v2(x, weight, bias, stride, padding)
+        # DCR_NATIVE_CONV_V3=1: double-buffered draft (round-2 A/B)
+        fwd = m.conv2d_nhwc_fwd_v3 if os.environ.get("DCR_NATIVE_CONV_V3") == "1" \
+            else m.conv2d_nhwc_fwd_v2
+        y = fwd(x, weight, bias, stride, padding)
         ctx.save_for_backward(x, weight)
         ctx.conf = (stride, padding, bias is not None)
         return y

@@ -15,6 +15,7 @@ loop the reference runs through libraries:
 from __future__ import annotations
 
 import math
+import os
 from typing import Optional
 
 import torch
@@ -172,7 +173,10 @@ class _FlashAttention(torch.autograd.Function):
     def forward(ctx, q, k, v, scale, causal):
         m = require_hip("attn")
         count_dispatch('attention')
-        o, lse = m.attn_fwd(q, k, v, scale, causal)
+        # DCR_ATTN_V2=1: bit-exact masked-tail-skip draft (round-2 A/B)
+        fwd = m.attn_fwd_v2 if os.environ.get("DCR_ATTN_V2") == "1" \
+            else m.attn_fwd
+        o, lse = fwd(q, k, v, scale, causal)
         ctx.save_for_backward(q, k, v, o, lse)
         ctx.scale = scale
         ctx.causal = causal
