@@ -28,3 +28,10 @@ python scripts/bench_attention.py
 # bf16-GEMM kNN vs fp32 (expect ~2x)
 python scripts/bench_search.py --repeat 2
 python scripts/bench_search.py --repeat 2 --bf16
+
+# whole-model A/B: each draft is also wired into the real model behind its
+# env var, so the full bench can A/B them directly (run separately — each
+# is a fresh process and pays the MIOpen find warmup):
+#   DCR_ATTN_V2=1        python bench.py --steps 10 --warmup 6
+#   DCR_NATIVE_CONV_V3=1 python bench.py --steps 10 --warmup 6
+#   DCR_DEV_ADAMW=1      python bench.py --steps 10 --warmup 6
