@@ -52,9 +52,14 @@ def parse_args():
     p.add_argument("--multiscale", default=False, type=lambda s: s in ("1", "true", "True"))
     p.add_argument("--pretrained", default="", type=str)
     p.add_argument("--similarity_metric", default="dotproduct", type=str)
-    p.add_argument("--num_loss_chunks", default=1, type=int)
-    p.add_argument("--numpatches", default=1, type=int)
-    p.add_argument("--isvit", action="store_true")
+    p.add_argument("--num_loss_chunks", default=1, type=int,
+                   help="(reference flag; splitloss here derives per-patch "
+                        "descriptors from the model's own token/spatial map "
+                        "instead of slicing the flat embedding)")
+    p.add_argument("--numpatches", default=1, type=int,
+                   help="(reference flag; see --num_loss_chunks)")
+    p.add_argument("--isvit", action="store_true",
+                   help="(reference flag; ViT-ness is detected from the model)")
     p.add_argument("--layer", default=1, type=int)
     p.add_argument("--stype", default="", type=str)
     p.add_argument("--keephead", action="store_true")
