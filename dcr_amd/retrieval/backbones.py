@@ -144,7 +144,8 @@ class VGG16(nn.Module):
     CFG = [64, 64, "M", 128, 128, "M", 256, 256, 256, "M",
            512, 512, 512, "M", 512, 512, 512, "M"]
 
-    def __init__(self, num_classes: int = 1000):
+    def __init__(self, num_classes: int = 1000,
+                 weights_path: str | None = None):
         super().__init__()
         layers, in_ch = [], 3
         for v in self.CFG:
@@ -159,6 +160,13 @@ class VGG16(nn.Module):
             nn.Linear(512 * 7 * 7, 4096), nn.ReLU(True), nn.Dropout(),
             nn.Linear(4096, 4096), nn.ReLU(True), nn.Dropout(),
             nn.Linear(4096, num_classes))
+        # torchvision-format ImageNet weights when available offline
+        # (reference metrics/ipr.py:41 downloads them via torchvision)
+        wp = Path(weights_path) if weights_path else \
+            Path("./pretrainedmodels") / "vgg16.pth"
+        if wp.exists():
+            sd = torch.load(str(wp), map_location="cpu", weights_only=True)
+            self.load_state_dict(sd, strict=False)
 
     def forward(self, x):
         x = self.features(x)
