@@ -25,6 +25,13 @@ def main():
         print(f"  exported ops ({len(kernels)}): {', '.join(kernels)}")
     if ops.dispatch_counts:
         print(f"dispatch counts: {dict(ops.dispatch_counts)}")
+    import os
+    gates = ["DCR_NATIVE_CONV", "DCR_NATIVE_CONV_V3", "DCR_NATIVE_CONV_BWD",
+             "DCR_ATTN_V2", "DCR_DEV_ADAMW", "DCR_PROFILE",
+             "DCR_AMD_ALLOW_FALLBACK"]
+    active = {g: os.environ[g] for g in gates if g in os.environ}
+    desc = active if active else "defaults (native conv on; round-2 drafts off)"
+    print(f"env gates: {desc}")
 
 
 if __name__ == "__main__":
