@@ -1,5 +1,5 @@
 # dcr_amd — common targets
-.PHONY: build test test-gpu bench bench8 sanitize validate-drafts clean
+.PHONY: build test test-gpu bench bench8 sanitize validate-drafts status clean
 
 build:            ## compile the gfx950 HIP extension in-tree
 	python -c "import __graft_entry__ as g; g.build()"
@@ -25,3 +25,6 @@ validate-drafts:  ## hardware-validate the gated round-2 draft kernels (GPU box)
 
 clean:
 	rm -rf dcr_amd/ops/_build dcr_amd/ops/_dcr_hip.so
+
+status:           ## build/kernel/env-gate status report
+	python -m dcr_amd
