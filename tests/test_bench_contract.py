@@ -11,6 +11,11 @@ ROOT = Path(__file__).parent.parent
 
 def _check_line(line: str, n_gpus: int):
     rec = json.loads(line)
+    # every field the driver's contract names must be present
+    required = {"metric", "value", "unit", "n_gpus", "steps", "warmup",
+                "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
+                "dtype", "data", "config"}
+    assert required <= rec.keys(), required - rec.keys()
     assert rec["metric"] == "sd21_256px_finetune_imgs_per_sec"
     assert rec["n_gpus"] == n_gpus
     assert rec["higher_is_better"] is True
@@ -19,6 +24,7 @@ def _check_line(line: str, n_gpus: int):
     assert rec["value"] > 0 and rec["ms_per_step"] > 0
     assert rec["config"]["global_batch"] == 2 * n_gpus
     assert rec["config"]["parallelism"] == f"dp{n_gpus}"
+    assert {"model", "global_batch", "seq_len"} <= rec["config"].keys()
     return rec
 
 
