@@ -51,11 +51,22 @@ def load_module(module: torch.nn.Module, path: Path, strict: bool = True):
     cfg_json = (path / "config.json").read_text() if (path / "config.json").exists() else "{}"
     wf = path / _weight_file(cfg_json)
     if not wf.exists():
-        # tolerate either name
-        cands = list(path.glob("*.safetensors"))
-        if not cands:
+        # tolerate either name, but deterministically: prefer known
+        # diffusers/transformers weight filenames, and refuse to guess
+        # between multiple unknown candidates (glob order is fs-dependent)
+        known = ("diffusion_pytorch_model.safetensors", "model.safetensors")
+        cands = sorted(path.glob("*.safetensors"))
+        preferred = [c for c in cands if c.name in known]
+        if preferred:
+            wf = preferred[0]
+        elif len(cands) == 1:
+            wf = cands[0]
+        elif not cands:
             raise FileNotFoundError(f"no safetensors weights under {path}")
-        wf = cands[0]
+        else:
+            raise FileNotFoundError(
+                f"ambiguous safetensors weights under {path}: "
+                f"{[c.name for c in cands]}")
     sd = load_file(str(wf))
     module.load_state_dict(sd, strict=strict)
     return module

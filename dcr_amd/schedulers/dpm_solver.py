@@ -96,7 +96,11 @@ class DPMSolverMultistepScheduler:
 
         K = float(-a_prev * torch.expm1(-h))
         c_x = float(s_prev / s_t)
-        if self.lower_order_nums < 1 or self.model_outputs[-2] is None or i + 1 >= len(self.timesteps):
+        # diffusers parity: lower_order_final only drops to first order on
+        # the last step when num_inference_steps < 15
+        final_first_order = (i + 1 >= len(self.timesteps)
+                             and len(self.timesteps) < 15)
+        if self.lower_order_nums < 1 or self.model_outputs[-2] is None or final_first_order:
             # first-order (DDIM-like) update in x0-parameterization
             if sample.is_cuda:
                 prev = ops.lincomb(sample, x0, c_x, K)

@@ -78,6 +78,7 @@ class Trainer:
         # bf16 does not.
         self.scaler = torch.amp.GradScaler("cuda") \
             if cfg.mixed_precision == "fp16" and torch.cuda.is_available() else None
+        self._finite_streak = 0  # consecutive finite fp16 steps (manual growth)
 
         self._build_models()
         self._build_data()
@@ -265,11 +266,16 @@ class Trainer:
                 raise ValueError(self.noise_scheduler.prediction_type)
 
         loss = F.mse_loss(model_pred.float(), target.float(), reduction="mean")
+        # reference parity: accelerator.backward(loss) divides by the
+        # accumulation count so accumulated grads are a mean, not a sum
+        # (reference diff_train.py:656 via accelerate)
+        accum = cfg.gradient_accumulation_steps
+        bwd_loss = loss / accum if accum > 1 else loss
         with self.prof.phase("backward"):
             if self.scaler is not None:
-                self.scaler.scale(loss).backward()
+                self.scaler.scale(bwd_loss).backward()
             else:
-                loss.backward()
+                bwd_loss.backward()
 
         if sync_gradients:
             with self.prof.phase("optimizer"):
@@ -291,11 +297,22 @@ class Trainer:
                         self.optimizer.clip_grad_norm_(cfg.max_grad_norm)
                         self.optimizer.step(lr=lr_step)
                     if self.scaler is not None:
-                        self.scaler.update()  # success: growth tracking
+                        # growth bookkeeping is done manually (we never call
+                        # scaler.step()/unscale_(), so a bare update() would
+                        # assert "No inf checks were recorded"): grow 2x after
+                        # growth_interval consecutive finite steps, matching
+                        # GradScaler defaults
+                        self._finite_streak += 1
+                        if self._finite_streak >= 2000:
+                            self.scaler.update(self.scaler.get_scale() * 2.0)
+                            self._finite_streak = 0
+                        else:
+                            self.scaler.update(self.scaler.get_scale())
                 else:
                     # fp16 overflow: skip the update, back the scale off
                     # (GradScaler semantics; reference utils_ret.py:834-860
                     # relies on torch's internal version of this)
+                    self._finite_streak = 0
                     self.scaler.update(self.scaler.get_scale() * 0.5)
                 self.optimizer.zero_grad()
             self.global_step += 1

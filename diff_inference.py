@@ -32,22 +32,24 @@ def resize(w_val, l_val, img):
 
 
 def prompt_augmentation(prompt, aug_style, tokenizer=None, repeat_num=2):
-    """Reference diff_inference.py:14-30."""
-    if aug_style == "rand_numb_add":
-        for _ in range(repeat_num):
-            randnum = np.random.choice(100000)
-            prompt = insert_rand_word(prompt, str(randnum))
-    elif aug_style == "rand_word_add":
-        for _ in range(repeat_num):
-            randword = tokenizer.decode(list(np.random.randint(49400, size=1)))
-            prompt = insert_rand_word(prompt, randword)
-    elif aug_style == "rand_word_repeat":
-        wordlist = prompt.split(" ")
-        for _ in range(repeat_num):
-            randword = np.random.choice(wordlist)
-            prompt = insert_rand_word(prompt, randword)
-    else:
-        raise Exception("This style of prompt augmnentation is not written")
+    """Inference-time prompt perturbation (semantics of reference
+    diff_inference.py:14-30): insert `repeat_num` extra tokens at random
+    word boundaries — random integers, random vocabulary words, or words
+    already present in the prompt."""
+    original_words = prompt.split(" ")
+
+    def pick_word() -> str:
+        if aug_style == "rand_numb_add":
+            return str(np.random.choice(100000))
+        if aug_style == "rand_word_add":
+            token_id = int(np.random.randint(49400))
+            return tokenizer.decode([token_id])
+        if aug_style == "rand_word_repeat":
+            return str(np.random.choice(original_words))
+        raise Exception(f"unknown prompt augmentation style: {aug_style!r}")
+
+    for _ in range(repeat_num):
+        prompt = insert_rand_word(prompt, pick_word())
     return prompt
 
 

@@ -74,7 +74,11 @@ def test_fp16_overflow_skips_step(tmp_path):
     p0 = tr.optimizer.flat_param.clone()
     tr.train_step(batch)
     assert not torch.equal(tr.optimizer.flat_param, p0)
-    assert tr.scaler.updates == [None]
+    # success path passes the scale explicitly (no torch inf-check records
+    # exist because scaler.step()/unscale_() are never used); growth only
+    # after 2000 consecutive finite steps
+    assert tr.scaler.updates == [2.0]
+    assert tr._finite_streak == 1
     # overflow path: tiny scale -> inv=1e300 -> inf grads -> skip + backoff
     tr.scaler = _FakeScaler(1e-300)
     p1 = tr.optimizer.flat_param.clone()
@@ -112,6 +116,36 @@ def test_gradient_accumulation_equivalence(tmp_path):
     assert not torch.equal(tr1.optimizer.flat_param, p0)
     assert not torch.equal(g_half, torch.zeros_like(g_half))
     del tr2
+
+
+def test_gradient_accumulation_is_mean(tmp_path):
+    """Accumulated grads are the MEAN over micro-steps (reference parity:
+    accelerate divides the loss by gradient_accumulation_steps before
+    backward), not the sum."""
+    cfg = tiny_cfg(tmp_path)
+    cfg.gradient_accumulation_steps = 2
+    tr = Trainer(cfg, device=torch.device("cpu"))
+    b = next(iter(tr.dataloader))
+
+    # identical RNG per micro-step -> identical per-micro grads; the mean
+    # of two equal grads equals one micro-step's grad at accum=1
+    torch.manual_seed(7)
+    tr.train_step(b, sync_gradients=False)
+    g1 = tr.optimizer.flat_grad.clone()
+    torch.manual_seed(7)
+    tr.train_step(b, sync_gradients=False)
+    g2 = tr.optimizer.flat_grad.clone()
+    # second micro-step added the same scaled grad again: g2 == 2*g1
+    assert torch.allclose(g2, 2 * g1, rtol=1e-5, atol=1e-7)
+
+    cfg1 = tiny_cfg(tmp_path)
+    cfg1.gradient_accumulation_steps = 1
+    tr1 = Trainer(cfg1, device=torch.device("cpu"))
+    tr1.optimizer.flat_param.copy_(tr.optimizer.flat_param)
+    torch.manual_seed(7)
+    tr1.train_step(b, sync_gradients=False)
+    # accum=2 total grad (mean over 2 identical micros) == accum=1 grad
+    assert torch.allclose(g2, tr1.optimizer.flat_grad, rtol=1e-5, atol=1e-7)
 
 
 def test_resume_roundtrip(tmp_path):
