@@ -26,7 +26,7 @@ def main():
     if ops.dispatch_counts:
         print(f"dispatch counts: {dict(ops.dispatch_counts)}")
     import os
-    gates = ["DCR_NATIVE_CONV", "DCR_NATIVE_CONV_V3", "DCR_NATIVE_CONV_BWD",
+    gates = ["DCR_NATIVE_CONV", "DCR_NATIVE_CONV_BWD", "DCR_NATIVE_GEMM",
              "DCR_ATTN_V2", "DCR_DEV_ADAMW", "DCR_PROFILE",
              "DCR_AMD_ALLOW_FALLBACK"]
     active = {g: os.environ[g] for g in gates if g in os.environ}

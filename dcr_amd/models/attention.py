@@ -5,7 +5,8 @@ Attention as run by the finetune hot loop (/root/reference/diff_train.py:644)
 and sampling (diff_inference.py:190). MI355X design: LayerNorm and the
 GEGLU gate are single fused HIP kernels; attention routes through
 dcr_amd.ops.attention (flash-style CDNA4 kernel on GPU); the QKV/out
-projections are plain rocBLAS GEMMs.
+projections route through the in-tree bf16 MFMA GEMM
+(dcr_amd/ops/hip/gemm.hip) on eligible shapes (DcrLinear).
 """
 from __future__ import annotations
 
@@ -16,6 +17,7 @@ import torch.nn as nn
 
 from .. import ops
 from .layers import GroupNormOp, LayerNormOp
+from ..ops.linear import DcrLinear
 
 
 class Attention(nn.Module):
@@ -36,10 +38,10 @@ class Attention(nn.Module):
         kv_dim = cross_attention_dim or query_dim
         self.heads = heads
         self.dim_head = dim_head
-        self.to_q = nn.Linear(query_dim, inner_dim, bias=bias)
-        self.to_k = nn.Linear(kv_dim, inner_dim, bias=bias)
-        self.to_v = nn.Linear(kv_dim, inner_dim, bias=bias)
-        self.to_out = nn.ModuleList([nn.Linear(inner_dim, query_dim, bias=out_bias),
+        self.to_q = DcrLinear(query_dim, inner_dim, bias=bias)
+        self.to_k = DcrLinear(kv_dim, inner_dim, bias=bias)
+        self.to_v = DcrLinear(kv_dim, inner_dim, bias=bias)
+        self.to_out = nn.ModuleList([DcrLinear(inner_dim, query_dim, bias=out_bias),
                                      nn.Dropout(dropout)])
 
     def forward(self, x: torch.Tensor, context: Optional[torch.Tensor] = None) -> torch.Tensor:
@@ -64,8 +66,8 @@ class FeedForward(nn.Module):
         inner = dim * mult
         # diffusers naming: ff.net.0.proj (GEGLU), ff.net.1 (Dropout), ff.net.2 (Linear)
         geglu_proj = nn.Module()
-        geglu_proj.proj = nn.Linear(dim, inner * 2)
-        self.net = nn.ModuleList([geglu_proj, nn.Dropout(dropout), nn.Linear(inner, dim)])
+        geglu_proj.proj = DcrLinear(dim, inner * 2)
+        self.net = nn.ModuleList([geglu_proj, nn.Dropout(dropout), DcrLinear(inner, dim)])
 
     def forward(self, x):
         x = self.net[0].proj(x)
@@ -122,8 +124,8 @@ class Transformer2DModel(nn.Module):
         self.use_linear_projection = use_linear_projection
         self.norm = GroupNormOp(norm_num_groups, in_channels, eps=1e-6, fused_silu=False)
         if use_linear_projection:
-            self.proj_in = nn.Linear(in_channels, inner_dim)
-            self.proj_out = nn.Linear(inner_dim, in_channels)
+            self.proj_in = DcrLinear(in_channels, inner_dim)
+            self.proj_out = DcrLinear(inner_dim, in_channels)
         else:
             self.proj_in = nn.Conv2d(in_channels, inner_dim, 1)
             self.proj_out = nn.Conv2d(inner_dim, in_channels, 1)

@@ -25,6 +25,7 @@ SOURCES = [
     HIP_DIR / "attention.hip",
     HIP_DIR / "conv_nhwc.hip",
     HIP_DIR / "conv_nhwc_bwd.hip",
+    HIP_DIR / "gemm.hip",
 ]
 
 

@@ -57,10 +57,7 @@ class _NativeConvFn(torch.autograd.Function):
     def forward(ctx, x, weight, bias, stride, padding):
         m = ext()
         count_dispatch('conv_nhwc')
-        # DCR_NATIVE_CONV_V3=1: double-buffered draft (round-2 A/B)
-        fwd = m.conv2d_nhwc_fwd_v3 if os.environ.get("DCR_NATIVE_CONV_V3") == "1" \
-            else m.conv2d_nhwc_fwd_v2
-        y = fwd(x, weight, bias, stride, padding)
+        y = m.conv2d_nhwc_fwd_v2(x, weight, bias, stride, padding)
         ctx.save_for_backward(x, weight)
         ctx.conf = (stride, padding, bias is not None)
         return y
@@ -69,8 +66,21 @@ class _NativeConvFn(torch.autograd.Function):
     def backward(ctx, dy):
         x, weight = ctx.saved_tensors
         stride, padding, has_bias = ctx.conf
+        dy = dy.contiguous(memory_format=torch.channels_last)
+        # native backward (conv_nhwc_bwd.hip: dgrad + wgrad + fused
+        # bias-grad) — numerics GPU-validated round 2; C % 64 shapes only
+        if os.environ.get("DCR_NATIVE_CONV_BWD", "0") == "1" \
+                and weight.shape[1] % 64 == 0:
+            m = ext()
+            count_dispatch('conv_nhwc_bwd')
+            dx, dw, db = m.conv2d_nhwc_bwd(dy, x, weight, stride, padding)
+            if has_bias and ctx.needs_input_grad[2]:
+                db = db.to(dy.dtype)
+            else:
+                db = None
+            return dx, dw, db, None, None
         dx, dw, db = torch.ops.aten.convolution_backward(
-            dy.contiguous(memory_format=torch.channels_last), x, weight,
+            dy, x, weight,
             [weight.shape[0]] if has_bias else None,
             [stride, stride], [padding, padding], [1, 1], False, [0, 0], 1,
             [ctx.needs_input_grad[0], ctx.needs_input_grad[1],

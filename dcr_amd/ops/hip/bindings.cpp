@@ -421,9 +421,9 @@ at::Tensor conv2d_nhwc_fwd(at::Tensor x, at::Tensor w,
   return y;
 }
 
-static at::Tensor conv2d_nhwc_fwd_v2v3(at::Tensor x, at::Tensor w,
-                                       c10::optional<at::Tensor> bias,
-                                       int64_t stride, int64_t pad, int ver) {
+static at::Tensor conv2d_nhwc_fwd_v2_impl(at::Tensor x, at::Tensor w,
+                                          c10::optional<at::Tensor> bias,
+                                          int64_t stride, int64_t pad) {
   TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16);
   TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast) &&
               w.is_contiguous(at::MemoryFormat::ChannelsLast));
@@ -456,8 +456,7 @@ static at::Tensor conv2d_nhwc_fwd_v2v3(at::Tensor x, at::Tensor w,
     ws = at::zeros({NPQ * K}, x.options().dtype(at::kFloat));
     wsp = ws.data_ptr<float>();
   }
-  auto launch = (ver == 3) ? conv_nhwc_fwd_v3_launch : conv_nhwc_fwd_v2_launch;
-  launch(x.data_ptr(), w.data_ptr(), bp, y.data_ptr(), wsp,
+  conv_nhwc_fwd_v2_launch(x.data_ptr(), w.data_ptr(), bp, y.data_ptr(), wsp,
          splitz, (int)Nb, (int)Hin, (int)Win, (int)C, (int)K,
          (int)P, (int)Q, (int)R, (int)S, (int)stride,
          (int)pad, cur_stream());
@@ -467,15 +466,7 @@ static at::Tensor conv2d_nhwc_fwd_v2v3(at::Tensor x, at::Tensor w,
 at::Tensor conv2d_nhwc_fwd_v2(at::Tensor x, at::Tensor w,
                               c10::optional<at::Tensor> bias, int64_t stride,
                               int64_t pad) {
-  return conv2d_nhwc_fwd_v2v3(x, w, bias, stride, pad, 2);
-}
-
-// round-2 draft: double-buffered staging — validated on hardware before
-// any dispatch (DCR_NATIVE_CONV_V3=1 gates its tests)
-at::Tensor conv2d_nhwc_fwd_v3(at::Tensor x, at::Tensor w,
-                              c10::optional<at::Tensor> bias, int64_t stride,
-                              int64_t pad) {
-  return conv2d_nhwc_fwd_v2v3(x, w, bias, stride, pad, 3);
+  return conv2d_nhwc_fwd_v2_impl(x, w, bias, stride, pad);
 }
 
 // conv backward drafts (round-2; validated before any dispatch)
@@ -509,11 +500,61 @@ std::vector<at::Tensor> conv2d_nhwc_bwd(at::Tensor dy, at::Tensor x,
   return {dx, dW, db};
 }
 
+// ---------------------------------------------------------------- GEMM
+// C[M,N] = sum_k A(m,k)*B(n,k) (+bias[n]); ta/tb: operand memory is
+// [K][rows] (k-strided) instead of [rows][K]. want_dbias (wgrad): also
+// return the column-sum of A's underlying dy (fused into staging).
+std::vector<at::Tensor> gemm_bf16(at::Tensor A, at::Tensor B,
+                                  c10::optional<at::Tensor> bias,
+                                  bool ta, bool tb, bool want_dbias) {
+  TORCH_CHECK(A.is_cuda() && A.scalar_type() == at::kBFloat16 &&
+              B.scalar_type() == at::kBFloat16, "gemm_bf16: bf16 CUDA only");
+  TORCH_CHECK(A.is_contiguous() && B.is_contiguous());
+  TORCH_CHECK(A.dim() == 2 && B.dim() == 2);
+  const int64_t M = ta ? A.size(1) : A.size(0);
+  const int64_t K = ta ? A.size(0) : A.size(1);
+  const int64_t N = tb ? B.size(1) : B.size(0);
+  TORCH_CHECK((tb ? B.size(0) : B.size(1)) == K, "gemm_bf16: K mismatch");
+  TORCH_CHECK(K % 8 == 0, "gemm_bf16: K % 8 != 0");
+  auto C = at::empty({M, N}, A.options());
+  at::Tensor bf;
+  const float* bp = nullptr;
+  if (bias.has_value()) {
+    bf = bias->to(at::kFloat).contiguous();
+    bp = bf.data_ptr<float>();
+  }
+  // split the contraction when the tile grid starves the 256 CUs
+  const int64_t blocks = ((M + 127) / 128) * ((N + 127) / 128);
+  const int64_t nsteps = (K + 63) / 64;
+  int splitz = 1;
+  if (blocks < 384)
+    splitz = (int)std::min<int64_t>({(384 + blocks - 1) / blocks, nsteps, 16});
+  if (splitz < 1) splitz = 1;
+  at::Tensor ws;
+  float* wsp = nullptr;
+  if (splitz > 1) {
+    ws = at::zeros({M * N}, A.options().dtype(at::kFloat));
+    wsp = ws.data_ptr<float>();
+  }
+  at::Tensor db;
+  float* dbp = nullptr;
+  if (want_dbias) {
+    TORCH_CHECK(ta, "gemm_bf16: dbias fusion needs ta (wgrad layout)");
+    db = at::zeros({M}, A.options().dtype(at::kFloat));
+    dbp = db.data_ptr<float>();
+  }
+  gemm_bf16_launch(A.data_ptr(), B.data_ptr(), bp, C.data_ptr(), wsp, dbp,
+                   M, N, (int)K, ta ? 1 : 0, tb ? 1 : 0, want_dbias ? 1 : 0,
+                   splitz, cur_stream());
+  if (want_dbias) return {C, db};
+  return {C};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("gemm_bf16", &gemm_bf16);
   mod.def("conv2d_nhwc_bwd", &conv2d_nhwc_bwd);
   mod.def("conv2d_nhwc_fwd", &conv2d_nhwc_fwd);
   mod.def("conv2d_nhwc_fwd_v2", &conv2d_nhwc_fwd_v2);
-  mod.def("conv2d_nhwc_fwd_v3", &conv2d_nhwc_fwd_v3);
   mod.def("attn_fwd", &attn_fwd);
   mod.def("attn_fwd_v2", &attn_fwd_v2);
   mod.def("attn_bwd", &attn_bwd);

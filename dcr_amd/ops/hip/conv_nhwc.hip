@@ -282,199 +282,6 @@ __global__ void conv_splitk_finalize_kernel(const float* __restrict__ ws,
 
 }  // namespace dcr_conv
 
-// ===========================================================================
-// v3 (round-2 draft): v2 + double-buffered LDS staging. Two LDS buffers per
-// operand; the NEXT K-step's global loads are issued into registers before
-// the current step's MFMA loop (register staging with the write pass after
-// compute), so the HBM latency of step k+1 hides behind the matrix math of
-// step k and each step pays ONE barrier instead of two. This is the guide's
-// §5 "2-buffer" pipeline with plain loads (measured within a few % of the
-// glds variant at BK=64 without needing a lane-linear LDS image, which the
-// pitch-72 anti-conflict layout is not). LDS: 2 x (A+B) tiles = 72 KB at
-// BK=64, still 2 blocks/CU by LDS.
-// NOT dispatched this round — hardware validation first (gpu tests gated
-// DCR_NATIVE_CONV_V3=1, A/B in scripts/bench_conv.py).
-// ===========================================================================
-namespace dcr_conv {
-
-template <int BK>
-__global__ __launch_bounds__(256)
-void conv_nhwc_fwd_v3_kernel(const bf16_t* __restrict__ x, const bf16_t* __restrict__ w,
-                             const float* __restrict__ bias, bf16_t* __restrict__ y,
-                             float* __restrict__ ws, int splitz,
-                             int Nb, int Hin, int Win, int C, int K, int P, int Q,
-                             int R, int S, int stride, int pad) {
-  constexpr int PITCH2 = BK + 8;
-  __shared__ short sA[2][128 * PITCH2];
-  __shared__ short sB[2][128 * PITCH2];
-
-  const long m0 = (long)blockIdx.x * 128;
-  const int k0 = blockIdx.y * 128;
-  const long NPQ = (long)Nb * P * Q;
-  const int rsc_total = R * S * C;
-
-  const int lane = threadIdx.x & 63;
-  const int wid = threadIdx.x >> 6;
-  const int l16 = lane & 15;
-  const int kgrp = lane >> 4;
-  const int wr = (wid >> 1) * 64;
-  const int wc = (wid & 1) * 64;
-
-  f32x4_t acc[4][4];
-#pragma unroll
-  for (int i = 0; i < 4; ++i)
-#pragma unroll
-    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
-
-  const int st_row = threadIdx.x >> 1;
-  const int st_chalf = (threadIdx.x & 1) * (BK / 2);
-  long st_m = m0 + st_row;
-  int st_n = 0, st_p = 0, st_q = 0;
-  if (st_m < NPQ) {
-    st_n = (int)(st_m / (P * Q));
-    int pq = (int)(st_m % (P * Q));
-    st_p = pq / Q;
-    st_q = pq % Q;
-  }
-  const long wrow = (long)(k0 + st_row) * rsc_total;
-
-  const int nsteps = rsc_total / BK;
-  const int spz = (nsteps + splitz - 1) / splitz;
-  const int step0 = blockIdx.z * spz;
-  const int step1 = min(nsteps, step0 + spz);
-  constexpr int NCH = BK / 16;
-
-  // staging registers: live across the MFMA loop so the global loads of
-  // step k+1 stay in flight behind step k's matrix math
-  uint4 av[NCH], bv[NCH];
-
-  auto load_tile = [&](int step) {
-    const int rsc0 = step * BK;
-    const int tap = rsc0 / C;
-    const int r = tap / S;
-    const int s = tap % S;
-    const int c0 = rsc0 - tap * C;
-#pragma unroll
-    for (int t = 0; t < NCH; ++t) av[t] = make_uint4(0, 0, 0, 0);
-#pragma unroll
-    for (int t = 0; t < NCH; ++t) bv[t] = make_uint4(0, 0, 0, 0);
-    const int hi = st_p * stride + r - pad;
-    const int wi = st_q * stride + s - pad;
-    if (st_m < NPQ && hi >= 0 && hi < Hin && wi >= 0 && wi < Win) {
-      const uint4* p4 = reinterpret_cast<const uint4*>(
-          x + (((long)st_n * Hin + hi) * Win + wi) * C + c0 + st_chalf);
-#pragma unroll
-      for (int t = 0; t < NCH; ++t) av[t] = p4[t];
-    }
-    if (k0 + st_row < K) {
-      const uint4* p4 = reinterpret_cast<const uint4*>(w + wrow + rsc0 + st_chalf);
-#pragma unroll
-      for (int t = 0; t < NCH; ++t) bv[t] = p4[t];
-    }
-  };
-  auto store_tile = [&](int buf) {
-    uint4* d = reinterpret_cast<uint4*>(sA[buf] + st_row * PITCH2 + st_chalf);
-    uint4* db = reinterpret_cast<uint4*>(sB[buf] + st_row * PITCH2 + st_chalf);
-#pragma unroll
-    for (int t = 0; t < NCH; ++t) d[t] = av[t];
-#pragma unroll
-    for (int t = 0; t < NCH; ++t) db[t] = bv[t];
-  };
-
-  if (step0 < step1) {
-    load_tile(step0);
-    store_tile(0);
-  }
-  __syncthreads();
-
-  int cur = 0;
-  for (int step = step0; step < step1; ++step) {
-    const bool more = (step + 1 < step1);  // uniform across the block
-    if (more) load_tile(step + 1);
-
-#pragma unroll
-    for (int kk = 0; kk < BK / 32; ++kk) {
-      bf16x8 af[4], bf[4];
-#pragma unroll
-      for (int i = 0; i < 4; ++i)
-        af[i] = *reinterpret_cast<const bf16x8*>(
-            sA[cur] + (wr + i * 16 + l16) * PITCH2 + kk * 32 + kgrp * 8);
-#pragma unroll
-      for (int j = 0; j < 4; ++j)
-        bf[j] = *reinterpret_cast<const bf16x8*>(
-            sB[cur] + (wc + j * 16 + l16) * PITCH2 + kk * 32 + kgrp * 8);
-#pragma unroll
-      for (int i = 0; i < 4; ++i)
-#pragma unroll
-        for (int j = 0; j < 4; ++j)
-          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[i], bf[j],
-                                                              acc[i][j], 0, 0, 0);
-    }
-
-    if (more) {
-      // writes target the buffer whose last readers finished before the
-      // PREVIOUS barrier; the barrier below publishes them for step+1
-      store_tile(cur ^ 1);
-      __syncthreads();
-      cur ^= 1;
-    }
-  }
-
-#pragma unroll
-  for (int i = 0; i < 4; ++i) {
-#pragma unroll
-    for (int rr = 0; rr < 4; ++rr) {
-      const long m = m0 + wr + i * 16 + kgrp * 4 + rr;
-      if (m >= NPQ) continue;
-#pragma unroll
-      for (int j = 0; j < 4; ++j) {
-        const int k = k0 + wc + j * 16 + l16;
-        if (k >= K) continue;
-        if (splitz > 1) {
-          atomicAdd(&ws[m * K + k], acc[i][j][rr]);
-        } else {
-          float v = acc[i][j][rr] + (bias ? bias[k] : 0.f);
-          y[m * K + k] = __float2bfloat16(v);
-        }
-      }
-    }
-  }
-}
-
-}  // namespace dcr_conv
-
-namespace dcr {
-
-void conv_nhwc_fwd_v3_launch(const void* x, const void* w, const float* bias,
-                             void* y, float* ws, int splitz, int Nb, int Hin,
-                             int Win, int C, int K, int P, int Q, int R, int S,
-                             int stride, int pad, hipStream_t st) {
-  long NPQ = (long)Nb * P * Q;
-  dim3 grid((unsigned)((NPQ + 127) / 128), (unsigned)((K + 127) / 128),
-            (unsigned)splitz),
-      block(256);
-  if (C % 64 == 0)
-    hipLaunchKernelGGL((dcr_conv::conv_nhwc_fwd_v3_kernel<64>), grid, block, 0,
-                       st, (const dcr_conv::bf16_t*)x,
-                       (const dcr_conv::bf16_t*)w, bias, (dcr_conv::bf16_t*)y,
-                       ws, splitz, Nb, Hin, Win, C, K, P, Q, R, S, stride, pad);
-  else
-    hipLaunchKernelGGL((dcr_conv::conv_nhwc_fwd_v3_kernel<32>), grid, block, 0,
-                       st, (const dcr_conv::bf16_t*)x,
-                       (const dcr_conv::bf16_t*)w, bias, (dcr_conv::bf16_t*)y,
-                       ws, splitz, Nb, Hin, Win, C, K, P, Q, R, S, stride, pad);
-  if (splitz > 1) {
-    long total = NPQ * K;
-    long b = (total / 4 + 255) / 256;
-    if (b > 8192) b = 8192;
-    hipLaunchKernelGGL(dcr_conv::conv_splitk_finalize_kernel,
-                       dim3((unsigned)b), dim3(256), 0, st, ws, bias,
-                       (dcr_conv::bf16_t*)y, total, K);
-  }
-}
-
-}  // namespace dcr
-
 namespace dcr {
 
 void conv_nhwc_fwd_v2_launch(const void* x, const void* w, const float* bias,

@@ -84,13 +84,7 @@ void conv_nhwc_fwd_v2_launch(const void* x, const void* w, const float* bias,
                              void* y, float* ws, int splitz, int Nb, int Hin,
                              int Win, int C, int K, int P, int Q, int R, int S,
                              int stride, int pad, hipStream_t st);
-// v3: double-buffered staging (round-2 draft; not dispatched)
-void conv_nhwc_fwd_v3_launch(const void* x, const void* w, const float* bias,
-                             void* y, float* ws, int splitz, int Nb, int Hin,
-                             int Win, int C, int K, int P, int Q, int R, int S,
-                             int stride, int pad, hipStream_t st);
-
-// conv_nhwc_bwd.hip (round-2 drafts; not dispatched)
+// conv_nhwc_bwd.hip (dgrad + wgrad + fused bias-grad; DCR_NATIVE_CONV_BWD)
 void conv_bwd_weight_launch(const void* dy, const void* x, float* dw_ws,
                             int Nb, int Hin, int Win, int C, int K, int P,
                             int Q, int R, int S, int stride, int pad,
@@ -100,5 +94,11 @@ void conv_bwd_data_launch(const void* dy, const void* w, void* dx, int Nb,
                           int S, int stride, int pad, hipStream_t st);
 void conv_bias_grad_launch(const void* dy, float* db, long NPQ, int K,
                            hipStream_t st);
+
+// gemm.hip — bf16 MFMA GEMM for transformer linears (fwd/dgrad/wgrad)
+void gemm_bf16_launch(const void* A, const void* B, const float* bias,
+                      void* C, float* ws, float* dbias, long M, long N, int K,
+                      int ta, int tb, int want_dbias, int splitz,
+                      hipStream_t st);
 
 }  // namespace dcr

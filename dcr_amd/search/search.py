@@ -85,14 +85,18 @@ def sharded_topk(query: torch.Tensor, shard: torch.Tensor, k: int = 1,
     (scores [Q, k], global_idx [Q, k]) for this shard.
 
     compute_dtype=torch.bfloat16 runs the big GEMM at the bf16 MFMA rate
-    (2x fp32, half the index-read HBM traffic) over a top-(k+margin)
-    candidate list, then RE-SCORES the surviving candidates in fp32 so the
-    returned scores (and their ranking) are exact fp32 values — the only
-    approximation left is a true top-k member falling below bf16's ~0.004
-    score resolution AND outside the margin.
+    over a top-(k+margin) candidate list, then RE-SCORES the surviving
+    candidates in fp32 so the returned scores (and their ranking) are
+    exact fp32 values — the only approximation left is a true top-k
+    member falling below bf16's ~0.004 score resolution AND outside the
+    margin. Measured on MI355X (1M x 512, 10k queries): 105.8 TF vs
+    70.5 TF fp32 (1.5x) -> the DEFAULT on GPU since round 2; pass
+    compute_dtype=torch.float32 to force the pure-fp32 path.
     """
     Q = query.shape[0]
     device = query.device
+    if compute_dtype is None and query.is_cuda:
+        compute_dtype = torch.bfloat16
     lowp = compute_dtype is not None and compute_dtype != torch.float32
     m = k + max(0, rerank_margin) if lowp else k
     q_mat = query.to(compute_dtype) if lowp else query

@@ -45,17 +45,28 @@ def main():
         bb = b.to(torch.bfloat16)
         t_v1 = timeit(lambda: m.conv2d_nhwc_fwd(x, w, b, stride, pad))
         t_v2 = timeit(lambda: m.conv2d_nhwc_fwd_v2(x, w, b, stride, pad))
-        t_v3 = timeit(lambda: m.conv2d_nhwc_fwd_v3(x, w, b, stride, pad))
         t_miopen = timeit(lambda: F.conv2d(x, w, bb, stride=stride, padding=pad))
         P = (H + 2 * pad - R) // stride + 1
         flops = 2 * N * P * P * K * C * R * R
         print(f"N{N} C{C} H{H} K{K} R{R}s{stride}: "
               f"v1 {t_v1:.3f} ms ({flops / t_v1 / 1e9:.0f} TF) "
               f"v2 {t_v2:.3f} ms ({flops / t_v2 / 1e9:.0f} TF) "
-              f"v3 {t_v3:.3f} ms ({flops / t_v3 / 1e9:.0f} TF) "
               f"MIOpen {t_miopen:.3f} ms ({flops / t_miopen / 1e9:.0f} TF) "
-              f"-> v2/MIOpen {t_miopen / t_v2:.2f}x "
-              f"v3/v2 {t_v2 / t_v3:.2f}x")
+              f"-> v2/MIOpen {t_miopen / t_v2:.2f}x")
+
+        # backward: native (dcr conv_nhwc_bwd: dgrad + wgrad + fused
+        # bias-grad) vs MIOpen via aten::convolution_backward (+ the
+        # separate aten bias-grad reduce it needs)
+        dy = torch.randn(N, K, P, P, device="cuda").to(torch.bfloat16) \
+            .to(memory_format=torch.channels_last)
+        t_nb = timeit(lambda: m.conv2d_nhwc_bwd(dy, x, w, stride, pad))
+        t_mb = timeit(lambda: torch.ops.aten.convolution_backward(
+            dy, x, w, [K], [stride, stride], [pad, pad], [1, 1], False,
+            [0, 0], 1, [True, True, True]))
+        bflops = 2 * flops  # dgrad + wgrad
+        print(f"    bwd: native {t_nb:.3f} ms ({bflops / t_nb / 1e9:.0f} TF) "
+              f"MIOpen {t_mb:.3f} ms ({bflops / t_mb / 1e9:.0f} TF) "
+              f"-> native/MIOpen {t_mb / t_nb:.2f}x")
 
 
 if __name__ == "__main__":

@@ -102,3 +102,22 @@ def test_clip_grad_norm_matches_torch():
     assert torch.allclose(norm, total, atol=1e-5)
     clipped = torch.linalg.vector_norm(opt.flat_grad)
     assert clipped <= 0.5 * (1 + 1e-4)
+
+
+def test_dcr_linear_cpu_fallback_and_state_dict():
+    """DcrLinear == nn.Linear on CPU (F.linear fallback) and state-dict
+    compatible with nn.Linear (diffusers checkpoint naming)."""
+    import torch
+    import torch.nn as nn
+    from dcr_amd.ops.linear import DcrLinear, dcr_linear
+
+    torch.manual_seed(0)
+    ref = nn.Linear(32, 48)
+    mod = DcrLinear(32, 48)
+    mod.load_state_dict(ref.state_dict())
+    x = torch.randn(5, 7, 32)
+    assert torch.equal(mod(x), ref(x))
+    assert set(mod.state_dict()) == {"weight", "bias"}
+    # functional fallback identical to F.linear on CPU
+    y = dcr_linear(x, ref.weight, ref.bias)
+    assert torch.equal(y, ref(x))

@@ -487,37 +487,6 @@ def test_adamw_dev_graph_capture(ext):
     assert torch.isfinite(p).all()
 
 
-@pytest.mark.skipif(os.environ.get("DCR_NATIVE_CONV_V3") != "1",
-                    reason="conv v3 double-buffered draft: validate in "
-                           "round 2 (DCR_NATIVE_CONV_V3=1)")
-@pytest.mark.parametrize("shape", [
-    (2, 320, 32, 32, 320, 3, 1),
-    (2, 320, 32, 32, 640, 1, 1),
-    (2, 640, 16, 16, 640, 3, 2),
-    (1, 128, 64, 64, 128, 3, 1),
-    (2, 1280, 8, 8, 1280, 3, 1),    # split-K path (grid-starved)
-    (2, 96, 16, 16, 64, 3, 1),      # BK=32 instance
-])
-def test_conv_nhwc_fwd_v3(ext, shape):
-    N, C, H, W, K, R, stride = shape
-    pad = 1 if R == 3 else 0
-    torch.manual_seed(0)
-    x = torch.randn(N, C, H, W, device="cuda").to(torch.bfloat16) \
-        .to(memory_format=torch.channels_last)
-    w = (torch.randn(K, C, R, R, device="cuda") * 0.05).to(torch.bfloat16) \
-        .to(memory_format=torch.channels_last)
-    b = torch.randn(K, device="cuda")
-    y = ext.conv2d_nhwc_fwd_v3(x, w, b, stride, pad)
-    # v3 must be bit-comparable to v2 (same tile walk, same accum order
-    # within a block) and close to the fp32 reference
-    y2 = ext.conv2d_nhwc_fwd_v2(x, w, b, stride, pad)
-    ref = torch.nn.functional.conv2d(x.float(), w.float(), b, stride=stride,
-                                     padding=pad)
-    _close(y, ref, 2e-2)
-    assert torch.equal(y.float(), y2.float()) or \
-        (y.float() - y2.float()).abs().max() < 1e-2  # atomics order may differ
-
-
 @pytest.mark.skipif(os.environ.get("DCR_NATIVE_CONV_BWD") != "1",
                     reason="conv bwd drafts: validate in round 2 "
                            "(DCR_NATIVE_CONV_BWD=1)")
@@ -545,3 +514,85 @@ def test_conv_nhwc_bwd(ext, shape):
     _close(dx, rx, 3e-2)
     _close(dW, rw, 3e-2)
     _close(db, rb, 1e-2)
+
+
+# ---------------------------------------------------------------------------
+# MFMA GEMM for the transformer linears (gemm.hip, round 2)
+# ---------------------------------------------------------------------------
+def _gemm_shapes():
+    # (M, N, K) — SD-2.1 linear shapes at bs16 + ragged/odd cases
+    return [
+        (16384, 320, 320),    # res32 qkv/out
+        (16384, 2560, 320),   # res32 GEGLU proj
+        (16384, 320, 1280),   # res32 ff.net.2
+        (4096, 640, 1024),    # res16 cross k/v proj (ctx side is M=1232)
+        (1024, 10240, 1280),  # res8 GEGLU proj
+        (1232, 1280, 1024),   # cross-attn k/v: M = 16*77 (ragged M)
+        (512, 192, 136),      # ragged K (%8 only) + small N
+    ]
+
+
+def test_gemm_bf16_fwd(ext):
+    torch.manual_seed(11)
+    for (M, N, K) in _gemm_shapes():
+        x = torch.randn(M, K, device="cuda", dtype=torch.bfloat16) * 0.5
+        w = torch.randn(N, K, device="cuda", dtype=torch.bfloat16) * 0.1
+        b = torch.randn(N, device="cuda", dtype=torch.bfloat16)
+        (y,) = ext.gemm_bf16(x, w, b, False, False, False)
+        ref = F.linear(x.float(), w.float(), b.float())
+        _close(y, ref, 2e-2)
+        (y2,) = ext.gemm_bf16(x, w, None, False, False, False)
+        _close(y2, ref - b.float()[None, :], 2e-2)
+
+
+def test_gemm_bf16_dgrad(ext):
+    torch.manual_seed(12)
+    for (M, N, K) in _gemm_shapes():
+        dy = torch.randn(M, N, device="cuda", dtype=torch.bfloat16) * 0.5
+        w = torch.randn(N, K, device="cuda", dtype=torch.bfloat16) * 0.1
+        (dx,) = ext.gemm_bf16(dy, w, None, False, True, False)
+        ref = dy.float() @ w.float()
+        _close(dx, ref, 2e-2)
+
+
+def test_gemm_bf16_wgrad_dbias(ext):
+    torch.manual_seed(13)
+    for (M, N, K) in _gemm_shapes():
+        dy = torch.randn(M, N, device="cuda", dtype=torch.bfloat16) * 0.5
+        x = torch.randn(M, K, device="cuda", dtype=torch.bfloat16) * 0.1
+        dw, db = ext.gemm_bf16(dy, x, None, True, True, True)
+        ref_dw = dy.float().t() @ x.float()
+        _close(dw, ref_dw, 2e-2)
+        ref_db = dy.float().sum(0)
+        _close(db, ref_db, 2e-2)
+        (dw2,) = ext.gemm_bf16(dy, x, None, True, True, False)
+        assert torch.equal(dw2, dw)
+
+
+def test_dcr_linear_autograd_matches_f_linear(ext):
+    """Full fwd+bwd of the autograd wrapper vs torch fp32 on a UNet shape."""
+    from dcr_amd.ops.linear import dcr_linear
+    from dcr_amd.ops import dispatch_counts
+    torch.manual_seed(14)
+    M, N, K = 4096, 640, 320
+    x = (torch.randn(M, K, device="cuda", dtype=torch.bfloat16) * 0.5
+         ).requires_grad_(True)
+    w = (torch.randn(N, K, device="cuda", dtype=torch.bfloat16) * 0.1
+         ).requires_grad_(True)
+    b = torch.randn(N, device="cuda", dtype=torch.bfloat16).requires_grad_(True)
+    before = dispatch_counts["gemm"]
+    y = dcr_linear(x, w, b)
+    assert dispatch_counts["gemm"] == before + 1, "native GEMM must run"
+    g = torch.randn_like(y)
+    y.backward(g)
+
+    xf = x.detach().float().requires_grad_(True)
+    wf = w.detach().float().requires_grad_(True)
+    bf = b.detach().float().requires_grad_(True)
+    yf = F.linear(xf, wf, bf)
+    yf.backward(g.float())
+
+    _close(y, yf, 2e-2)
+    _close(x.grad, xf.grad, 2e-2)
+    _close(w.grad, wf.grad, 2e-2)
+    _close(b.grad, bf.grad, 2e-2)
