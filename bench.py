@@ -54,7 +54,10 @@ def main():
     if os.environ.get("DCR_CONV_BENCHMARK") == "1":
         torch.backends.cudnn.benchmark = True  # MIOpen find-mode tuning
     use_cuda = torch.cuda.is_available()
-    device = torch.device("cuda", local) if use_cuda else torch.device("cpu")
+    # local % device_count: lets a world-2 smoke run share one leased GPU
+    # (RCCL multi-rank-per-device) without a special path
+    device = torch.device("cuda", local % max(torch.cuda.device_count(), 1)) \
+        if use_cuda else torch.device("cpu")
     if use_cuda:
         torch.cuda.set_device(device)
 
