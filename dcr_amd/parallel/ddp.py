@@ -8,9 +8,10 @@ are sized (default 64 MiB) so per-bucket all-reduce launched DURING
 backward hides under the remaining backward compute, and RCCL spreads
 channels over the links.
 
-Works with FusedAdamW's flat gradient buffer: every param.grad is a view
-into one contiguous arena, so a bucket is a contiguous slice — zero-copy
-all-reduce, no flatten/unflatten traffic. Gradient accumulation parity
+Works with FusedAdamW's flat gradient arena: each bucket hook batch-
+copies its params' autograd grads into the contiguous arena slice
+(gather_grads) and all-reduces that slice — no per-param add kernels,
+no flatten/unflatten traffic. Gradient accumulation parity
 (diff_train.py:618): set `require_backward_grad_sync=False` on non-sync
 micro-steps to skip collectives entirely.
 """
@@ -89,7 +90,10 @@ class GradBucketAllReduce:
             self._param_bucket[id(p)] = b
 
     def _launch(self, b: _Bucket):
-        # SUM then post-scale (bf16 grads: pre-divide would lose mantissa)
+        # batched copy of this bucket's autograd grads into the arena
+        # slice, then SUM all-reduce + post-scale (bf16 grads: pre-divide
+        # would lose mantissa)
+        self.opt.gather_grads(b.params)
         sl = self.opt.flat_grad[b.start:b.end]
         b.work = dist.all_reduce(sl, group=self.pg, async_op=True)
 
@@ -103,6 +107,8 @@ class GradBucketAllReduce:
 
     def finalize(self):
         """Call after backward(), before optimizer.step()."""
+        # any grads not yet moved by a bucket hook (world 1: all of them)
+        self.opt.gather_grads()
         if not self._enabled or not self.require_backward_grad_sync:
             self._reset()
             return

@@ -111,6 +111,7 @@ def test_gradient_accumulation_equivalence(tmp_path):
     p0 = tr1.optimizer.flat_param.clone()
     tr1.train_step(b, sync_gradients=False)
     assert torch.equal(tr1.optimizer.flat_param, p0)
+    tr1.optimizer.gather_grads()
     g_half = tr1.optimizer.flat_grad.clone()
     tr1.train_step(b, sync_gradients=True)
     assert not torch.equal(tr1.optimizer.flat_param, p0)
@@ -131,9 +132,11 @@ def test_gradient_accumulation_is_mean(tmp_path):
     # of two equal grads equals one micro-step's grad at accum=1
     torch.manual_seed(7)
     tr.train_step(b, sync_gradients=False)
+    tr.optimizer.gather_grads()
     g1 = tr.optimizer.flat_grad.clone()
     torch.manual_seed(7)
     tr.train_step(b, sync_gradients=False)
+    tr.optimizer.gather_grads()
     g2 = tr.optimizer.flat_grad.clone()
     # second micro-step added the same scaled grad again: g2 == 2*g1
     assert torch.allclose(g2, 2 * g1, rtol=1e-5, atol=1e-7)
@@ -144,6 +147,7 @@ def test_gradient_accumulation_is_mean(tmp_path):
     tr1.optimizer.flat_param.copy_(tr.optimizer.flat_param)
     torch.manual_seed(7)
     tr1.train_step(b, sync_gradients=False)
+    tr1.optimizer.gather_grads()
     # accum=2 total grad (mean over 2 identical micros) == accum=1 grad
     assert torch.allclose(g2, tr1.optimizer.flat_grad, rtol=1e-5, atol=1e-7)
 
