@@ -565,12 +565,15 @@ def test_gemm_bf16_wgrad_dbias(ext):
         _close(dw, ref_dw, 2e-2)
         ref_db = dy.float().sum(0)
         _close(db, ref_db, 2e-2)
+        # split-K over the contraction uses fp32 atomics, so two runs are
+        # equal only to atomics-reordering rounding (like rocBLAS/MIOpen)
         (dw2,) = ext.gemm_bf16(dy, x, None, True, True, False)
-        assert torch.equal(dw2, dw)
+        _close(dw2, ref_dw, 2e-2)
 
 
 def test_dcr_linear_autograd_matches_f_linear(ext):
-    """Full fwd+bwd of the autograd wrapper vs torch fp32 on a UNet shape."""
+    """Full fwd+bwd of the hybrid wrapper vs torch fp32 on a UNet shape:
+    rocBLAS fwd/dgrad + native MFMA wgrad with fused bias-grad."""
     from dcr_amd.ops.linear import dcr_linear
     from dcr_amd.ops import dispatch_counts
     torch.manual_seed(14)
@@ -580,11 +583,12 @@ def test_dcr_linear_autograd_matches_f_linear(ext):
     w = (torch.randn(N, K, device="cuda", dtype=torch.bfloat16) * 0.1
          ).requires_grad_(True)
     b = torch.randn(N, device="cuda", dtype=torch.bfloat16).requires_grad_(True)
-    before = dispatch_counts["gemm"]
+    before = dispatch_counts["gemm_wgrad"]
     y = dcr_linear(x, w, b)
-    assert dispatch_counts["gemm"] == before + 1, "native GEMM must run"
     g = torch.randn_like(y)
     y.backward(g)
+    assert dispatch_counts["gemm_wgrad"] == before + 1, \
+        "native wgrad must run in its win regime"
 
     xf = x.detach().float().requires_grad_(True)
     wf = w.detach().float().requires_grad_(True)
