@@ -229,6 +229,23 @@ def attention(
                 v.reshape(-1, v.shape[-2], 64).contiguous(),
                 scale, causal)
             return out.reshape(shp)
+        if q.dtype == torch.bfloat16 and q.shape[-1] in (40, 80, 160) \
+                and not (torch.is_grad_enabled() and
+                         (q.requires_grad or k.requires_grad or v.requires_grad)):
+            # SD-1.4 head dims (sd_mitigation / diff_inference sampling,
+            # reference sd_mitigation.py:46) — forward-only HIP kernel
+            m = require_hip("attn_gen")
+            count_dispatch('attention_gen')
+            if layout == "blhd":
+                return m.attn_fwd_gen(q.contiguous(), k.contiguous(),
+                                      v.contiguous(), scale, causal)
+            d = q.shape[-1]
+            shp = q.shape
+            out = m.attn_fwd_gen(
+                q.reshape(-1, shp[-2], d).contiguous(),
+                k.reshape(-1, k.shape[-2], d).contiguous(),
+                v.reshape(-1, v.shape[-2], d).contiguous(), scale, causal)
+            return out.reshape(shp)
         if layout == "blhd":
             out = _attention_math(q.permute(0, 2, 1, 3), k.permute(0, 2, 1, 3),
                                   v.permute(0, 2, 1, 3), scale, causal)

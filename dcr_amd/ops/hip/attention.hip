@@ -661,6 +661,175 @@ __global__ void mfma_probe_kernel(const bf16_t* __restrict__ A,
   for (int r = 0; r < 4; ++r) C[((l >> 4) * 4 + r) * 16 + (l & 15)] = c[r];
 }
 
+// ==========================================================================
+// Generalized-head-dim forward (SD-1.4 / sd_mitigation parity: head_dim
+// 40/80/160 — /root/reference/sd_mitigation.py:46; inference-only, so no
+// backward). Same tile discipline as attn_fwd_kernel with the head dim
+// padded to DP = ceil(D/32)*32 (zero-padded contraction contributes 0 to
+// QK^T) and LDS pitch DP+8 bf16 — every (DP+8)*2 B row stride lands the
+// 16 rows of a ds_read_b128 lane group on 16 distinct banks (stride/4
+// mod 64 has gcd 4 with 64 for DP in {64,96,160}).
+// ==========================================================================
+template <int D, int DP, int GP>
+__global__ __launch_bounds__(256)
+void attn_fwd_gen_kernel(const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
+                         const bf16_t* __restrict__ v, bf16_t* __restrict__ o,
+                         float* __restrict__ lse, int Lq, int Lk, int H,
+                         float scale, int causal) {
+  __shared__ short sQ[TILE * GP];
+  __shared__ short sK[TILE * GP];
+  __shared__ short sVT[DP * PITCH];     // [dim][key], key pitch 72
+  __shared__ short sP[TILE * PITCH];
+
+  const int bh = blockIdx.y;
+  const int b = bh / H, h = bh % H;
+  const int q0 = blockIdx.x * TILE;
+  const long rs = (long)H * D;            // [B,L,H,D] row stride
+  const bf16_t* qp = q + (((long)b * Lq + q0) * H + h) * D;
+  const bf16_t* kp = k + ((long)b * Lk * H + h) * D;
+  const bf16_t* vp = v + ((long)b * Lk * H + h) * D;
+  bf16_t* op = o + ((long)b * Lq * H + h) * D;
+
+  const int t = threadIdx.x;
+  // cooperative [64 rows][DP] load, 4 threads/row, 8-elem units (D%8==0)
+  auto load_rows = [&](const bf16_t* src, int valid, short* dst) {
+    const int row = t >> 2;
+    const bf16_t* sp_ = src + (long)row * rs;
+#pragma unroll
+    for (int c = (t & 3) * 8; c < DP; c += 32) {
+      uint4 a = make_uint4(0, 0, 0, 0);
+      if (row < valid && c + 8 <= D)
+        a = *reinterpret_cast<const uint4*>(sp_ + c);
+      *reinterpret_cast<uint4*>(dst + row * GP + c) = a;
+    }
+  };
+  auto load_vt = [&](const bf16_t* src, int valid) {
+    const int row = t >> 2;
+    const bf16_t* sp_ = src + (long)row * rs;
+#pragma unroll
+    for (int c0 = (t & 3) * 8; c0 < DP; c0 += 32) {
+      short vv[8];
+      if (row < valid && c0 + 8 <= D) {
+        *reinterpret_cast<uint4*>(vv) = *reinterpret_cast<const uint4*>(sp_ + c0);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) vv[j] = 0;
+      }
+#pragma unroll
+      for (int j = 0; j < 8; ++j) sVT[(c0 + j) * PITCH + row] = vv[j];
+    }
+  };
+
+  load_rows(qp, Lq - q0, sQ);
+
+  const int lane = t & 63;
+  const int wid = t >> 6;
+  const int l16 = lane & 15;
+  const int kgrp = lane >> 4;
+  const int wrow0 = wid * 16;
+  constexpr int NK = DP / 32;             // QK^T contraction chunks
+  constexpr int ND = DP / 16;             // PV output d-subtiles
+
+  float row_max[4] = {-1e30f, -1e30f, -1e30f, -1e30f};
+  float row_sum[4] = {0.f, 0.f, 0.f, 0.f};
+  f32x4_t acc_o[ND];
+#pragma unroll
+  for (int i = 0; i < ND; ++i) acc_o[i] = {0.f, 0.f, 0.f, 0.f};
+
+  const int kv_end = causal ? min(Lk, q0 + TILE) : Lk;
+
+  for (int kv0 = 0; kv0 < kv_end; kv0 += TILE) {
+    __syncthreads();
+    load_rows(kp + (long)kv0 * rs, Lk - kv0, sK);
+    load_vt(vp + (long)kv0 * rs, Lk - kv0);
+    __syncthreads();
+
+    bf16x8 qf[NK];
+#pragma unroll
+    for (int ck = 0; ck < NK; ++ck)
+      qf[ck] = *reinterpret_cast<const bf16x8*>(
+          sQ + (wrow0 + l16) * GP + ck * 32 + kgrp * 8);
+    f32x4_t s_frag[4];
+#pragma unroll
+    for (int ns = 0; ns < 4; ++ns) {
+      f32x4_t acc = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int ck = 0; ck < NK; ++ck) {
+        bf16x8 kf = *reinterpret_cast<const bf16x8*>(
+            sK + (ns * 16 + l16) * GP + ck * 32 + kgrp * 8);
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[ck], kf, acc, 0, 0, 0);
+      }
+      s_frag[ns] = acc;
+    }
+
+    float tile_max[4] = {-1e30f, -1e30f, -1e30f, -1e30f};
+#pragma unroll
+    for (int ns = 0; ns < 4; ++ns) {
+      const int key = kv0 + ns * 16 + l16;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float sv = s_frag[ns][r] * scale;
+        const int qrow = q0 + wrow0 + kgrp * 4 + r;
+        if (key >= Lk || (causal && key > qrow)) sv = -1e30f;
+        s_frag[ns][r] = sv;
+        tile_max[r] = fmaxf(tile_max[r], sv);
+      }
+    }
+    float alpha[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      float tm = qmax(tile_max[r]);
+      float mnew = fmaxf(row_max[r], tm);
+      alpha[r] = (mnew <= -1e29f) ? 1.f : __expf(row_max[r] - mnew);
+      row_max[r] = mnew;
+    }
+
+    float psum[4] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int ns = 0; ns < 4; ++ns) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float pv = (s_frag[ns][r] <= -1e29f)
+                       ? 0.f : __expf(s_frag[ns][r] - row_max[r]);
+        psum[r] += pv;
+        sP[(wrow0 + kgrp * 4 + r) * PITCH + ns * 16 + l16] = f2bf_rne(pv);
+      }
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) row_sum[r] = row_sum[r] * alpha[r] + qsum(psum[r]);
+
+#pragma unroll
+    for (int ds_ = 0; ds_ < ND; ++ds_)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) acc_o[ds_][r] *= alpha[r];
+
+    bf16x8 pf0 = frag(sP, wrow0 + l16, kgrp * 8);
+    bf16x8 pf1 = frag(sP, wrow0 + l16, kgrp * 8 + 32);
+#pragma unroll
+    for (int ds_ = 0; ds_ < ND; ++ds_) {
+      acc_o[ds_] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          pf0, frag(sVT, ds_ * 16 + l16, kgrp * 8), acc_o[ds_], 0, 0, 0);
+      acc_o[ds_] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          pf1, frag(sVT, ds_ * 16 + l16, kgrp * 8 + 32), acc_o[ds_], 0, 0, 0);
+    }
+  }
+
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int qrow = q0 + wrow0 + kgrp * 4 + r;
+    if (qrow >= Lq) continue;
+    const float inv = (row_sum[r] > 0.f) ? 1.f / row_sum[r] : 0.f;
+#pragma unroll
+    for (int ds_ = 0; ds_ < ND; ++ds_) {
+      const int d = ds_ * 16 + l16;
+      if (d < D)
+        op[(long)qrow * rs + d] = __float2bfloat16(acc_o[ds_][r] * inv);
+    }
+    if (l16 == 0 && lse != nullptr)
+      lse[(long)bh * Lq + qrow] = row_max[r] + __logf(fmaxf(row_sum[r], 1e-30f));
+  }
+}
+
 }  // namespace dcr_attn
 
 // ==========================================================================
@@ -713,6 +882,34 @@ void attn_bwd_launch(const void* q, const void* k, const void* v,
                      (const dcr_attn::bf16_t*)v, (const dcr_attn::bf16_t*)dO,
                      lse, delta, (dcr_attn::bf16_t*)dQ, Lq, Lk, H, scale,
                      causal ? 1 : 0);
+}
+
+// generalized head-dim forward (SD-1.4 40/80/160; inference-only)
+void attn_fwd_gen_launch(const void* q, const void* k, const void* v, void* o,
+                         float* lse, int BH, int Lq, int Lk, int H, int D,
+                         float scale, bool causal, hipStream_t s) {
+  dim3 grid((Lq + TILE - 1) / TILE, BH), block(256);
+  const auto* qq = (const dcr_attn::bf16_t*)q;
+  const auto* kk = (const dcr_attn::bf16_t*)k;
+  const auto* vv = (const dcr_attn::bf16_t*)v;
+  auto* oo = (dcr_attn::bf16_t*)o;
+  const int c = causal ? 1 : 0;
+  switch (D) {
+    case 40:
+      hipLaunchKernelGGL((dcr_attn::attn_fwd_gen_kernel<40, 64, 72>), grid,
+                         block, 0, s, qq, kk, vv, oo, lse, Lq, Lk, H, scale, c);
+      break;
+    case 80:
+      hipLaunchKernelGGL((dcr_attn::attn_fwd_gen_kernel<80, 96, 104>), grid,
+                         block, 0, s, qq, kk, vv, oo, lse, Lq, Lk, H, scale, c);
+      break;
+    case 160:
+      hipLaunchKernelGGL((dcr_attn::attn_fwd_gen_kernel<160, 160, 168>), grid,
+                         block, 0, s, qq, kk, vv, oo, lse, Lq, Lk, H, scale, c);
+      break;
+    default:
+      break;  // binding guards D
+  }
 }
 
 void mfma_probe_launch(const void* A, const void* B, float* C, hipStream_t s) {

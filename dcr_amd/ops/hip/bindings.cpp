@@ -349,6 +349,31 @@ std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
   return {o, lse};
 }
 
+// generalized head-dim forward (SD-1.4 40/80/160; inference-only — no
+// backward: sd_mitigation / diff_inference sampling runs under no_grad)
+at::Tensor attn_fwd_gen(at::Tensor q, at::Tensor k, at::Tensor v,
+                        double scale, bool causal) {
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16,
+              "attn_gen: bf16 CUDA only");
+  TORCH_CHECK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
+  const int64_t D = q.size(-1);
+  TORCH_CHECK(D == 40 || D == 80 || D == 160,
+              "attn_gen: head_dim 40/80/160 only (64 has the main kernel)");
+  int64_t B, H, Lq, Lk;
+  if (q.dim() == 4) {
+    B = q.size(0); Lq = q.size(1); H = q.size(2); Lk = k.size(1);
+    TORCH_CHECK(k.dim() == 4 && k.size(2) == H && k.size(-1) == D);
+  } else {
+    TORCH_CHECK(q.dim() == 3);
+    B = q.size(0); Lq = q.size(1); H = 1; Lk = k.size(1);
+  }
+  auto o = at::empty_like(q);
+  attn_fwd_gen_launch(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
+                      nullptr, (int)(B * H), (int)Lq, (int)Lk, (int)H, (int)D,
+                      (float)scale, causal, cur_stream());
+  return o;
+}
+
 // round-2 draft: bit-exact masked-tail skip — validated vs attn_fwd on
 // hardware before any dispatch (DCR_ATTN_V2=1 gates its tests)
 std::vector<at::Tensor> attn_fwd_v2(at::Tensor q, at::Tensor k, at::Tensor v,
@@ -556,6 +581,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("conv2d_nhwc_fwd", &conv2d_nhwc_fwd);
   mod.def("conv2d_nhwc_fwd_v2", &conv2d_nhwc_fwd_v2);
   mod.def("attn_fwd", &attn_fwd);
+  mod.def("attn_fwd_gen", &attn_fwd_gen);
   mod.def("attn_fwd_v2", &attn_fwd_v2);
   mod.def("attn_bwd", &attn_bwd);
   mod.def("mfma_probe", &mfma_probe);

@@ -9,8 +9,12 @@
 //            LDS per-channel sums -> per-group partials -> global atomics
 //            into a [N, G, 2] fp32 workspace. No per-element atomics.
 //   finalize:[N*G] -> mean/rstd.
-//   apply:   flat vec4 elementwise (fully coalesced NHWC walk).
+//   apply:   flat vecV elementwise (fully coalesced NHWC walk).
 // Backward mirrors it (stats also emit per-channel dw/db).
+//
+// All four streaming kernels are templated on the vector width V: 8
+// bf16/f16 elements = 16 B/lane (the HBM3E coalescing sweet spot, guide
+// G13) whenever C % 8 == 0 (every SD-2.1/VAE channel count), else 4.
 
 #include "dcr_common.h"
 
@@ -18,8 +22,36 @@ using namespace dcr;
 
 namespace dcr_nhwc {
 
+template <typename T, int V>
+__device__ __forceinline__ void loadv(const T* __restrict__ p, float (&o)[V]) {
+  if constexpr (V == 8) {
+    f32x8 v = load8<T>(p);
+#pragma unroll
+    for (int k = 0; k < 4; ++k) { o[k] = (&v.lo.x)[k]; o[4 + k] = (&v.hi.x)[k]; }
+  } else {
+    f32x4 v = load4<T>(p);
+#pragma unroll
+    for (int k = 0; k < 4; ++k) o[k] = (&v.x)[k];
+  }
+}
+
+template <typename T, int V>
+__device__ __forceinline__ void storev(T* __restrict__ p, const float (&i)[V]) {
+  if constexpr (V == 8) {
+    f32x8 v;
+#pragma unroll
+    for (int k = 0; k < 4; ++k) { (&v.lo.x)[k] = i[k]; (&v.hi.x)[k] = i[4 + k]; }
+    store8<T>(p, v);
+  } else {
+    f32x4 v;
+#pragma unroll
+    for (int k = 0; k < 4; ++k) (&v.x)[k] = i[k];
+    store4<T>(p, v);
+  }
+}
+
 // --------------------------------------------------------------- stats fwd
-template <typename T>
+template <typename T, int V>
 __global__ void gn_nhwc_stats_kernel(const T* __restrict__ x, float* __restrict__ ws,
                                      int N, int R, int C, int G, int rows_per_blk) {
   extern __shared__ float smem[];          // [2*C]
@@ -31,45 +63,43 @@ __global__ void gn_nhwc_stats_kernel(const T* __restrict__ x, float* __restrict_
   const long base = (long)n * R * C;
   const int Cg = C / G;
 
-  if (C >= (int)blockDim.x * 4) {
-    // wide-C: thread t owns channels [4t, 4t+4) per stripe (full util)
-    for (int c4 = threadIdx.x * 4; c4 < C; c4 += blockDim.x * 4) {
-      f32x4 a = {0.f, 0.f, 0.f, 0.f}, b = {0.f, 0.f, 0.f, 0.f};
-      for (int r = r0; r < r1; ++r) {
-        f32x4 v = load4<T>(x + base + (long)r * C + c4);
+  if (C >= (int)blockDim.x * V) {
+    // wide-C: thread t owns channels [V*t, V*t+V) per stripe (full util)
+    for (int c0 = threadIdx.x * V; c0 < C; c0 += blockDim.x * V) {
+      float a[V], b[V], v[V];
 #pragma unroll
-        for (int k = 0; k < 4; ++k) {
-          (&a.x)[k] += (&v.x)[k];
-          (&b.x)[k] += (&v.x)[k] * (&v.x)[k];
-        }
+      for (int k = 0; k < V; ++k) { a[k] = 0.f; b[k] = 0.f; }
+      for (int r = r0; r < r1; ++r) {
+        loadv<T, V>(x + base + (long)r * C + c0, v);
+#pragma unroll
+        for (int k = 0; k < V; ++k) { a[k] += v[k]; b[k] += v[k] * v[k]; }
       }
 #pragma unroll
-      for (int k = 0; k < 4; ++k) { s1[c4 + k] = (&a.x)[k]; s2[c4 + k] = (&b.x)[k]; }
+      for (int k = 0; k < V; ++k) { s1[c0 + k] = a[k]; s2[c0 + k] = b[k]; }
     }
     __syncthreads();
   } else {
-    // narrow-C (C < 1024): split threads over (channel-vec, row group) so
-    // all 256 lanes stay busy; combine row groups via LDS atomics.
+    // narrow-C: split threads over (channel-vec, row group) so all lanes
+    // stay busy; combine row groups via LDS atomics.
     for (int c = threadIdx.x; c < 2 * C; c += blockDim.x) smem[c] = 0.f;
     __syncthreads();
-    const int tpr = C / 4;
+    const int tpr = C / V;
     const int rpar = (int)blockDim.x / tpr;
-    const int cidx = (threadIdx.x % tpr) * 4;
+    const int cidx = (threadIdx.x % tpr) * V;
     const int rgrp = threadIdx.x / tpr;
     if (rgrp < rpar) {
-      f32x4 a = {0.f, 0.f, 0.f, 0.f}, b = {0.f, 0.f, 0.f, 0.f};
-      for (int r = r0 + rgrp; r < r1; r += rpar) {
-        f32x4 v = load4<T>(x + base + (long)r * C + cidx);
+      float a[V], b[V], v[V];
 #pragma unroll
-        for (int k = 0; k < 4; ++k) {
-          (&a.x)[k] += (&v.x)[k];
-          (&b.x)[k] += (&v.x)[k] * (&v.x)[k];
-        }
+      for (int k = 0; k < V; ++k) { a[k] = 0.f; b[k] = 0.f; }
+      for (int r = r0 + rgrp; r < r1; r += rpar) {
+        loadv<T, V>(x + base + (long)r * C + cidx, v);
+#pragma unroll
+        for (int k = 0; k < V; ++k) { a[k] += v[k]; b[k] += v[k] * v[k]; }
       }
 #pragma unroll
-      for (int k = 0; k < 4; ++k) {
-        atomicAdd(&s1[cidx + k], (&a.x)[k]);
-        atomicAdd(&s2[cidx + k], (&b.x)[k]);
+      for (int k = 0; k < V; ++k) {
+        atomicAdd(&s1[cidx + k], a[k]);
+        atomicAdd(&s2[cidx + k], b[k]);
       }
     }
     __syncthreads();
@@ -94,39 +124,39 @@ __global__ void gn_finalize_kernel(const float* __restrict__ ws,
 }
 
 // --------------------------------------------------------------- apply fwd
-template <typename T, typename WT, bool SILU>
+template <typename T, typename WT, bool SILU, int V>
 __global__ void gn_nhwc_apply_kernel(const T* __restrict__ x, const WT* __restrict__ w,
                                      const WT* __restrict__ b,
                                      const float* __restrict__ mean,
                                      const float* __restrict__ rstd,
                                      T* __restrict__ y, long total, int R, int C, int G) {
   const int Cg = C / G;
-  const long nvec = total / 4;
+  const long nvec = total / V;
   const long RC = (long)R * C;
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
        i += (long)gridDim.x * blockDim.x) {
-    long e = i * 4;
+    long e = i * V;
     long n = e / RC;
-    int c = (int)(e % C);                  // C % 4 == 0: 4 consecutive c
-    f32x4 xv = load4<T>(x + e);
-    f32x4 wv = load4<WT>(w + c);
-    f32x4 bv = load4<WT>(b + c);
-    f32x4 o;
+    int c = (int)(e % C);                  // C % V == 0: V consecutive c
+    float xv[V], wv[V], bv[V], o[V];
+    loadv<T, V>(x + e, xv);
+    loadv<WT, V>(w + c, wv);
+    loadv<WT, V>(b + c, bv);
 #pragma unroll
-    for (int k = 0; k < 4; ++k) {
+    for (int k = 0; k < V; ++k) {
       int g = (c + k) / Cg;
       float m = mean[n * G + g];
       float rs = rstd[n * G + g];
-      float z = (( &xv.x)[k] - m) * rs * (&wv.x)[k] + (&bv.x)[k];
-      (&o.x)[k] = SILU ? silu(z) : z;
+      float z = (xv[k] - m) * rs * wv[k] + bv[k];
+      o[k] = SILU ? silu(z) : z;
     }
-    store4<T>(y + e, o);
+    storev<T, V>(y + e, o);
   }
 }
 
 // --------------------------------------------------------------- stats bwd
 // per-channel dw/db; per-group S1 = sum(w*dz), S2 = sum(w*dz*yhat)
-template <typename T, typename WT, bool SILU>
+template <typename T, typename WT, bool SILU, int V>
 __global__ void gn_nhwc_bwd_stats_kernel(const T* __restrict__ dy, const T* __restrict__ x,
                                          const WT* __restrict__ w, const WT* __restrict__ b_,
                                          const float* __restrict__ mean,
@@ -144,36 +174,37 @@ __global__ void gn_nhwc_bwd_stats_kernel(const T* __restrict__ dy, const T* __re
   const int Cg = C / G;
 
   // (same wide/narrow split as the forward stats kernel)
-  const bool wide = C >= (int)blockDim.x * 4;
-  int tpr = wide ? (int)blockDim.x : C / 4;
+  const bool wide = C >= (int)blockDim.x * V;
+  int tpr = wide ? (int)blockDim.x : C / V;
   int rpar = wide ? 1 : (int)blockDim.x / tpr;
   int rgrp = wide ? 0 : (int)threadIdx.x / tpr;
   if (!wide) {
     for (int c = threadIdx.x; c < 2 * C; c += blockDim.x) smem[c] = 0.f;
     __syncthreads();
   }
-  for (int c4 = (wide ? (int)threadIdx.x * 4 : ((int)threadIdx.x % tpr) * 4);
-       c4 < C; c4 += (wide ? (int)blockDim.x * 4 : C + 1)) {
+  for (int c0 = (wide ? (int)threadIdx.x * V : ((int)threadIdx.x % tpr) * V);
+       c0 < C; c0 += (wide ? (int)blockDim.x * V : C + 1)) {
     if (rgrp >= rpar) break;
-    float m[4], rs[4], wc[4], bc[4];
-    float a[4] = {0, 0, 0, 0}, bb[4] = {0, 0, 0, 0};
-    float dwc[4] = {0, 0, 0, 0}, dbc[4] = {0, 0, 0, 0};
+    float m[V], rs[V], wc[V], bc[V];
+    float a[V], bb[V], dwc[V], dbc[V];
 #pragma unroll
-    for (int k = 0; k < 4; ++k) {
-      const int g = (c4 + k) / Cg;
+    for (int k = 0; k < V; ++k) {
+      const int g = (c0 + k) / Cg;
       m[k] = mean[(long)n * G + g];
       rs[k] = rstd[(long)n * G + g];
-      wc[k] = to_f32<WT>(w[c4 + k]);
-      bc[k] = to_f32<WT>(b_[c4 + k]);
+      wc[k] = to_f32<WT>(w[c0 + k]);
+      bc[k] = to_f32<WT>(b_[c0 + k]);
+      a[k] = bb[k] = dwc[k] = dbc[k] = 0.f;
     }
     for (int r = r0 + rgrp; r < r1; r += rpar) {
-      long idx = base + (long)r * C + c4;
-      f32x4 xv = load4<T>(x + idx);
-      f32x4 gv = load4<T>(dy + idx);
+      long idx = base + (long)r * C + c0;
+      float xv[V], gv[V];
+      loadv<T, V>(x + idx, xv);
+      loadv<T, V>(dy + idx, gv);
 #pragma unroll
-      for (int k = 0; k < 4; ++k) {
-        float yh = ((&xv.x)[k] - m[k]) * rs[k];
-        float dz = (&gv.x)[k];
+      for (int k = 0; k < V; ++k) {
+        float yh = (xv[k] - m[k]) * rs[k];
+        float dz = gv[k];
         if (SILU) dz *= dsilu(yh * wc[k] + bc[k]);
         float gx = dz * wc[k];
         a[k] += gx;
@@ -183,16 +214,16 @@ __global__ void gn_nhwc_bwd_stats_kernel(const T* __restrict__ dy, const T* __re
       }
     }
 #pragma unroll
-    for (int k = 0; k < 4; ++k) {
+    for (int k = 0; k < V; ++k) {
       if (wide) {
-        sa[c4 + k] = a[k];
-        sb[c4 + k] = bb[k];
+        sa[c0 + k] = a[k];
+        sb[c0 + k] = bb[k];
       } else {
-        atomicAdd(&sa[c4 + k], a[k]);
-        atomicAdd(&sb[c4 + k], bb[k]);
+        atomicAdd(&sa[c0 + k], a[k]);
+        atomicAdd(&sb[c0 + k], bb[k]);
       }
-      atomicAdd(&dw[c4 + k], dwc[k]);
-      atomicAdd(&db[c4 + k], dbc[k]);
+      atomicAdd(&dw[c0 + k], dwc[k]);
+      atomicAdd(&db[c0 + k], dbc[k]);
     }
   }
   __syncthreads();
@@ -205,7 +236,7 @@ __global__ void gn_nhwc_bwd_stats_kernel(const T* __restrict__ dy, const T* __re
 }
 
 // --------------------------------------------------------------- apply bwd
-template <typename T, typename WT, bool SILU>
+template <typename T, typename WT, bool SILU, int V>
 __global__ void gn_nhwc_bwd_apply_kernel(const T* __restrict__ dy, const T* __restrict__ x,
                                          const WT* __restrict__ w, const WT* __restrict__ b_,
                                          const float* __restrict__ mean,
@@ -214,32 +245,32 @@ __global__ void gn_nhwc_bwd_apply_kernel(const T* __restrict__ dy, const T* __re
                                          T* __restrict__ dx, long total, int R, int C,
                                          int G, float inv_L) {
   const int Cg = C / G;
-  const long nvec = total / 4;
+  const long nvec = total / V;
   const long RC = (long)R * C;
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
        i += (long)gridDim.x * blockDim.x) {
-    long e = i * 4;
+    long e = i * V;
     long n = e / RC;
     int c = (int)(e % C);
-    f32x4 xv = load4<T>(x + e);
-    f32x4 gv = load4<T>(dy + e);
-    f32x4 wv = load4<WT>(w + c);
-    f32x4 bv = load4<WT>(b_ + c);
-    f32x4 o;
+    float xv[V], gv[V], wv[V], bv[V], o[V];
+    loadv<T, V>(x + e, xv);
+    loadv<T, V>(dy + e, gv);
+    loadv<WT, V>(w + c, wv);
+    loadv<WT, V>(b_ + c, bv);
 #pragma unroll
-    for (int k = 0; k < 4; ++k) {
+    for (int k = 0; k < V; ++k) {
       int g = (c + k) / Cg;
       float m = mean[n * G + g];
       float rs = rstd[n * G + g];
       float m1 = ws[(n * G + g) * 2] * inv_L;
       float m2 = ws[(n * G + g) * 2 + 1] * inv_L;
-      float yh = ((&xv.x)[k] - m) * rs;
-      float dz = (&gv.x)[k];
-      if (SILU) dz *= dsilu(yh * (&wv.x)[k] + (&bv.x)[k]);
-      float gx = dz * (&wv.x)[k];
-      (&o.x)[k] = rs * (gx - m1 - yh * m2);
+      float yh = (xv[k] - m) * rs;
+      float dz = gv[k];
+      if (SILU) dz *= dsilu(yh * wv[k] + bv[k]);
+      float gx = dz * wv[k];
+      o[k] = rs * (gx - m1 - yh * m2);
     }
-    store4<T>(dx + e, o);
+    storev<T, V>(dx + e, o);
   }
 }
 
@@ -260,28 +291,39 @@ static inline int nhwc_row_chunks(int N, int R) {
   return chunks;
 }
 
-template <typename T, typename WT>
-static void gn_nhwc_fwd_t(const void* x, const void* w, const void* b, void* y,
+template <typename T, typename WT, int V>
+static void gn_nhwc_fwd_v(const void* x, const void* w, const void* b, void* y,
                           float* ws, float* mean, float* rstd, int N, int R,
                           int C, int G, float eps, bool silu, hipStream_t s) {
   int chunks = nhwc_row_chunks(N, R);
   int rows_per_blk = (R + chunks - 1) / chunks;
   dim3 grid(chunks, N), block(256);
   size_t lds = 2 * (size_t)C * sizeof(float);
-  hipLaunchKernelGGL((dcr_nhwc::gn_nhwc_stats_kernel<T>), grid, block, lds, s,
+  hipLaunchKernelGGL((dcr_nhwc::gn_nhwc_stats_kernel<T, V>), grid, block, lds, s,
                      (const T*)x, ws, N, R, C, G, rows_per_blk);
   long NG = (long)N * G;
   float inv_L = 1.f / ((float)R * (C / G));
   hipLaunchKernelGGL(dcr_nhwc::gn_finalize_kernel, dim3((NG + 255) / 256),
                      dim3(256), 0, s, ws, mean, rstd, NG, inv_L, eps);
   long total = (long)N * R * C;
-  dim3 agrid((int)min((total / 4 + 255) / 256, (long)8192)), ablock(256);
+  dim3 agrid((int)min((total / V + 255) / 256, (long)8192)), ablock(256);
   if (silu)
-    hipLaunchKernelGGL((dcr_nhwc::gn_nhwc_apply_kernel<T, WT, true>), agrid, ablock, 0, s,
+    hipLaunchKernelGGL((dcr_nhwc::gn_nhwc_apply_kernel<T, WT, true, V>), agrid, ablock, 0, s,
                        (const T*)x, (const WT*)w, (const WT*)b, mean, rstd, (T*)y, total, R, C, G);
   else
-    hipLaunchKernelGGL((dcr_nhwc::gn_nhwc_apply_kernel<T, WT, false>), agrid, ablock, 0, s,
+    hipLaunchKernelGGL((dcr_nhwc::gn_nhwc_apply_kernel<T, WT, false, V>), agrid, ablock, 0, s,
                        (const T*)x, (const WT*)w, (const WT*)b, mean, rstd, (T*)y, total, R, C, G);
+}
+
+template <typename T, typename WT>
+static void gn_nhwc_fwd_t(const void* x, const void* w, const void* b, void* y,
+                          float* ws, float* mean, float* rstd, int N, int R,
+                          int C, int G, float eps, bool silu, hipStream_t s) {
+  // 16 B/lane when the channel count admits it (f32 is already 16 B at V=4)
+  if (C % 8 == 0 && sizeof(T) == 2)
+    gn_nhwc_fwd_v<T, WT, 8>(x, w, b, y, ws, mean, rstd, N, R, C, G, eps, silu, s);
+  else
+    gn_nhwc_fwd_v<T, WT, 4>(x, w, b, y, ws, mean, rstd, N, R, C, G, eps, silu, s);
 }
 
 void gn_nhwc_fwd_launch(DType dt, const void* x, const void* w, const void* b,
@@ -301,8 +343,8 @@ void gn_nhwc_fwd_launch(DType dt, const void* x, const void* w, const void* b,
   }
 }
 
-template <typename T, typename WT>
-static void gn_nhwc_bwd_t(const void* dy, const void* x, const void* w,
+template <typename T, typename WT, int V>
+static void gn_nhwc_bwd_v(const void* dy, const void* x, const void* w,
                           const void* b, const float* mean, const float* rstd,
                           float* ws, void* dx, float* dw, float* db, int N,
                           int R, int C, int G, bool silu, hipStream_t s) {
@@ -311,24 +353,35 @@ static void gn_nhwc_bwd_t(const void* dy, const void* x, const void* w,
   dim3 grid(chunks, N), block(256);
   size_t lds = 2 * (size_t)C * sizeof(float);
   if (silu)
-    hipLaunchKernelGGL((dcr_nhwc::gn_nhwc_bwd_stats_kernel<T, WT, true>), grid, block, lds, s,
+    hipLaunchKernelGGL((dcr_nhwc::gn_nhwc_bwd_stats_kernel<T, WT, true, V>), grid, block, lds, s,
                        (const T*)dy, (const T*)x, (const WT*)w, (const WT*)b, mean, rstd, ws, dw, db,
                        N, R, C, G, rows_per_blk);
   else
-    hipLaunchKernelGGL((dcr_nhwc::gn_nhwc_bwd_stats_kernel<T, WT, false>), grid, block, lds, s,
+    hipLaunchKernelGGL((dcr_nhwc::gn_nhwc_bwd_stats_kernel<T, WT, false, V>), grid, block, lds, s,
                        (const T*)dy, (const T*)x, (const WT*)w, (const WT*)b, mean, rstd, ws, dw, db,
                        N, R, C, G, rows_per_blk);
   long total = (long)N * R * C;
   float inv_L = 1.f / ((float)R * (C / G));
-  dim3 agrid((int)min((total / 4 + 255) / 256, (long)8192)), ablock(256);
+  dim3 agrid((int)min((total / V + 255) / 256, (long)8192)), ablock(256);
   if (silu)
-    hipLaunchKernelGGL((dcr_nhwc::gn_nhwc_bwd_apply_kernel<T, WT, true>), agrid, ablock, 0, s,
+    hipLaunchKernelGGL((dcr_nhwc::gn_nhwc_bwd_apply_kernel<T, WT, true, V>), agrid, ablock, 0, s,
                        (const T*)dy, (const T*)x, (const WT*)w, (const WT*)b, mean, rstd, ws, (T*)dx,
                        total, R, C, G, inv_L);
   else
-    hipLaunchKernelGGL((dcr_nhwc::gn_nhwc_bwd_apply_kernel<T, WT, false>), agrid, ablock, 0, s,
+    hipLaunchKernelGGL((dcr_nhwc::gn_nhwc_bwd_apply_kernel<T, WT, false, V>), agrid, ablock, 0, s,
                        (const T*)dy, (const T*)x, (const WT*)w, (const WT*)b, mean, rstd, ws, (T*)dx,
                        total, R, C, G, inv_L);
+}
+
+template <typename T, typename WT>
+static void gn_nhwc_bwd_t(const void* dy, const void* x, const void* w,
+                          const void* b, const float* mean, const float* rstd,
+                          float* ws, void* dx, float* dw, float* db, int N,
+                          int R, int C, int G, bool silu, hipStream_t s) {
+  if (C % 8 == 0 && sizeof(T) == 2)
+    gn_nhwc_bwd_v<T, WT, 8>(dy, x, w, b, mean, rstd, ws, dx, dw, db, N, R, C, G, silu, s);
+  else
+    gn_nhwc_bwd_v<T, WT, 4>(dy, x, w, b, mean, rstd, ws, dx, dw, db, N, R, C, G, silu, s);
 }
 
 void gn_nhwc_bwd_launch(DType dt, const void* dy, const void* x, const void* w,

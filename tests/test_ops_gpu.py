@@ -600,3 +600,52 @@ def test_dcr_linear_autograd_matches_f_linear(ext):
     _close(x.grad, xf.grad, 2e-2)
     _close(w.grad, wf.grad, 2e-2)
     _close(b.grad, bf.grad, 2e-2)
+
+
+# ---------------------------------------------------------------------------
+# Generalized head-dim attention forward (SD-1.4 40/80/160, round 2)
+# ---------------------------------------------------------------------------
+@pytest.mark.parametrize("D", [40, 80, 160])
+@pytest.mark.parametrize("Lq,Lk", [(64, 64), (256, 77), (1024, 1024), (100, 33)])
+def test_attn_fwd_gen(ext, D, Lq, Lk):
+    torch.manual_seed(D + Lq)
+    B, H = 2, 3
+    q = torch.randn(B, Lq, H, D, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(B, Lk, H, D, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(B, Lk, H, D, device="cuda", dtype=torch.bfloat16)
+    scale = 1.0 / D ** 0.5
+    o = ext.attn_fwd_gen(q, k, v, scale, False)
+    ref = F.scaled_dot_product_attention(
+        q.permute(0, 2, 1, 3).float(), k.permute(0, 2, 1, 3).float(),
+        v.permute(0, 2, 1, 3).float(), scale=scale).permute(0, 2, 1, 3)
+    _close(o, ref, 2e-2)
+
+
+def test_attn_fwd_gen_causal(ext):
+    torch.manual_seed(5)
+    B, H, L, D = 2, 4, 128, 80
+    q = torch.randn(B, L, H, D, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(B, L, H, D, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(B, L, H, D, device="cuda", dtype=torch.bfloat16)
+    scale = 1.0 / D ** 0.5
+    o = ext.attn_fwd_gen(q, k, v, scale, True)
+    ref = F.scaled_dot_product_attention(
+        q.permute(0, 2, 1, 3).float(), k.permute(0, 2, 1, 3).float(),
+        v.permute(0, 2, 1, 3).float(), is_causal=True,
+        scale=scale).permute(0, 2, 1, 3)
+    _close(o, ref, 2e-2)
+
+
+def test_attention_dispatch_sd14_head_dims(ext):
+    """ops.attention routes SD-1.4 inference shapes through the gen
+    kernel, not the composite rocBLAS+softmax fallback."""
+    from dcr_amd import ops
+    from dcr_amd.ops import dispatch_counts
+    before = dispatch_counts["attention_gen"]
+    with torch.no_grad():
+        q = torch.randn(2, 256, 8, 40, device="cuda", dtype=torch.bfloat16)
+        k = torch.randn(2, 77, 8, 40, device="cuda", dtype=torch.bfloat16)
+        v = torch.randn(2, 77, 8, 40, device="cuda", dtype=torch.bfloat16)
+        out = ops.attention(q, k, v, layout="blhd")
+    assert out.shape == q.shape
+    assert dispatch_counts["attention_gen"] == before + 1
