@@ -192,9 +192,10 @@ class _FlashAttention(torch.autograd.Function):
 
 
 def _attention_math(q, k, v, scale, causal):
-    """Composite path (rocBLAS GEMMs + native softmax) for shapes the
-    flash kernel does not cover (head_dim != 64, e.g. SD-1.4's 40/80/160
-    and the VAE's single 512-d head). No Triton involved."""
+    """Composite path (rocBLAS GEMMs + native softmax) for shapes no HIP
+    kernel covers (now only the VAE's single 512-d head and training-mode
+    non-64 dims; SD-1.4's 40/80/160 run the gen kernel). No Triton."""
+    count_dispatch('attention_math')
     s = (q.float() @ k.float().transpose(-2, -1)) * scale
     if causal:
         Lq, Lk = q.shape[-2], k.shape[-2]

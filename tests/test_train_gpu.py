@@ -55,20 +55,28 @@ def test_sd21_train_step_gpu(tmp_path):
 
 
 def test_sd14_forward_gpu():
-    """SD-1.4 family (sd_mitigation's model): head_dim 40/80/160 takes the
-    composite attention path; norms/GEGLU still run the HIP kernels."""
+    """SD-1.4 family (sd_mitigation's model): head_dim 40/80/160 runs the
+    generalized HIP attention kernel (round 2) — no composite
+    rocBLAS+softmax fallback on the sampling path."""
     from dcr_amd.models import (AutoencoderKL, CLIPTextModel, CLIPTextConfig,
                                 UNet2DConditionModel, UNetConfig, VAEConfig)
+    from dcr_amd.ops import dispatch_counts
     torch.manual_seed(0)
     unet = UNet2DConditionModel(UNetConfig.sd14()).cuda().to(torch.bfloat16)
     te = CLIPTextModel(CLIPTextConfig.sd14()).cuda().to(torch.bfloat16)
     ids = torch.randint(0, 49408, (2, 77), device="cuda")
+    gen0 = dispatch_counts["attention_gen"]
+    math0 = dispatch_counts["attention_math"]
     with torch.no_grad():
         emb = te(ids)[0]
         out = unet(torch.randn(2, 4, 32, 32, device="cuda").bfloat16(),
                    torch.tensor([10, 500], device="cuda"), emb)
     assert out.shape == (2, 4, 32, 32)
     assert torch.isfinite(out.float()).all()
+    assert dispatch_counts["attention_gen"] > gen0, \
+        "SD-1.4 attention must dispatch the gen HIP kernel"
+    assert dispatch_counts["attention_math"] == math0, \
+        "no composite fallback on the SD-1.4 forward"
 
 
 def test_unet_fwd_hip_matches_cpu_reference():
