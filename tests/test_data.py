@@ -158,8 +158,8 @@ def test_bool_flag():
 
 
 def test_load_real_clip_tokenizer(tmp_path):
-    """load_tokenizer uses transformers CLIPTokenizer when vocab files
-    exist (reference tokenizer path, diff_train.py:371-383)."""
+    """load_tokenizer uses the in-house CLIP BPE when vocab files exist
+    (reference tokenizer path, diff_train.py:371-383)."""
     vocab = {"<|startoftext|>": 0, "<|endoftext|>": 1}
     # minimal BPE vocab: byte-level symbols + a merged token
     for i, ch in enumerate("abcdefghijklmnopqrstuvwxyz"):
@@ -168,14 +168,14 @@ def test_load_real_clip_tokenizer(tmp_path):
     vocab["ab</w>"] = 60
     (tmp_path / "vocab.json").write_text(json.dumps(vocab))
     (tmp_path / "merges.txt").write_text("#version: 0.2\na b</w>\n")
-    from dcr_amd.data.tokenizer import load_tokenizer
+    from dcr_amd.data.tokenizer import CLIPBPETokenizer, load_tokenizer
     tok = load_tokenizer(tmp_path, model_max_length=77)
-    from transformers import CLIPTokenizer
-    assert isinstance(tok, CLIPTokenizer)
+    assert isinstance(tok, CLIPBPETokenizer)  # in-house BPE since round 2
     out = tok("ab", truncation=True, padding="max_length", max_length=16,
               return_tensors="pt")
     assert out.input_ids.shape == (1, 16)
     assert out.input_ids[0, 0].item() == 0  # bos
+    assert out.input_ids[0, 1].item() == vocab["ab</w>"]
 
 
 def test_transforms_shapes_and_range():
@@ -221,3 +221,61 @@ def test_dataset_with_real_clip_tokenizer(tmp_path, image_folder):
                                 size=64)
     ex = ds[0]
     assert ex["instance_prompt_ids"].shape == (1, 77)
+
+
+def test_clip_bpe_matches_transformers(tmp_path):
+    """In-house CLIP BPE (dcr_amd.data.tokenizer.CLIPBPETokenizer) produces
+    token-for-token the ids transformers' CLIPTokenizer produces on the
+    same vocab/merges files (constructed here — no network)."""
+    import json
+    from dcr_amd.data.tokenizer import CLIPBPETokenizer, _bytes_to_unicode
+
+    b2u = _bytes_to_unicode()
+    chars = list(b2u.values())
+    vocab = {}
+    for c in chars:
+        vocab[c] = len(vocab)
+    for c in chars:
+        vocab[c + "</w>"] = len(vocab)
+    merges = [("t", "h"), ("th", "e</w>"), ("a", "n"), ("an", "d</w>"),
+              ("i", "n"), ("in", "g</w>"), ("r", "e"), ("o", "n</w>"),
+              ("c", "a"), ("ca", "t</w>"), ("th", "e"), ("the", "re</w>")]
+    for a, b in merges:
+        for tok in (a, b, a + b):
+            if tok not in vocab:
+                vocab[tok] = len(vocab)
+    vocab["<|startoftext|>"] = len(vocab)
+    vocab["<|endoftext|>"] = len(vocab)
+    (tmp_path / "vocab.json").write_text(json.dumps(vocab))
+    (tmp_path / "merges.txt").write_text(
+        "#version: 0.2\n" + "\n".join(f"{a} {b}" for a, b in merges) + "\n")
+
+    ours = CLIPBPETokenizer(tmp_path / "vocab.json", tmp_path / "merges.txt")
+    from transformers import CLIPTokenizer
+    theirs = CLIPTokenizer(str(tmp_path / "vocab.json"),
+                           str(tmp_path / "merges.txt"))
+
+    samples = [
+        "the cat sat on the mat",
+        "A photograph of THE re-opening, and nothing else!",
+        "there and then: 123 cats running, in-the rain...",
+        "captain's log 42; the ending?",
+        "",
+    ]
+    for s in samples:
+        ref = theirs(s)["input_ids"]
+        got = ours.encode_words(s)
+        assert [vocab["<|startoftext|>"]] + got + [vocab["<|endoftext|>"]] \
+            == ref, (s, got, ref)
+
+    # padded batch surface (what the training pipeline calls)
+    out = ours(samples[:2], max_length=16).input_ids
+    assert out.shape == (2, 16)
+    assert (out[:, 0] == vocab["<|startoftext|>"]).all()
+
+    # save/load round-trip keeps ids
+    ours.save_pretrained(tmp_path / "rt")
+    from dcr_amd.data.tokenizer import load_tokenizer
+    re_tok = load_tokenizer(tmp_path / "rt")
+    assert type(re_tok).__name__ == "CLIPBPETokenizer"
+    assert re_tok.encode_words(samples[0]) == ours.encode_words(samples[0])

@@ -186,3 +186,55 @@ def test_clip_quick_gelu_differs_from_gelu():
     m.act = "gelu"
     yg = m(x)
     assert not torch.allclose(yq, yg)
+
+
+def test_sd21_state_dict_matches_golden_manifest():
+    """Every SD-2.1 module's state-dict keys AND shapes match the
+    committed diffusers-naming manifest (tests/golden/), pinning the
+    checkpoint-interop surface (diffusers layout contract).
+    The manifest's UNet totals 865,910,724 params = diffusers'
+    stabilityai/stable-diffusion-2-1 UNet."""
+    import json
+    from pathlib import Path
+    from dcr_amd.models import (AutoencoderKL, CLIPTextConfig, CLIPTextModel,
+                                UNet2DConditionModel, UNetConfig, VAEConfig)
+    man = json.loads((Path(__file__).parent / "golden" /
+                      "sd21_state_dict_manifest.json").read_text())
+    mods = {
+        "unet": UNet2DConditionModel(UNetConfig.sd21()),
+        "vae": AutoencoderKL(VAEConfig.sd()),
+        "text_encoder": CLIPTextModel(CLIPTextConfig.sd21()),
+    }
+    for name, mod in mods.items():
+        got = {k: list(v.shape) for k, v in mod.state_dict().items()}
+        assert got == man[name], (
+            name,
+            sorted(set(got) ^ set(man[name]))[:10] or "shape mismatch")
+    n = sum(p.numel() for p in mods["unet"].parameters())
+    assert n == 865_910_724, n
+
+
+def test_tiny_forward_matches_golden():
+    """Fixed-seed tiny UNet + VAE forward reproduces the committed golden
+    tensors bit-for-bit on CPU — guards model numerics across rounds
+    (SURVEY §4.2 module-parity strategy under the no-network constraint)."""
+    from pathlib import Path
+    from dcr_amd.models import (AutoencoderKL, UNet2DConditionModel,
+                                UNetConfig, VAEConfig)
+    g = torch.load(Path(__file__).parent / "golden" / "tiny_forward_golden.pt",
+                   weights_only=True)
+    torch.manual_seed(1234)
+    un = UNet2DConditionModel(UNetConfig.tiny()).eval()
+    with torch.no_grad():
+        out = un(g["x"], g["t"], g["ehs"])
+    assert torch.allclose(out, g["unet_out"], rtol=1e-5, atol=1e-6), \
+        (out - g["unet_out"]).abs().max()
+
+    torch.manual_seed(1234)
+    vae = AutoencoderKL(VAEConfig.tiny()).eval()
+    with torch.no_grad():
+        lat = vae.encode(g["vae_in"]).latent_dist.mean
+        dec = vae.decode(lat)
+    dec = getattr(dec, "sample", dec)
+    assert torch.allclose(lat, g["vae_lat"], rtol=1e-5, atol=1e-6)
+    assert torch.allclose(dec, g["vae_dec"], rtol=1e-5, atol=1e-6)
