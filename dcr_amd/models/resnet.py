@@ -3,7 +3,8 @@
 Reference behavior: diffusers ResnetBlock2D / Downsample2D / Upsample2D as
 exercised by the finetune loop (/root/reference/diff_train.py:644) —
 rebuilt on dcr_amd ops: GroupNorm+SiLU is one fused HIP kernel
-(SURVEY.md §2.4.A), convs go through MIOpen via torch.
+(SURVEY.md §2.4.A), convs go through the native implicit-GEMM kernel (ops/conv.py)
+with the time-embedding and residual adds fused into its epilogue.
 """
 from __future__ import annotations
 
@@ -50,15 +51,20 @@ class ResnetBlock2D(nn.Module):
 
     def forward(self, x: torch.Tensor, temb: Optional[torch.Tensor] = None) -> torch.Tensor:
         h = self.norm1(x)          # fused GN+SiLU
-        h = self.conv1(h)
+        # time-embedding projection rides conv1's epilogue (one fused
+        # add instead of a broadcast elementwise pass); the residual add
+        # rides conv2's epilogue the same way (ops/conv.py)
+        tv = None
         if self.time_emb_proj is not None and temb is not None:
-            h = h + self.time_emb_proj(F.silu(temb))[:, :, None, None]
+            tv = self.time_emb_proj(F.silu(temb))
+        h = self.conv1(h, temb=tv)
         h = self.norm2(h)          # fused GN+SiLU
         h = self.dropout(h)
-        h = self.conv2(h)
-        if self.conv_shortcut is not None:
-            x = self.conv_shortcut(x)
-        return (x + h) / self.output_scale_factor
+        sc = self.conv_shortcut(x) if self.conv_shortcut is not None else x
+        h = self.conv2(h, res=sc)
+        if self.output_scale_factor != 1.0:
+            h = h / self.output_scale_factor
+        return h
 
 
 class Downsample2D(nn.Module):

@@ -53,20 +53,30 @@ def _eligible(x: torch.Tensor, m: nn.Conv2d) -> bool:
 
 
 class _NativeConvFn(torch.autograd.Function):
+    """Native conv fwd with optional fused epilogue adds:
+    res — ResnetBlock2D's residual (d_res = dy, no extra kernel);
+    temb — the [N, K] time-embedding projection broadcast
+    (d_temb = dy.sum over H, W — what autograd's broadcast-add backward
+    did anyway). Each fusion deletes one full elementwise pass."""
+
     @staticmethod
-    def forward(ctx, x, weight, bias, stride, padding):
+    def forward(ctx, x, weight, bias, stride, padding, res, temb):
         m = ext()
         count_dispatch('conv_nhwc')
-        y = m.conv2d_nhwc_fwd_v2(x, weight, bias, stride, padding)
+        y = m.conv2d_nhwc_fwd_v2(x, weight, bias, stride, padding,
+                                 res, temb)
         ctx.save_for_backward(x, weight)
-        ctx.conf = (stride, padding, bias is not None)
+        ctx.conf = (stride, padding, bias is not None,
+                    res is not None, temb is not None)
         return y
 
     @staticmethod
     def backward(ctx, dy):
         x, weight = ctx.saved_tensors
-        stride, padding, has_bias = ctx.conf
+        stride, padding, has_bias, has_res, has_temb = ctx.conf
         dy = dy.contiguous(memory_format=torch.channels_last)
+        d_res = dy if has_res else None
+        d_temb = dy.sum(dim=(2, 3)) if has_temb else None
         # native backward (conv_nhwc_bwd.hip: dgrad + wgrad + fused
         # bias-grad) — numerics GPU-validated round 2; C % 64 shapes only
         if os.environ.get("DCR_NATIVE_CONV_BWD", "0") == "1" \
@@ -78,22 +88,44 @@ class _NativeConvFn(torch.autograd.Function):
                 db = db.to(dy.dtype)
             else:
                 db = None
-            return dx, dw, db, None, None
+            return dx, dw, db, None, None, d_res, d_temb
         dx, dw, db = torch.ops.aten.convolution_backward(
             dy, x, weight,
             [weight.shape[0]] if has_bias else None,
             [stride, stride], [padding, padding], [1, 1], False, [0, 0], 1,
             [ctx.needs_input_grad[0], ctx.needs_input_grad[1],
              has_bias and ctx.needs_input_grad[2]])
-        return dx, dw, db, None, None
+        return dx, dw, db, None, None, d_res, d_temb
 
 
 class Conv2d(nn.Conv2d):
     """nn.Conv2d with the native MI355X forward when eligible (identical
-    parameters/state-dict; backward via MIOpen either way)."""
+    parameters/state-dict; backward via MIOpen either way). `res`/`temb`
+    optionally fuse a residual / per-(n,k) time-embedding add into the
+    native epilogue; ineligible shapes fall back to the unfused ops."""
 
-    def forward(self, x):
+    def forward(self, x, res=None, temb=None):
         if _enabled() and _eligible(x, self):
-            return _NativeConvFn.apply(x, self.weight, self.bias,
-                                       self.stride[0], self.padding[0])
-        return super().forward(x)
+            fuse_res = res is None or (
+                res.dtype == torch.bfloat16
+                and res.is_contiguous(memory_format=torch.channels_last))
+            fuse_temb = temb is None or (temb.dtype == torch.bfloat16
+                                         and temb.is_contiguous())
+            if fuse_res and fuse_temb:
+                return _NativeConvFn.apply(x, self.weight, self.bias,
+                                           self.stride[0], self.padding[0],
+                                           res, temb)
+            y = _NativeConvFn.apply(x, self.weight, self.bias,
+                                    self.stride[0], self.padding[0],
+                                    None, None)
+            if temb is not None:
+                y = y + temb[:, :, None, None]
+            if res is not None:
+                y = y + res
+            return y
+        y = super().forward(x)
+        if temb is not None:
+            y = y + temb[:, :, None, None]
+        if res is not None:
+            y = y + res
+        return y

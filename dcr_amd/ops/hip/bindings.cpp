@@ -448,7 +448,9 @@ at::Tensor conv2d_nhwc_fwd(at::Tensor x, at::Tensor w,
 
 static at::Tensor conv2d_nhwc_fwd_v2_impl(at::Tensor x, at::Tensor w,
                                           c10::optional<at::Tensor> bias,
-                                          int64_t stride, int64_t pad) {
+                                          int64_t stride, int64_t pad,
+                                          c10::optional<at::Tensor> res,
+                                          c10::optional<at::Tensor> temb) {
   TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16);
   TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast) &&
               w.is_contiguous(at::MemoryFormat::ChannelsLast));
@@ -481,8 +483,23 @@ static at::Tensor conv2d_nhwc_fwd_v2_impl(at::Tensor x, at::Tensor w,
     ws = at::zeros({NPQ * K}, x.options().dtype(at::kFloat));
     wsp = ws.data_ptr<float>();
   }
+  const void* rp = nullptr;
+  const void* tp = nullptr;
+  if (res.has_value()) {
+    TORCH_CHECK(res->scalar_type() == at::kBFloat16 &&
+                res->is_contiguous(at::MemoryFormat::ChannelsLast) &&
+                res->sizes() == y.sizes(), "conv res: NHWC bf16 same shape");
+    rp = res->data_ptr();
+  }
+  if (temb.has_value()) {
+    TORCH_CHECK(temb->scalar_type() == at::kBFloat16 &&
+                temb->is_contiguous() && temb->dim() == 2 &&
+                temb->size(0) == Nb && temb->size(1) == K,
+                "conv temb: contiguous bf16 [N, K]");
+    tp = temb->data_ptr();
+  }
   conv_nhwc_fwd_v2_launch(x.data_ptr(), w.data_ptr(), bp, y.data_ptr(), wsp,
-         splitz, (int)Nb, (int)Hin, (int)Win, (int)C, (int)K,
+         splitz, rp, tp, (int)Nb, (int)Hin, (int)Win, (int)C, (int)K,
          (int)P, (int)Q, (int)R, (int)S, (int)stride,
          (int)pad, cur_stream());
   return y;
@@ -490,8 +507,10 @@ static at::Tensor conv2d_nhwc_fwd_v2_impl(at::Tensor x, at::Tensor w,
 
 at::Tensor conv2d_nhwc_fwd_v2(at::Tensor x, at::Tensor w,
                               c10::optional<at::Tensor> bias, int64_t stride,
-                              int64_t pad) {
-  return conv2d_nhwc_fwd_v2_impl(x, w, bias, stride, pad);
+                              int64_t pad,
+                              c10::optional<at::Tensor> res = c10::nullopt,
+                              c10::optional<at::Tensor> temb = c10::nullopt) {
+  return conv2d_nhwc_fwd_v2_impl(x, w, bias, stride, pad, res, temb);
 }
 
 // conv backward drafts (round-2; validated before any dispatch)
@@ -579,7 +598,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("gemm_bf16", &gemm_bf16);
   mod.def("conv2d_nhwc_bwd", &conv2d_nhwc_bwd);
   mod.def("conv2d_nhwc_fwd", &conv2d_nhwc_fwd);
-  mod.def("conv2d_nhwc_fwd_v2", &conv2d_nhwc_fwd_v2);
+  mod.def("conv2d_nhwc_fwd_v2", &conv2d_nhwc_fwd_v2,
+          pybind11::arg("x"), pybind11::arg("w"), pybind11::arg("bias"),
+          pybind11::arg("stride"), pybind11::arg("pad"),
+          pybind11::arg("res") = c10::nullopt,
+          pybind11::arg("temb") = c10::nullopt);
   mod.def("attn_fwd", &attn_fwd);
   mod.def("attn_fwd_gen", &attn_fwd_gen);
   mod.def("attn_fwd_v2", &attn_fwd_v2);

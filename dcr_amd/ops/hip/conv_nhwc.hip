@@ -143,11 +143,17 @@ namespace dcr_conv {
 // splitz > 1: blockIdx.z covers a slice of the rsc steps; fp32 partials
 // are atomically accumulated into ws[NPQ*K] and a finalize kernel adds
 // bias + casts (grid starvation fix for the 8x8/16x16 shapes).
+// res: optional residual [NPQ, K] (NHWC tensor) added in the epilogue
+// (ResnetBlock2D's `x + h`); temb: optional per-(n,k) bf16 bias [N, K]
+// (the time-embedding projection broadcast) — both fuse away a full
+// elementwise pass over the output tensor.
 template <int BK>
 __global__ __launch_bounds__(256)
 void conv_nhwc_fwd_v2_kernel(const bf16_t* __restrict__ x, const bf16_t* __restrict__ w,
                              const float* __restrict__ bias, bf16_t* __restrict__ y,
                              float* __restrict__ ws, int splitz,
+                             const bf16_t* __restrict__ res,
+                             const bf16_t* __restrict__ temb,
                              int Nb, int Hin, int Win, int C, int K, int P, int Q,
                              int R, int S, int stride, int pad) {
   constexpr int PITCH2 = BK + 8;           // 16-lane b128 groups: 16 banks
@@ -248,6 +254,7 @@ void conv_nhwc_fwd_v2_kernel(const bf16_t* __restrict__ x, const bf16_t* __restr
     }
   }
 
+  const int PQ = P * Q;
 #pragma unroll
   for (int i = 0; i < 4; ++i) {
 #pragma unroll
@@ -262,6 +269,8 @@ void conv_nhwc_fwd_v2_kernel(const bf16_t* __restrict__ x, const bf16_t* __restr
           atomicAdd(&ws[m * K + k], acc[i][j][rr]);
         } else {
           float v = acc[i][j][rr] + (bias ? bias[k] : 0.f);
+          if (res) v += __bfloat162float(res[m * K + k]);
+          if (temb) v += __bfloat162float(temb[(m / PQ) * K + k]);
           y[m * K + k] = __float2bfloat16(v);
         }
       }
@@ -271,11 +280,15 @@ void conv_nhwc_fwd_v2_kernel(const bf16_t* __restrict__ x, const bf16_t* __restr
 
 __global__ void conv_splitk_finalize_kernel(const float* __restrict__ ws,
                                             const float* __restrict__ bias,
+                                            const bf16_t* __restrict__ res,
+                                            const bf16_t* __restrict__ temb,
                                             bf16_t* __restrict__ y, long total,
-                                            int K) {
+                                            int K, int PQ) {
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += (long)gridDim.x * blockDim.x) {
     float v = ws[i] + (bias ? bias[i % K] : 0.f);
+    if (res) v += __bfloat162float(res[i]);
+    if (temb) v += __bfloat162float(temb[(i / ((long)K * PQ)) * K + i % K]);
     y[i] = __float2bfloat16(v);
   }
 }
@@ -285,30 +298,35 @@ __global__ void conv_splitk_finalize_kernel(const float* __restrict__ ws,
 namespace dcr {
 
 void conv_nhwc_fwd_v2_launch(const void* x, const void* w, const float* bias,
-                             void* y, float* ws, int splitz, int Nb, int Hin,
-                             int Win, int C, int K, int P, int Q, int R, int S,
-                             int stride, int pad, hipStream_t st) {
+                             void* y, float* ws, int splitz, const void* res,
+                             const void* temb, int Nb, int Hin, int Win, int C,
+                             int K, int P, int Q, int R, int S, int stride,
+                             int pad, hipStream_t st) {
   long NPQ = (long)Nb * P * Q;
   dim3 grid((unsigned)((NPQ + 127) / 128), (unsigned)((K + 127) / 128),
             (unsigned)splitz),
       block(256);
+  const auto* rp = (const dcr_conv::bf16_t*)res;
+  const auto* tp = (const dcr_conv::bf16_t*)temb;
   if (C % 64 == 0)
     hipLaunchKernelGGL((dcr_conv::conv_nhwc_fwd_v2_kernel<64>), grid, block, 0,
                        st, (const dcr_conv::bf16_t*)x,
                        (const dcr_conv::bf16_t*)w, bias, (dcr_conv::bf16_t*)y,
-                       ws, splitz, Nb, Hin, Win, C, K, P, Q, R, S, stride, pad);
+                       ws, splitz, rp, tp, Nb, Hin, Win, C, K, P, Q, R, S,
+                       stride, pad);
   else
     hipLaunchKernelGGL((dcr_conv::conv_nhwc_fwd_v2_kernel<32>), grid, block, 0,
                        st, (const dcr_conv::bf16_t*)x,
                        (const dcr_conv::bf16_t*)w, bias, (dcr_conv::bf16_t*)y,
-                       ws, splitz, Nb, Hin, Win, C, K, P, Q, R, S, stride, pad);
+                       ws, splitz, rp, tp, Nb, Hin, Win, C, K, P, Q, R, S,
+                       stride, pad);
   if (splitz > 1) {
     long total = NPQ * K;
     long b = (total / 4 + 255) / 256;
     if (b > 8192) b = 8192;
     hipLaunchKernelGGL(dcr_conv::conv_splitk_finalize_kernel,
-                       dim3((unsigned)b), dim3(256), 0, st, ws, bias,
-                       (dcr_conv::bf16_t*)y, total, K);
+                       dim3((unsigned)b), dim3(256), 0, st, ws, bias, rp, tp,
+                       (dcr_conv::bf16_t*)y, total, K, P * Q);
   }
 }
 

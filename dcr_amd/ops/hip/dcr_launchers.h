@@ -85,9 +85,10 @@ void conv_nhwc_fwd_launch(const void* x, const void* w, const float* bias,
                           int P, int Q, int R, int S, int stride, int pad,
                           hipStream_t st);
 void conv_nhwc_fwd_v2_launch(const void* x, const void* w, const float* bias,
-                             void* y, float* ws, int splitz, int Nb, int Hin,
-                             int Win, int C, int K, int P, int Q, int R, int S,
-                             int stride, int pad, hipStream_t st);
+                             void* y, float* ws, int splitz, const void* res,
+                             const void* temb, int Nb, int Hin, int Win, int C,
+                             int K, int P, int Q, int R, int S, int stride,
+                             int pad, hipStream_t st);
 // conv_nhwc_bwd.hip (dgrad + wgrad + fused bias-grad; DCR_NATIVE_CONV_BWD)
 void conv_bwd_weight_launch(const void* dy, const void* x, float* dw_ws,
                             int Nb, int Hin, int Win, int C, int K, int P,
