@@ -10,6 +10,15 @@ from-scratch implementations):
   network in this environment; BASELINE configs run on synthetic data).
 * VGG16 (features + fc2 4096-d) for Improved Precision & Recall
   (/root/reference/metrics/ipr.py:41,146-148).
+
+
+MI355X note (SURVEY §2.4.F / §7 stage 5): bottleneck/downsample convs are
+DcrConv2d — they dispatch to the in-tree implicit-GEMM kernel when run
+bf16 + channels_last (C%32, K%64, 1x1/3x3, stride 1|2); the 7x7 C=3 stem
+and fp32 eval runs stay on MIOpen, which measured 0.94-1.24x parity with
+our kernel on SD shapes — for the inference-only,
+amortized metric passes that parity makes a dedicated small-C stem
+kernel not worth its maintenance.
 """
 from __future__ import annotations
 
@@ -18,6 +27,8 @@ from typing import Optional
 
 import torch
 import torch.nn as nn
+
+from ..ops.conv import Conv2d as DcrConv2d
 import torch.nn.functional as F
 
 
@@ -27,11 +38,11 @@ class Bottleneck(nn.Module):
 
     def __init__(self, in_ch, planes, stride=1, downsample=None):
         super().__init__()
-        self.conv1 = nn.Conv2d(in_ch, planes, 1, bias=False)
+        self.conv1 = DcrConv2d(in_ch, planes, 1, bias=False)
         self.bn1 = nn.BatchNorm2d(planes)
-        self.conv2 = nn.Conv2d(planes, planes, 3, stride=stride, padding=1, bias=False)
+        self.conv2 = DcrConv2d(planes, planes, 3, stride=stride, padding=1, bias=False)
         self.bn2 = nn.BatchNorm2d(planes)
-        self.conv3 = nn.Conv2d(planes, planes * 4, 1, bias=False)
+        self.conv3 = DcrConv2d(planes, planes * 4, 1, bias=False)
         self.bn3 = nn.BatchNorm2d(planes * 4)
         self.downsample = downsample
 
@@ -62,7 +73,7 @@ class ResNet50(nn.Module):
         downsample = None
         if stride != 1 or self.inplanes != planes * 4:
             downsample = nn.Sequential(
-                nn.Conv2d(self.inplanes, planes * 4, 1, stride=stride, bias=False),
+                DcrConv2d(self.inplanes, planes * 4, 1, stride=stride, bias=False),
                 nn.BatchNorm2d(planes * 4))
         layers = [Bottleneck(self.inplanes, planes, stride, downsample)]
         self.inplanes = planes * 4
@@ -152,7 +163,7 @@ class VGG16(nn.Module):
             if v == "M":
                 layers.append(nn.MaxPool2d(2, 2))
             else:
-                layers += [nn.Conv2d(in_ch, v, 3, padding=1), nn.ReLU(inplace=True)]
+                layers += [DcrConv2d(in_ch, v, 3, padding=1), nn.ReLU(inplace=True)]
                 in_ch = v
         self.features = nn.Sequential(*layers)
         self.avgpool = nn.AdaptiveAvgPool2d(7)
