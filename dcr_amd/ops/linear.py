@@ -8,16 +8,16 @@ diffusers) through cuBLAS. Round-2 measurement on MI355X
 * fwd / dgrad: the in-tree 128x128 2-barrier MFMA kernel reaches
   77-373 TF = 0.42-0.59x rocBLAS/Tensile on the SD-2.1 shapes — Tensile
   keeps those passes.
-* wgrad + fused bias-grad: dW = dy^T @ x with the bias column-sum fused
-  into the staging pass measures 0.98-1.13x (vs Tensile mm + separate
-  fp32 sum) when the contraction (batch*tokens) is >= 4096 — i.e. the
-  res32/res16 transformer levels — and loses below that. So the wgrad
-  pass dispatches natively exactly in its winning regime, which also
-  deletes those layers' separate aten bias-grad reduce kernels
-  (VERDICT r01 "aten glue" item).
+* wgrad + fused bias-grad measured 0.98-1.13x in the ISOLATED microbench
+  at contraction >= 4096, but the whole-model profile
+  (profiles/r02_prof_head.md) shows it costing 40 ms/6-steps vs the
+  ~25 ms of Tensile wgrad + aten bias-reduce it replaced — the
+  microbench's rocBLAS side was inflated by an fp32 cast in the
+  reference timing. Net in-model loss, so native linear dispatch is
+  DEFAULT OFF.
 
-DCR_NATIVE_GEMM=0 opts out entirely; DCR_NATIVE_GEMM=full forces all
-three passes native (kernel iteration / benching only).
+DCR_NATIVE_GEMM=1 enables the hybrid (native wgrad) path;
+DCR_NATIVE_GEMM=full forces all three passes native (benching only).
 """
 from __future__ import annotations
 
@@ -32,7 +32,7 @@ from . import use_hip, require_hip, count_dispatch
 
 
 def _mode() -> str:
-    return os.environ.get("DCR_NATIVE_GEMM", "1")
+    return os.environ.get("DCR_NATIVE_GEMM", "0")
 
 
 def _wgrad_eligible(x2d: torch.Tensor, weight: torch.Tensor) -> bool:
