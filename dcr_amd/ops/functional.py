@@ -173,9 +173,13 @@ class _FlashAttention(torch.autograd.Function):
     def forward(ctx, q, k, v, scale, causal):
         m = require_hip("attn")
         count_dispatch('attention')
-        # DCR_ATTN_V2=1: bit-exact masked-tail-skip draft (round-2 A/B)
-        fwd = m.attn_fwd_v2 if os.environ.get("DCR_ATTN_V2") == "1" \
-            else m.attn_fwd
+        # DCR_ATTN_V2 / DCR_ATTN_V3: env-gated draft schedules (A/B)
+        if os.environ.get("DCR_ATTN_V3") == "1":
+            fwd = m.attn_fwd_v3
+        elif os.environ.get("DCR_ATTN_V2") == "1":
+            fwd = m.attn_fwd_v2
+        else:
+            fwd = m.attn_fwd
         o, lse = fwd(q, k, v, scale, causal)
         ctx.save_for_backward(q, k, v, o, lse)
         ctx.scale = scale

@@ -662,6 +662,186 @@ __global__ void mfma_probe_kernel(const bf16_t* __restrict__ A,
 }
 
 // ==========================================================================
+// Forward v3 (round-2 draft, DCR_ATTN_V3): same math as v1 with the
+// guide's T14 async-STAGE split + T5 setprio applied to the L >= 1024
+// regime where v1 trails AOTriton SDPA (182 vs 385 TF @1024):
+//   * K/V tiles double-buffered; the NEXT tile's global loads are issued
+//     into registers BEFORE this tile's MFMA/softmax phase (HBM latency
+//     hides under compute), written to the alternate LDS buffer after it
+//   * ONE barrier per K/V tile instead of two
+//   * s_setprio(1) around the MFMA clusters (T5: phase-split scheduling)
+// ==========================================================================
+__global__ __launch_bounds__(256)
+void attn_fwd_v3_kernel(const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
+                        const bf16_t* __restrict__ v, bf16_t* __restrict__ o,
+                        float* __restrict__ lse, int Lq, int Lk, int H,
+                        float scale, int causal) {
+  __shared__ short sQ[TILE * PITCH];
+  __shared__ short sK[2][TILE * PITCH];
+  __shared__ short sVT[2][TILE * PITCH];
+  __shared__ short sP[TILE * PITCH];
+
+  const int bh = blockIdx.y;
+  const int b = bh / H, h = bh % H;
+  const int q0 = blockIdx.x * TILE;
+  const long qrs = (long)H * DHEAD;
+  const bf16_t* qp = q + (((long)b * Lq + q0) * H + h) * DHEAD;
+  const bf16_t* kp = k + ((long)b * Lk * H + h) * DHEAD;
+  const bf16_t* vp = v + ((long)b * Lk * H + h) * DHEAD;
+  bf16_t* op = o + ((long)b * Lq * H + h) * DHEAD;
+
+  const int t = threadIdx.x;
+  const int st_row = t >> 2;
+  const int st_col = (t & 3) * 16;
+
+  // register staging: K rows as 2 uint4, V rows as 16 shorts (pre-split
+  // for the transposed write)
+  uint4 ka, kb;
+  short vv[16];
+
+  auto issue_tile = [&](int kv0) {
+    const int valid = Lk - kv0;
+    ka = make_uint4(0, 0, 0, 0);
+    kb = ka;
+    if (st_row < valid) {
+      const uint4* pk = reinterpret_cast<const uint4*>(
+          kp + ((long)(kv0 + st_row)) * qrs + st_col);
+      ka = pk[0];
+      kb = pk[1];
+      const uint4* pv = reinterpret_cast<const uint4*>(
+          vp + ((long)(kv0 + st_row)) * qrs + st_col);
+      *reinterpret_cast<uint4*>(vv) = pv[0];
+      *reinterpret_cast<uint4*>(vv + 8) = pv[1];
+    } else {
+#pragma unroll
+      for (int j = 0; j < 16; ++j) vv[j] = 0;
+    }
+  };
+  auto write_tile = [&](int buf) {
+    uint4* d = reinterpret_cast<uint4*>(sK[buf] + st_row * PITCH + st_col);
+    d[0] = ka;
+    d[1] = kb;
+#pragma unroll
+    for (int j = 0; j < 16; ++j)
+      sVT[buf][(st_col + j) * PITCH + st_row] = vv[j];
+  };
+
+  load_tile(qp, qrs, Lq - q0, sQ);
+
+  const int lane = t & 63;
+  const int wid = t >> 6;
+  const int l16 = lane & 15;
+  const int kgrp = lane >> 4;
+  const int wrow0 = wid * 16;
+
+  float row_max[4] = {-1e30f, -1e30f, -1e30f, -1e30f};
+  float row_sum[4] = {0.f, 0.f, 0.f, 0.f};
+  f32x4_t acc_o[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) acc_o[i] = {0.f, 0.f, 0.f, 0.f};
+
+  const int kv_end = causal ? min(Lk, q0 + TILE) : Lk;
+
+  issue_tile(0);
+  write_tile(0);
+  __syncthreads();
+
+  int cur = 0;
+  for (int kv0 = 0; kv0 < kv_end; kv0 += TILE) {
+    const bool more = (kv0 + TILE < kv_end);
+    if (more) issue_tile(kv0 + TILE);
+
+    bf16x8 qf0 = frag(sQ, wrow0 + l16, kgrp * 8);
+    bf16x8 qf1 = frag(sQ, wrow0 + l16, kgrp * 8 + 32);
+    f32x4_t s_frag[4];
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int ns = 0; ns < 4; ++ns) {
+      f32x4_t acc = {0.f, 0.f, 0.f, 0.f};
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          qf0, frag(sK[cur], ns * 16 + l16, kgrp * 8), acc, 0, 0, 0);
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          qf1, frag(sK[cur], ns * 16 + l16, kgrp * 8 + 32), acc, 0, 0, 0);
+      s_frag[ns] = acc;
+    }
+    __builtin_amdgcn_s_setprio(0);
+
+    float tile_max[4] = {-1e30f, -1e30f, -1e30f, -1e30f};
+#pragma unroll
+    for (int ns = 0; ns < 4; ++ns) {
+      const int key = kv0 + ns * 16 + l16;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float sv = s_frag[ns][r] * scale;
+        const int qrow = q0 + wrow0 + kgrp * 4 + r;
+        if (key >= Lk || (causal && key > qrow)) sv = -1e30f;
+        s_frag[ns][r] = sv;
+        tile_max[r] = fmaxf(tile_max[r], sv);
+      }
+    }
+    float alpha[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      float tm = qmax(tile_max[r]);
+      float mnew = fmaxf(row_max[r], tm);
+      alpha[r] = (mnew <= -1e29f) ? 1.f : __expf(row_max[r] - mnew);
+      row_max[r] = mnew;
+    }
+
+    float psum[4] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int ns = 0; ns < 4; ++ns) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float pv = (s_frag[ns][r] <= -1e29f)
+                       ? 0.f : __expf(s_frag[ns][r] - row_max[r]);
+        psum[r] += pv;
+        sP[(wrow0 + kgrp * 4 + r) * PITCH + ns * 16 + l16] = f2bf_rne(pv);
+      }
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) row_sum[r] = row_sum[r] * alpha[r] + qsum(psum[r]);
+
+#pragma unroll
+    for (int ds_ = 0; ds_ < 4; ++ds_)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) acc_o[ds_][r] *= alpha[r];
+
+    bf16x8 pf0 = frag(sP, wrow0 + l16, kgrp * 8);
+    bf16x8 pf1 = frag(sP, wrow0 + l16, kgrp * 8 + 32);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int ds_ = 0; ds_ < 4; ++ds_) {
+      acc_o[ds_] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          pf0, frag(sVT[cur], ds_ * 16 + l16, kgrp * 8), acc_o[ds_], 0, 0, 0);
+      acc_o[ds_] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          pf1, frag(sVT[cur], ds_ * 16 + l16, kgrp * 8 + 32), acc_o[ds_], 0, 0, 0);
+    }
+    __builtin_amdgcn_s_setprio(0);
+
+    if (more) {
+      // previous readers of buf^1 finished before the LAST barrier, so
+      // the write needs no pre-barrier; the one barrier publishes it
+      write_tile(cur ^ 1);
+      __syncthreads();
+      cur ^= 1;
+    }
+  }
+
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int qrow = q0 + wrow0 + kgrp * 4 + r;
+    if (qrow >= Lq) continue;
+    const float inv = (row_sum[r] > 0.f) ? 1.f / row_sum[r] : 0.f;
+#pragma unroll
+    for (int ds_ = 0; ds_ < 4; ++ds_)
+      op[(long)qrow * qrs + ds_ * 16 + l16] = __float2bfloat16(acc_o[ds_][r] * inv);
+    if (l16 == 0 && lse != nullptr)
+      lse[(long)bh * Lq + qrow] = row_max[r] + __logf(fmaxf(row_sum[r], 1e-30f));
+  }
+}
+
+// ==========================================================================
 // Generalized-head-dim forward (SD-1.4 / sd_mitigation parity: head_dim
 // 40/80/160 — /root/reference/sd_mitigation.py:46; inference-only, so no
 // backward). Same tile discipline as attn_fwd_kernel with the head dim
@@ -855,6 +1035,17 @@ void attn_fwd_v2_launch(const void* q, const void* k, const void* v, void* o,
                         bool causal, hipStream_t s) {
   dim3 grid((Lq + TILE - 1) / TILE, BH), block(256);
   hipLaunchKernelGGL(dcr_attn::attn_fwd_v2_kernel, grid, block, 0, s,
+                     (const dcr_attn::bf16_t*)q, (const dcr_attn::bf16_t*)k,
+                     (const dcr_attn::bf16_t*)v, (dcr_attn::bf16_t*)o, lse,
+                     Lq, Lk, H, scale, causal ? 1 : 0);
+}
+
+// v3: T14 async-stage + one barrier/tile + setprio (round-2 draft)
+void attn_fwd_v3_launch(const void* q, const void* k, const void* v, void* o,
+                        float* lse, int BH, int Lq, int Lk, int H, float scale,
+                        bool causal, hipStream_t s) {
+  dim3 grid((Lq + TILE - 1) / TILE, BH), block(256);
+  hipLaunchKernelGGL(dcr_attn::attn_fwd_v3_kernel, grid, block, 0, s,
                      (const dcr_attn::bf16_t*)q, (const dcr_attn::bf16_t*)k,
                      (const dcr_attn::bf16_t*)v, (dcr_attn::bf16_t*)o, lse,
                      Lq, Lk, H, scale, causal ? 1 : 0);

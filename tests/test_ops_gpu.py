@@ -703,3 +703,22 @@ def test_resnet_block_fused_matches_unfused(ext):
     assert xg.grad is not None and torch.isfinite(xg.grad.float()).all()
     assert tg.grad is not None and torch.isfinite(tg.grad.float()).all()
     assert blk.time_emb_proj.weight.grad is not None
+
+
+@pytest.mark.parametrize("Lq,Lk", [(64, 64), (256, 77), (1024, 1024),
+                                   (100, 33)])
+def test_attn_fwd_v3_matches_v1(ext, Lq, Lk):
+    """v3 (async-stage + 1-barrier + setprio draft) computes the same
+    math as v1 in the same order -> bit-equal outputs."""
+    torch.manual_seed(Lq + Lk)
+    B, H, D = 2, 3, 64
+    q = torch.randn(B, Lq, H, D, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(B, Lk, H, D, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(B, Lk, H, D, device="cuda", dtype=torch.bfloat16)
+    o1, l1 = ext.attn_fwd(q, k, v, 0.125, False)
+    o3, l3 = ext.attn_fwd_v3(q, k, v, 0.125, False)
+    assert torch.equal(o1, o3)
+    assert torch.equal(l1, l3)
+    oc1, lc1 = ext.attn_fwd(q, k, v[:, :Lk], 0.125, True)
+    oc3, lc3 = ext.attn_fwd_v3(q, k, v[:, :Lk], 0.125, True)
+    assert torch.equal(oc1, oc3)
