@@ -374,8 +374,8 @@ at::Tensor attn_fwd_gen(at::Tensor q, at::Tensor k, at::Tensor v,
   return o;
 }
 
-// round-2 draft: bit-exact masked-tail skip — validated vs attn_fwd on
-// hardware before any dispatch (DCR_ATTN_V2=1 gates its tests)
+// round-2 draft: T14 async K/V staging + one barrier per tile + setprio
+// (DCR_ATTN_V3=1 gates its tests and dispatch)
 std::vector<at::Tensor> attn_fwd_v3(at::Tensor q, at::Tensor k, at::Tensor v,
                                     double scale, bool causal) {
   TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16,
@@ -388,7 +388,8 @@ std::vector<at::Tensor> attn_fwd_v3(at::Tensor q, at::Tensor k, at::Tensor v,
   attn_fwd_v3_launch(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
                      lse.data_ptr<float>(), (int)(B * H), (int)Lq, (int)Lk,
                      (int)H, (float)scale, causal, cur_stream());
-  return {o, lse}
+  return {o, lse};
+}
 
 std::vector<at::Tensor> attn_fwd_v2(at::Tensor q, at::Tensor k, at::Tensor v,
                                     double scale, bool causal) {
