@@ -670,6 +670,14 @@ __global__ void mfma_probe_kernel(const bf16_t* __restrict__ A,
 //     hides under compute), written to the alternate LDS buffer after it
 //   * ONE barrier per K/V tile instead of two
 //   * s_setprio(1) around the MFMA clusters (T5: phase-split scheduling)
+//
+// MEASURED (r02c5, bit-equal vs v1): 0.75-0.82x of v1 at L >= 256 —
+// the second K/V buffer pushes LDS to 110 KB -> 1 block/CU (v1: 73 KB,
+// 2 blocks/CU) and the lost wave-level overlap outweighs the staging
+// pipeline, exactly like the conv v3 draft. NOT dispatched; kept as the
+// documented negative result. The path to SDPA-class throughput at
+// L >= 1024 is the restructured schedule (swapped QK^T, in-register
+// softmax), not staging grafts on this structure.
 // ==========================================================================
 __global__ __launch_bounds__(256)
 void attn_fwd_v3_kernel(const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
