@@ -850,6 +850,193 @@ void attn_fwd_v3_kernel(const bf16_t* __restrict__ q, const bf16_t* __restrict__
 }
 
 // ==========================================================================
+// Forward v4 (round-2, DCR_ATTN_V4): restructured schedule for the
+// L >= 256 self-attention shapes where the v1 structure trails AOTriton
+// SDPA. Following the guide's 8-warp ladder:
+//   * 8 waves x 32 q-rows = 256 q-rows per block; K/V staging amortizes
+//     over 4x the rows of v1 and Q lives in registers.
+//   * swapped QK^T — mfma_f32_32x32x16_bf16(A=K, B=Q) gives S^T[key][q]
+//     with each lane holding ONE q column (l&31): the entire online
+//     softmax is lane-local (31 fmax/adds + one lane^32 exchange), no
+//     quarter-wave shuffle reductions and no sP LDS round-trip.
+//   * P stays in registers: bf16 pack + v_permlane32_swap assembles the
+//     PV B-fragments directly (guide T12 layout algebra).
+//   * PV is also swapped — mfma(A=V^T, B=P) accumulates O^T[d][q] so the
+//     alpha rescale stays lane-local too.
+// Fragment maps (guide §3, gfx950): 32x32x16 A/B: lane l holds row l&31,
+// k = (l>>5)*8 + j; C/D: col = l&31, row = (r&3) + 8*(r>>2) + 4*(l>>5).
+// ==========================================================================
+typedef __attribute__((ext_vector_type(16))) float f32x16_t;
+
+__global__ __launch_bounds__(512)
+void attn_fwd_v4_kernel(const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
+                        const bf16_t* __restrict__ v, bf16_t* __restrict__ o,
+                        float* __restrict__ lse, int Lq, int Lk, int H,
+                        float scale, int causal) {
+  __shared__ short sK[TILE * PITCH];    // [64 keys][72] (d contiguous)
+  __shared__ short sVT[TILE * PITCH];   // [64 d][72]   (keys contiguous)
+
+  const int bh = blockIdx.y;
+  const int b = bh / H, h = bh % H;
+  const int q0 = blockIdx.x * 256;
+  const long qrs = (long)H * DHEAD;
+  const bf16_t* kp = k + ((long)b * Lk * H + h) * DHEAD;
+  const bf16_t* vp = v + ((long)b * Lk * H + h) * DHEAD;
+  bf16_t* op = o + ((long)b * Lq * H + h) * DHEAD;
+
+  const int t = threadIdx.x;
+  const int wid = t >> 6;
+  const int lane = t & 63;
+  const int l32 = lane & 31;
+  const int hi = lane >> 5;
+  const int qrow = q0 + wid * 32 + l32;  // this lane's q row (all regs)
+
+  // Q fragments in registers: qf[c] = Q[qrow][c*16 + hi*8 .. +8]
+  bf16x8 qf[4];
+  if (qrow < Lq) {
+    const bf16_t* qp = q + (((long)b * Lq + qrow) * H + h) * DHEAD;
+#pragma unroll
+    for (int c = 0; c < 4; ++c)
+      qf[c] = *reinterpret_cast<const bf16x8*>(qp + c * 16 + hi * 8);
+  } else {
+#pragma unroll
+    for (int c = 0; c < 4; ++c) qf[c] = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+  }
+
+  // cooperative staging: 512 threads, 8 threads/row, one uint4 each
+  const int st_row = t >> 3;
+  const int st_col = (t & 7) * 8;
+
+  float m_run = -1e30f, l_run = 0.f;
+  f32x16_t oacc[2];
+#pragma unroll
+  for (int dt = 0; dt < 2; ++dt)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) oacc[dt][r] = 0.f;
+
+  const int kv_end = causal ? min(Lk, q0 + 256) : Lk;
+
+  for (int kv0 = 0; kv0 < kv_end; kv0 += TILE) {
+    __syncthreads();
+    {
+      const int valid = Lk - kv0;
+      uint4 kv4 = make_uint4(0, 0, 0, 0);
+      short vv[8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) vv[j] = 0;
+      if (st_row < valid) {
+        kv4 = *reinterpret_cast<const uint4*>(
+            kp + (long)(kv0 + st_row) * qrs + st_col);
+        *reinterpret_cast<uint4*>(vv) = *reinterpret_cast<const uint4*>(
+            vp + (long)(kv0 + st_row) * qrs + st_col);
+      }
+      *reinterpret_cast<uint4*>(sK + st_row * PITCH + st_col) = kv4;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        sVT[(st_col + j) * PITCH + st_row] = vv[j];
+    }
+    __syncthreads();
+
+    // S^T = scale * (K Q^T): two 32-key subtiles, contraction D in 16s
+    f32x16_t sacc[2];
+#pragma unroll
+    for (int st = 0; st < 2; ++st) {
+      f32x16_t acc;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) acc[r] = 0.f;
+#pragma unroll
+      for (int c = 0; c < 4; ++c) {
+        bf16x8 kf = *reinterpret_cast<const bf16x8*>(
+            sK + (st * 32 + l32) * PITCH + c * 16 + hi * 8);
+        acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[c], acc, 0, 0, 0);
+      }
+      sacc[st] = acc;
+    }
+
+    // mask + lane-local online softmax (this lane owns row qrow)
+    float tmax = -1e30f;
+#pragma unroll
+    for (int st = 0; st < 2; ++st)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int key = kv0 + st * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+        float sv = sacc[st][r] * scale;
+        if (key >= Lk || (causal && key > qrow)) sv = -1e30f;
+        sacc[st][r] = sv;
+        tmax = fmaxf(tmax, sv);
+      }
+    tmax = fmaxf(tmax, __shfl_xor(tmax, 32, 64));
+    const float mnew = fmaxf(m_run, tmax);
+    const float alpha = (mnew <= -1e29f) ? 1.f : __expf(m_run - mnew);
+    m_run = mnew;
+
+    float psum = 0.f;
+#pragma unroll
+    for (int st = 0; st < 2; ++st)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        float pv = (sacc[st][r] <= -1e29f) ? 0.f
+                                           : __expf(sacc[st][r] - m_run);
+        sacc[st][r] = pv;
+        psum += pv;
+      }
+    psum += __shfl_xor(psum, 32, 64);
+    l_run = l_run * alpha + psum;
+
+#pragma unroll
+    for (int dt = 0; dt < 2; ++dt)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) oacc[dt][r] *= alpha;
+
+    // P -> bf16 fragments via pack + permlane32_swap, then O^T += V^T P
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      const int st = c >> 1;
+      const int r8 = 8 * (c & 1);
+      unsigned u[4];
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        __hip_bfloat16 lo = __float2bfloat16(sacc[st][r8 + 2 * i]);
+        __hip_bfloat16 hi_ = __float2bfloat16(sacc[st][r8 + 2 * i + 1]);
+        u[i] = (unsigned)*reinterpret_cast<unsigned short*>(&lo) |
+               ((unsigned)*reinterpret_cast<unsigned short*>(&hi_) << 16);
+      }
+      auto s0 = __builtin_amdgcn_permlane32_swap(u[0], u[2], false, false);
+      auto s1 = __builtin_amdgcn_permlane32_swap(u[1], u[3], false, false);
+      unsigned w[4] = {(unsigned)s0[0], (unsigned)s1[0],
+                       (unsigned)s0[1], (unsigned)s1[1]};
+      bf16x8 pfrag = *reinterpret_cast<const bf16x8*>(w);
+#pragma unroll
+      for (int dt = 0; dt < 2; ++dt) {
+        bf16x8 vf = *reinterpret_cast<const bf16x8*>(
+            sVT + (dt * 32 + l32) * PITCH + c * 16 + hi * 8);
+        oacc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, pfrag,
+                                                           oacc[dt], 0, 0, 0);
+      }
+    }
+  }
+
+  if (qrow >= Lq) return;
+  const float inv = (l_run > 0.f) ? 1.f / l_run : 0.f;
+#pragma unroll
+  for (int dt = 0; dt < 2; ++dt)
+#pragma unroll
+    for (int rq = 0; rq < 4; ++rq) {
+      const int d0 = dt * 32 + 8 * rq + 4 * hi;
+      ushort4 pk;
+#pragma unroll
+      for (int e = 0; e < 4; ++e) {
+        __hip_bfloat16 hv = __float2bfloat16(oacc[dt][rq * 4 + e] * inv);
+        (&pk.x)[e] = *reinterpret_cast<unsigned short*>(&hv);
+      }
+      *reinterpret_cast<uint2*>(op + (long)qrow * qrs + d0) =
+          *reinterpret_cast<uint2*>(&pk);
+    }
+  if (hi == 0 && lse != nullptr)
+    lse[(long)bh * Lq + qrow] = m_run + __logf(fmaxf(l_run, 1e-30f));
+}
+
+// ==========================================================================
 // Generalized-head-dim forward (SD-1.4 / sd_mitigation parity: head_dim
 // 40/80/160 — /root/reference/sd_mitigation.py:46; inference-only, so no
 // backward). Same tile discipline as attn_fwd_kernel with the head dim
@@ -1054,6 +1241,17 @@ void attn_fwd_v3_launch(const void* q, const void* k, const void* v, void* o,
                         bool causal, hipStream_t s) {
   dim3 grid((Lq + TILE - 1) / TILE, BH), block(256);
   hipLaunchKernelGGL(dcr_attn::attn_fwd_v3_kernel, grid, block, 0, s,
+                     (const dcr_attn::bf16_t*)q, (const dcr_attn::bf16_t*)k,
+                     (const dcr_attn::bf16_t*)v, (dcr_attn::bf16_t*)o, lse,
+                     Lq, Lk, H, scale, causal ? 1 : 0);
+}
+
+// v4: swapped-QK^T in-register-softmax schedule (round-2)
+void attn_fwd_v4_launch(const void* q, const void* k, const void* v, void* o,
+                        float* lse, int BH, int Lq, int Lk, int H, float scale,
+                        bool causal, hipStream_t s) {
+  dim3 grid((Lq + 255) / 256, BH), block(512);
+  hipLaunchKernelGGL(dcr_attn::attn_fwd_v4_kernel, grid, block, 0, s,
                      (const dcr_attn::bf16_t*)q, (const dcr_attn::bf16_t*)k,
                      (const dcr_attn::bf16_t*)v, (dcr_attn::bf16_t*)o, lse,
                      Lq, Lk, H, scale, causal ? 1 : 0);

@@ -376,6 +376,20 @@ at::Tensor attn_fwd_gen(at::Tensor q, at::Tensor k, at::Tensor v,
 
 // round-2 draft: T14 async K/V staging + one barrier per tile + setprio
 // (DCR_ATTN_V3=1 gates its tests and dispatch)
+std::vector<at::Tensor> attn_fwd_v4(at::Tensor q, at::Tensor k, at::Tensor v,
+                                    double scale, bool causal) {
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16,
+              "attn: bf16 CUDA only");
+  TORCH_CHECK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
+  int64_t B, H, Lq, Lk;
+  attn_dims(q, k, B, H, Lq, Lk);
+  auto o = at::empty_like(q);
+  auto lse = at::empty({B * H, Lq}, q.options().dtype(at::kFloat));
+  attn_fwd_v4_launch(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
+                     lse.data_ptr<float>(), (int)(B * H), (int)Lq, (int)Lk,
+                     (int)H, (float)scale, causal, cur_stream());
+  return {o, lse}
+
 std::vector<at::Tensor> attn_fwd_v3(at::Tensor q, at::Tensor k, at::Tensor v,
                                     double scale, bool causal) {
   TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16,
@@ -622,6 +636,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("attn_fwd_gen", &attn_fwd_gen);
   mod.def("attn_fwd_v2", &attn_fwd_v2);
   mod.def("attn_fwd_v3", &attn_fwd_v3);
+  mod.def("attn_fwd_v4", &attn_fwd_v4);
   mod.def("attn_bwd", &attn_bwd);
   mod.def("mfma_probe", &mfma_probe);
   mod.def("groupnorm_silu_fwd", &groupnorm_silu_fwd);

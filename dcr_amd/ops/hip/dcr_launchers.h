@@ -72,6 +72,10 @@ void attn_fwd_v2_launch(const void* q, const void* k, const void* v, void* o,
 void attn_fwd_v3_launch(const void* q, const void* k, const void* v, void* o,
                         float* lse, int BH, int Lq, int Lk, int H, float scale,
                         bool causal, hipStream_t s);
+// v4: swapped-QK^T in-register-softmax schedule (round-2)
+void attn_fwd_v4_launch(const void* q, const void* k, const void* v, void* o,
+                        float* lse, int BH, int Lq, int Lk, int H, float scale,
+                        bool causal, hipStream_t s);
 // generalized head-dim forward (SD-1.4 40/80/160; inference-only)
 void attn_fwd_gen_launch(const void* q, const void* k, const void* v, void* o,
                          float* lse, int BH, int Lq, int Lk, int H, int D,

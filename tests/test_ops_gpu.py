@@ -722,3 +722,26 @@ def test_attn_fwd_v3_matches_v1(ext, Lq, Lk):
     oc1, lc1 = ext.attn_fwd(q, k, v[:, :Lk], 0.125, True)
     oc3, lc3 = ext.attn_fwd_v3(q, k, v[:, :Lk], 0.125, True)
     assert torch.equal(oc1, oc3)
+
+
+@pytest.mark.parametrize("Lq,Lk,causal", [(256, 256, False), (1024, 1024, False),
+                                          (256, 77, False), (300, 100, False),
+                                          (512, 512, True)])
+def test_attn_fwd_v4_matches_ref(ext, Lq, Lk, causal):
+    """v4 (swapped-QK^T in-register softmax, 32x32 MFMA): vs fp32 SDPA
+    (random asymmetric data catches any fragment-layout transpose) and
+    LSE vs v1."""
+    torch.manual_seed(Lq + Lk)
+    B, H, D = 2, 3, 64
+    q = torch.randn(B, Lq, H, D, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(B, Lk, H, D, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(B, Lk, H, D, device="cuda", dtype=torch.bfloat16)
+    scale = 1.0 / D ** 0.5
+    o4, l4 = ext.attn_fwd_v4(q, k, v, scale, causal)
+    ref = F.scaled_dot_product_attention(
+        q.permute(0, 2, 1, 3).float(), k.permute(0, 2, 1, 3).float(),
+        v.permute(0, 2, 1, 3).float(), is_causal=causal,
+        scale=scale).permute(0, 2, 1, 3)
+    _close(o4, ref, 2e-2)
+    o1, l1 = ext.attn_fwd(q, k, v, scale, causal)
+    assert (l4 - l1).abs().max().item() < 1e-4
