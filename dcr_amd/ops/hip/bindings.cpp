@@ -388,7 +388,8 @@ std::vector<at::Tensor> attn_fwd_v4(at::Tensor q, at::Tensor k, at::Tensor v,
   attn_fwd_v4_launch(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
                      lse.data_ptr<float>(), (int)(B * H), (int)Lq, (int)Lk,
                      (int)H, (float)scale, causal, cur_stream());
-  return {o, lse}
+  return {o, lse};
+}
 
 std::vector<at::Tensor> attn_fwd_v3(at::Tensor q, at::Tensor k, at::Tensor v,
                                     double scale, bool causal) {
