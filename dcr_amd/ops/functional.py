@@ -173,11 +173,18 @@ class _FlashAttention(torch.autograd.Function):
     def forward(ctx, q, k, v, scale, causal):
         m = require_hip("attn")
         count_dispatch('attention')
-        # DCR_ATTN_V2 / DCR_ATTN_V3: env-gated draft schedules (A/B)
+        # schedule dispatch (measured, gpurun_out/r02c7): the swapped-QK^T
+        # in-register-softmax v4 wins at Lq >= 256 (1.17x @256, 1.27x
+        # @1024, 1.73x @4096 vs v1); the 64x64-tile v1 stays for short
+        # sequences and CLIP's causal 77. v4's LSE is bit-identical so
+        # the shared FlashAttention-2 backward applies to both.
+        # DCR_ATTN_V2/V3: env-gated draft schedules (A/B only).
         if os.environ.get("DCR_ATTN_V3") == "1":
             fwd = m.attn_fwd_v3
         elif os.environ.get("DCR_ATTN_V2") == "1":
             fwd = m.attn_fwd_v2
+        elif q.shape[1] >= 256 and os.environ.get("DCR_ATTN_V4", "1") != "0":
+            fwd = m.attn_fwd_v4
         else:
             fwd = m.attn_fwd
         o, lse = fwd(q, k, v, scale, causal)
