@@ -121,3 +121,33 @@ def test_dcr_linear_cpu_fallback_and_state_dict():
     # functional fallback identical to F.linear on CPU
     y = dcr_linear(x, ref.weight, ref.bias)
     assert torch.equal(y, ref(x))
+
+
+def test_fused_adamw_preserves_channels_last():
+    """Flattening must keep channels_last conv weights channels_last —
+    a plain view_as silently reverted them to NCHW, costing a per-call
+    aten weight relayout and disabling the native conv for every trained
+    module (found via scripts/profile_aten.py, round 2)."""
+    import torch
+    from dcr_amd.ops.adamw import FusedAdamW
+
+    conv = torch.nn.Conv2d(8, 16, 3, padding=1)
+    conv.to(memory_format=torch.channels_last)
+    w0 = conv.weight.detach().clone()
+    opt = FusedAdamW(conv.parameters(), lr=1e-2)
+    assert conv.weight.is_contiguous(memory_format=torch.channels_last)
+    assert conv.weight.data_ptr() == opt.flat_param.data_ptr()  # still a view
+    assert torch.equal(conv.weight, w0)
+
+    x = torch.randn(2, 8, 4, 4).to(memory_format=torch.channels_last)
+    conv(x).pow(2).mean().backward()
+    opt.step()
+    assert conv.weight.is_contiguous(memory_format=torch.channels_last)
+    assert not torch.equal(conv.weight, w0)
+
+    # gathered grads land in the right arena slots (layout-aligned views)
+    opt.zero_grad()
+    conv(x).pow(2).mean().backward()
+    g_auto = conv.weight.grad.clone()
+    opt.gather_grads()
+    assert torch.allclose(opt._grad_views[0], g_auto, atol=1e-7)
