@@ -211,9 +211,12 @@ class FusedAdamW:
 
     @torch.no_grad()
     def clip_grad_norm_(self, max_norm: float) -> torch.Tensor:
-        """Global L2 grad clip (reference: diff_train.py:657-663, max 1.0)."""
+        """Global L2 grad clip (reference: diff_train.py:657-663, max 1.0).
+        fp32 accumulation via the dtype arg — `.float()` would
+        materialize a full fp32 copy of the 1.7 GB bf16 arena first
+        (~1 ms/step, profile_aten r02c8)."""
         self.gather_grads()  # no-op when finalize() already ran
-        norm = torch.linalg.vector_norm(self.flat_grad.float())
+        norm = torch.linalg.vector_norm(self.flat_grad, dtype=torch.float32)
         scale = max_norm / (norm + 1e-6)
         if float(norm) > max_norm:
             self.flat_grad.mul_(scale.to(self.flat_grad.dtype))
