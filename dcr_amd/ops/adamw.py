@@ -116,12 +116,34 @@ class FusedAdamW:
 
     # -- torch.optim-ish surface ------------------------------------------
     def zero_grad(self, set_to_none: bool = False):
-        # one arena-wide zero (unused/frozen slices must read 0), then
-        # autograd re-allocates per-param grads next backward
-        self.flat_grad.zero_()
+        # the arena-wide zero exists so slices whose params produce no
+        # grad read 0 — when EVERY param was gathered last cycle (the
+        # normal training case) the next cycle's first gather per param
+        # is a copy, so the zero is skippable (1.7 GB write saved).
+        # _ensure_cold_slices() covers the rare dynamic-graph case where
+        # a later cycle leaves some params ungathered.
+        if len(self._gathered) != len(self.params):
+            self.flat_grad.zero_()
+            self._skipped_zero = False
+        else:
+            self._skipped_zero = True
         self._gathered.clear()
         for p in self.params:
             p.grad = None
+
+    def _ensure_cold_slices(self):
+        """After the final gather of a cycle whose zero was skipped, any
+        param that produced no grad this cycle still holds LAST cycle's
+        grad in its arena slice — zero exactly those."""
+        if not getattr(self, "_skipped_zero", False):
+            return
+        if len(self._gathered) == len(self.params):
+            return
+        stale = [self._view_of[id(p)] for p in self.params
+                 if id(p) not in self._gathered]
+        if stale:
+            torch._foreach_zero_(stale)
+        self._skipped_zero = False
 
     @torch.no_grad()
     def gather_grads(self, params=None):
@@ -154,6 +176,7 @@ class FusedAdamW:
     @torch.no_grad()
     def step(self, lr: float | None = None):
         self.gather_grads()  # no-op when finalize() already ran
+        self._ensure_cold_slices()
         if lr is not None:
             self.lr = lr
         self.step_count += 1
@@ -198,6 +221,7 @@ class FusedAdamW:
         4-byte lr write when the schedule moves."""
         assert self.hyper is not None, "built without device_state=True"
         self.gather_grads()  # no-op when finalize() already ran
+        self._ensure_cold_slices()
         if lr is not None and lr != self.lr:
             self.lr = lr
             self.hyper[0].fill_(lr)  # outside any captured graph
@@ -216,6 +240,7 @@ class FusedAdamW:
         materialize a full fp32 copy of the 1.7 GB bf16 arena first
         (~1 ms/step, profile_aten r02c8)."""
         self.gather_grads()  # no-op when finalize() already ran
+        self._ensure_cold_slices()
         norm = torch.linalg.vector_norm(self.flat_grad, dtype=torch.float32)
         scale = max_norm / (norm + 1e-6)
         if float(norm) > max_norm:

@@ -109,6 +109,7 @@ class GradBucketAllReduce:
         """Call after backward(), before optimizer.step()."""
         # any grads not yet moved by a bucket hook (world 1: all of them)
         self.opt.gather_grads()
+        self.opt._ensure_cold_slices()
         if not self._enabled or not self.require_backward_grad_sync:
             self._reset()
             return

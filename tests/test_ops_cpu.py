@@ -151,3 +151,29 @@ def test_fused_adamw_preserves_channels_last():
     g_auto = conv.weight.grad.clone()
     opt.gather_grads()
     assert torch.allclose(opt._grad_views[0], g_auto, atol=1e-7)
+
+
+def test_fused_adamw_zero_skip_and_stale_slice_guard():
+    """zero_grad skips the arena zero after a full-coverage cycle; a
+    later partial-coverage cycle must still read 0 (not last cycle's
+    grad) for params that produced no grad."""
+    import torch
+    from dcr_amd.ops.adamw import FusedAdamW
+
+    a = torch.nn.Linear(4, 4)
+    b = torch.nn.Linear(4, 4)
+    opt = FusedAdamW(list(a.parameters()) + list(b.parameters()), lr=0.0,
+                     weight_decay=0.0)
+    x = torch.randn(2, 4)
+    # cycle 1: both modules produce grads
+    (a(x).sum() + b(x).sum()).backward()
+    opt.step()
+    opt.zero_grad()
+    assert opt._skipped_zero  # full coverage -> skip
+    # cycle 2: only module a runs
+    a(x).pow(2).sum().backward()
+    opt.gather_grads()
+    opt._ensure_cold_slices()
+    for p in b.parameters():
+        v = opt._view_of[id(p)]
+        assert torch.count_nonzero(v) == 0, "stale grads must be zeroed"

@@ -296,7 +296,9 @@ def test_gradient_checkpointing(tmp_path):
     b = next(iter(tr.dataloader))
     loss = tr.train_step(b)
     assert torch.isfinite(loss)
-    assert tr.optimizer.flat_grad.abs().sum() == 0  # zeroed after step
+    # zero_grad contract: autograd grads released; the arena zero is
+    # skipped after a full-coverage cycle (next gathers are copies)
+    assert all(p.grad is None for p in tr.optimizer.params)
 
 
 def test_trainsubset(tmp_path):
