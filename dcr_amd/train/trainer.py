@@ -91,6 +91,7 @@ class Trainer:
         self.global_step = 0
         self.tracker = None
         self.prof = PhaseProfiler()  # enabled via DCR_PROFILE=1
+        self._graph = None  # hipGraph-captured step (enable_hipgraph)
 
     # ------------------------------------------------------------------
     def _build_models(self):
@@ -217,8 +218,92 @@ class Trainer:
         self.ddp = GradBucketAllReduce(self.optimizer, bucket_mb=cfg.ddp_bucket_mb)
 
     # ------------------------------------------------------------------
+    # hipGraph-captured step: the whole micro-step (VAE encode, noise,
+    # text encode, UNet fwd/bwd, grad gather, device-state AdamW with
+    # fused clip) replays as ONE graph launch — the per-step host work
+    # is two H2D batch copies and a 4-byte lr write. Philox RNG is
+    # graph-safe (each replay draws fresh noise/timesteps).
+    def enable_hipgraph(self, batch) -> None:
+        cfg = self.cfg
+        assert self.device.type == "cuda", "hipGraph capture needs a GPU"
+        assert self.pure_bf16 and self.scaler is None, \
+            "graph capture: pure_bf16 only (no GradScaler host branches)"
+        assert cfg.gradient_accumulation_steps == 1
+        assert cfg.mixup_noise_lam == 0, "mixup draws host RNG per step"
+        assert self.world == 1, "multi-rank capture not enabled yet"
+        if self.optimizer.hyper is None:
+            # device-state mode: per-step scalars live on device so the
+            # captured kernels read them indirectly (lr updated by a
+            # 4-byte write OUTSIDE the graph)
+            self.optimizer.hyper = torch.tensor(
+                [self.optimizer.lr, 1, 1, 1, 1, 1, 0, 0],
+                dtype=torch.float32, device=self.device)
+        self._g_pix = batch["pixel_values"].to(self.device).clone()
+        if cfg.channels_last:
+            self._g_pix = self._g_pix.contiguous(
+                memory_format=torch.channels_last)
+        self._g_ids = batch["input_ids"].to(self.device).clone()
+
+        self.optimizer.zero_grad()
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            for _ in range(3):
+                self._capture_body()
+                self.optimizer.zero_grad()
+        torch.cuda.current_stream().wait_stream(side)
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            self._g_loss = self._capture_body()
+        self._graph = g
+
+    def _capture_body(self) -> torch.Tensor:
+        cfg = self.cfg
+        with torch.no_grad():
+            latents = self.vae.encode(
+                self._g_pix.to(self.weight_dtype)).latent_dist.sample()
+            latents = latents * self.vae.config.scaling_factor
+            noise = torch.randn_like(latents)
+            timesteps = torch.randint(
+                0, self.noise_scheduler.num_train_timesteps,
+                (latents.shape[0],), device=self.device, dtype=torch.long)
+            noisy_latents = self.noise_scheduler.add_noise(
+                latents, noise, timesteps)
+            encoder_hidden_states = self.text_encoder(self._g_ids)[0]
+        if cfg.rand_noise_lam > 0:
+            encoder_hidden_states = encoder_hidden_states + \
+                cfg.rand_noise_lam * torch.randn_like(encoder_hidden_states)
+        model_pred = self.unet(noisy_latents, timesteps, encoder_hidden_states)
+        if self.noise_scheduler.prediction_type == "epsilon":
+            target = noise
+        else:
+            target = self.noise_scheduler.get_velocity(latents, noise, timesteps)
+        loss = F.mse_loss(model_pred.float(), target.float(), reduction="mean")
+        loss.backward()
+        self.optimizer.gather_grads()
+        self.optimizer.step_dev()  # clip fused; scalars from hyper[]
+        return loss.detach()
+
+    def _graph_replay(self, batch) -> torch.Tensor:
+        cfg = self.cfg
+        pix = batch["pixel_values"].to(self.device, non_blocking=True)
+        if cfg.channels_last:
+            pix = pix.contiguous(memory_format=torch.channels_last)
+        self._g_pix.copy_(pix, non_blocking=True)
+        self._g_ids.copy_(batch["input_ids"].to(self.device), non_blocking=True)
+        lr_step = get_lr(cfg, self.global_step, self.world)
+        if lr_step != self.optimizer.lr:
+            self.optimizer.lr = lr_step
+            self.optimizer.hyper[0].fill_(lr_step)
+        self._graph.replay()
+        self.optimizer.step_count += 1
+        self.global_step += 1
+        return self._g_loss
+
     def train_step(self, batch, sync_gradients: bool = True) -> torch.Tensor:
         """One micro-step; returns the (detached) loss."""
+        if self._graph is not None and sync_gradients:
+            return self._graph_replay(batch)
         cfg = self.cfg
         device_type = self.device.type
         autocast_on = self.weight_dtype != torch.float32 and not self.pure_bf16
