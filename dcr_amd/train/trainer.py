@@ -253,7 +253,11 @@ class Trainer:
                 self.optimizer.zero_grad()
         torch.cuda.current_stream().wait_stream(side)
         g = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(g):
+        # thread_local: only THIS thread's HIP calls are captured —
+        # background threads (pinned-memory, profiler watchdogs) would
+        # otherwise trip hipErrorStreamCaptureUnsupported under the
+        # default global mode
+        with torch.cuda.graph(g, capture_error_mode="thread_local"):
             self._g_loss = self._capture_body()
         self._graph = g
 
