@@ -291,8 +291,11 @@ def add_noise(
     m = ext()
     if use_hip(x0) and m is not None:
         count_dispatch('add_noise')
-        return m.add_noise(x0.contiguous(), noise.contiguous(),
-                           alphas_cumprod.to(x0.device, torch.float32), t.contiguous())
+        ac = alphas_cumprod if (alphas_cumprod.device == x0.device and
+                                alphas_cumprod.dtype == torch.float32) \
+            else alphas_cumprod.to(x0.device, torch.float32)
+        return m.add_noise(x0.contiguous(), noise.contiguous(), ac,
+                           t.contiguous())
     sa, sb = _gather_sqrt(alphas_cumprod.to(x0.device), t, x0.dim())
     return (sa * x0.float() + sb * noise.float()).to(x0.dtype)
 
@@ -305,8 +308,11 @@ def get_velocity(
 
     m = ext()
     if use_hip(x0) and m is not None:
-        return m.get_velocity(x0.contiguous(), noise.contiguous(),
-                              alphas_cumprod.to(x0.device, torch.float32), t.contiguous())
+        ac = alphas_cumprod if (alphas_cumprod.device == x0.device and
+                                alphas_cumprod.dtype == torch.float32) \
+            else alphas_cumprod.to(x0.device, torch.float32)
+        return m.get_velocity(x0.contiguous(), noise.contiguous(), ac,
+                              t.contiguous())
     sa, sb = _gather_sqrt(alphas_cumprod.to(x0.device), t, x0.dim())
     return (sa * noise.float() - sb * x0.float()).to(x0.dtype)
 

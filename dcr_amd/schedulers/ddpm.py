@@ -59,13 +59,23 @@ class DDPMScheduler:
         self.prediction_type = c.prediction_type
         self.init_noise_sigma = 1.0
 
+    def _ac_on(self, device) -> torch.Tensor:
+        # device-cached alphas_cumprod: a per-call pageable H2D copy is
+        # both wasteful and hipGraph-capture-illegal (r02c11)
+        cache = getattr(self, "_ac_cache", None)
+        if cache is None or cache.device != device:
+            self._ac_cache = self.alphas_cumprod.to(device, torch.float32)
+        return self._ac_cache
+
     def add_noise(self, original_samples: torch.Tensor, noise: torch.Tensor,
                   timesteps: torch.Tensor) -> torch.Tensor:
-        return ops.add_noise(original_samples, noise, self.alphas_cumprod, timesteps)
+        return ops.add_noise(original_samples, noise,
+                             self._ac_on(original_samples.device), timesteps)
 
     def get_velocity(self, sample: torch.Tensor, noise: torch.Tensor,
                      timesteps: torch.Tensor) -> torch.Tensor:
-        return ops.get_velocity(sample, noise, self.alphas_cumprod, timesteps)
+        return ops.get_velocity(sample, noise,
+                                self._ac_on(sample.device), timesteps)
 
     # ancestral DDPM sampling step (used mainly by tests; inference uses DDIM/DPM)
     def step(self, model_output: torch.Tensor, timestep: int, sample: torch.Tensor,
