@@ -97,11 +97,17 @@ def main():
     for _ in range(args.warmup):
         step()
 
-    # DCR_HIPGRAPH=1: capture the whole micro-step as one hipGraph after
-    # the MIOpen-find warmup; timed steps then replay it (1-GPU only)
-    if use_cuda and n_gpus == 1 and os.environ.get("DCR_HIPGRAPH") == "1":
-        trainer.enable_hipgraph(batch)
-        step()  # one replay outside the timed window
+    # Whole-step hipGraph capture after the MIOpen-find warmup; timed
+    # steps then replay it (1-GPU only; measured 217.2 vs 213-215 eager,
+    # r02c11). DCR_HIPGRAPH=0 opts out; any capture failure falls back
+    # to the eager step so the bench always completes.
+    if use_cuda and n_gpus == 1 and os.environ.get("DCR_HIPGRAPH", "1") != "0":
+        try:
+            trainer.enable_hipgraph(batch)
+            step()  # one replay outside the timed window
+        except Exception as e:  # noqa: BLE001
+            print(f"hipGraph capture unavailable ({type(e).__name__}: {e}); "
+                  f"benching the eager step", file=sys.stderr)
 
     if dist_utils.is_dist():
         dist_utils.barrier()
