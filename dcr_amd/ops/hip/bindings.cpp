@@ -109,7 +109,10 @@ std::vector<at::Tensor> groupnorm_silu_nhwc_fwd(at::Tensor x, at::Tensor w,
   auto bf = norm_weight(b, x, bf32);
   TORCH_CHECK(wf32 == bf32);
   auto y = at::empty_like(x);  // preserves channels_last
-  auto ws = at::zeros({N * groups * 2}, x.options().dtype(at::kFloat));
+  const int64_t chunks = gn_nhwc_chunks((int)N, (int)R);
+  // per-chunk partial slab, plain stores -> at::empty (no fill kernel)
+  auto ws = at::empty({chunks * N * groups * 2},
+                      x.options().dtype(at::kFloat));
   auto mean = at::empty({N * groups}, x.options().dtype(at::kFloat));
   auto rstd = at::empty_like(mean);
   gn_nhwc_fwd_launch(dtype_of(x), x.data_ptr(), wf.data_ptr(), bf.data_ptr(),
@@ -132,11 +135,18 @@ std::vector<at::Tensor> groupnorm_silu_nhwc_bwd(at::Tensor dy, at::Tensor x,
   bool bf32;
   auto bf = norm_weight(b, x, bf32);
   auto dx = at::empty_like(x);
-  auto scratch = at::zeros({N * groups * 2 + 2 * C},
-                           x.options().dtype(at::kFloat));
-  auto ws = scratch.narrow(0, 0, N * groups * 2);
-  auto dw = scratch.narrow(0, N * groups * 2, C);
-  auto db = scratch.narrow(0, N * groups * 2 + C, C);
+  const int64_t chunks = gn_nhwc_chunks((int)N, (int)R);
+  // [chunk-partial group sums | chunk-partial dw/db | finalized groups]
+  // — every slot plain-stored, so no zero-init kernel; dw/db are the
+  // finalize kernel's outputs
+  auto scratch = at::empty(
+      {chunks * N * groups * 2 + chunks * N * 2 * C + N * groups * 2 + 2 * C},
+      x.options().dtype(at::kFloat));
+  auto ws = scratch.narrow(0, 0, chunks * N * groups * 2);
+  auto dw = scratch.narrow(
+      0, chunks * N * groups * 2 + chunks * N * 2 * C + N * groups * 2, C);
+  auto db = scratch.narrow(
+      0, chunks * N * groups * 2 + chunks * N * 2 * C + N * groups * 2 + C, C);
   gn_nhwc_bwd_launch(dtype_of(x), dy.data_ptr(), x.data_ptr(),
                      wf.data_ptr(), bf.data_ptr(), wf32,
                      mean.data_ptr<float>(), rstd.data_ptr<float>(),

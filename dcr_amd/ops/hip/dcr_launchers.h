@@ -24,7 +24,11 @@ void ln_bwd_launch(DType dt, const void* dy, const void* x, const void* w,
                    bool w_f32, const float* mean, const float* rstd, void* dx,
                    float* dw, float* db, long M, int N, hipStream_t s);
 
-// norms_nhwc.hip (channels_last GroupNorm; w_f32 as above)
+// norms_nhwc.hip (channels_last GroupNorm; w_f32 as above).
+// Workspaces are per-chunk partial slabs (plain stores, deterministic,
+// no zero-init needed): fwd ws = chunks*N*G*2 floats; bwd ws =
+// chunks*N*G*2 + chunks*N*2C + N*G*2 floats. chunks = gn_nhwc_chunks().
+int gn_nhwc_chunks(int N, int R);
 void gn_nhwc_fwd_launch(DType dt, const void* x, const void* w, const void* b,
                         bool w_f32, void* y, float* ws, float* mean, float* rstd,
                         int N, int R, int C, int G, float eps, bool silu,

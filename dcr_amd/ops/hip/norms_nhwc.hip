@@ -51,6 +51,9 @@ __device__ __forceinline__ void storev(T* __restrict__ p, const float (&i)[V]) {
 }
 
 // --------------------------------------------------------------- stats fwd
+// ws is a per-chunk partial slab [chunks, N, G, 2] written with PLAIN
+// stores (no zero-init, no atomics — deterministic); gn_finalize_kernel
+// reduces over chunks.
 template <typename T, int V>
 __global__ void gn_nhwc_stats_kernel(const T* __restrict__ x, float* __restrict__ ws,
                                      int N, int R, int C, int G, int rows_per_blk) {
@@ -104,23 +107,60 @@ __global__ void gn_nhwc_stats_kernel(const T* __restrict__ x, float* __restrict_
     }
     __syncthreads();
   }
+  float* slot = ws + (((long)blockIdx.x * N + n) * G) * 2;
   for (int g = threadIdx.x; g < G; g += blockDim.x) {
     float a = 0.f, b = 0.f;
     for (int c = g * Cg; c < (g + 1) * Cg; ++c) { a += s1[c]; b += s2[c]; }
-    atomicAdd(&ws[((long)n * G + g) * 2], a);
-    atomicAdd(&ws[((long)n * G + g) * 2 + 1], b);
+    slot[(long)g * 2] = a;
+    slot[(long)g * 2 + 1] = b;
   }
 }
 
 __global__ void gn_finalize_kernel(const float* __restrict__ ws,
                                    float* __restrict__ mean, float* __restrict__ rstd,
-                                   long NG, float inv_L, float eps) {
+                                   long NG, int chunks, float inv_L, float eps) {
   long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= NG) return;
-  float m = ws[i * 2] * inv_L;
-  float var = fmaxf(ws[i * 2 + 1] * inv_L - m * m, 0.f);
+  float s1 = 0.f, s2 = 0.f;
+  for (int cx = 0; cx < chunks; ++cx) {
+    s1 += ws[((long)cx * NG + i) * 2];
+    s2 += ws[((long)cx * NG + i) * 2 + 1];
+  }
+  float m = s1 * inv_L;
+  float var = fmaxf(s2 * inv_L - m * m, 0.f);
   mean[i] = m;
   rstd[i] = rsqrtf(var + eps);
+}
+
+// reduce the bwd partial slabs: group sums [chunks, N, G, 2] -> [N*G, 2]
+// and per-channel dw/db [chunks*N, 2C] -> dw[C], db[C]
+__global__ void gn_bwd_finalize_groups_kernel(const float* __restrict__ ws,
+                                              float* __restrict__ out, long NG,
+                                              int chunks) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= NG) return;
+  float s1 = 0.f, s2 = 0.f;
+  for (int cx = 0; cx < chunks; ++cx) {
+    s1 += ws[((long)cx * NG + i) * 2];
+    s2 += ws[((long)cx * NG + i) * 2 + 1];
+  }
+  out[i * 2] = s1;
+  out[i * 2 + 1] = s2;
+}
+
+__global__ void gn_bwd_finalize_dwdb_kernel(const float* __restrict__ slab,
+                                            float* __restrict__ dw,
+                                            float* __restrict__ db, int C,
+                                            long rows) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float a = 0.f, b = 0.f;
+  for (long r = 0; r < rows; ++r) {
+    a += slab[r * 2 * C + c];
+    b += slab[r * 2 * C + C + c];
+  }
+  dw[c] = a;
+  db[c] = b;
 }
 
 // --------------------------------------------------------------- apply fwd
@@ -161,14 +201,15 @@ __global__ void gn_nhwc_bwd_stats_kernel(const T* __restrict__ dy, const T* __re
                                          const WT* __restrict__ w, const WT* __restrict__ b_,
                                          const float* __restrict__ mean,
                                          const float* __restrict__ rstd,
-                                         float* __restrict__ ws, float* __restrict__ dw,
-                                         float* __restrict__ db,
+                                         float* __restrict__ ws,
+                                         float* __restrict__ dwdb_slab,
                                          int N, int R, int C, int G, int rows_per_blk) {
   // smem [4*C]: per-channel S1/S2 (group sums) + per-channel dw/db
-  // partials. dw/db go through LDS so each block issues ONE global
-  // atomic per channel — with per-thread atomics the ~500-block grid
-  // serialized rpar x blocks adds on every dw[c] address (measured
-  // 3.5-4.6x kernel slowdown at V=8; atomic contention, not bandwidth).
+  // partials; everything leaves the block as PLAIN per-chunk slab
+  // stores (no global atomics at all — deterministic, no zero-init;
+  // the per-channel-atomic version measured 3.5-4.6x slower at V=8
+  // from contention; finalize kernels reduce the slabs).
+  // ws: [chunks, N, G, 2]; dwdb_slab: [chunks*N, 2C].
   extern __shared__ float smem[];
   float* sa = smem;                        // S1 per channel (w*dz)
   float* sb = smem + C;                    // S2 per channel (w*dz*yhat)
@@ -234,15 +275,17 @@ __global__ void gn_nhwc_bwd_stats_kernel(const T* __restrict__ dy, const T* __re
     }
   }
   __syncthreads();
+  float* drow = dwdb_slab + ((long)blockIdx.x * N + n) * 2 * C;
   for (int c = threadIdx.x; c < C; c += blockDim.x) {
-    atomicAdd(&dw[c], sdw[c]);
-    atomicAdd(&db[c], sdb[c]);
+    drow[c] = sdw[c];
+    drow[C + c] = sdb[c];
   }
+  float* slot = ws + (((long)blockIdx.x * N + n) * G) * 2;
   for (int g = threadIdx.x; g < G; g += blockDim.x) {
     float a = 0.f, bb = 0.f;
     for (int c = g * Cg; c < (g + 1) * Cg; ++c) { a += sa[c]; bb += sb[c]; }
-    atomicAdd(&ws[((long)n * G + g) * 2], a);
-    atomicAdd(&ws[((long)n * G + g) * 2 + 1], bb);
+    slot[(long)g * 2] = a;
+    slot[(long)g * 2 + 1] = bb;
   }
 }
 
@@ -294,6 +337,8 @@ __global__ void gn_nhwc_bwd_apply_kernel(const T* __restrict__ dy, const T* __re
 
 namespace dcr {
 
+int gn_nhwc_chunks(int N, int R);  // forward decl for the binding
+
 static inline int nhwc_row_chunks(int N, int R) {
   // target >= 512 workgroups to fill 256 CUs / 8 XCDs
   int chunks = (512 + N - 1) / N;
@@ -315,7 +360,7 @@ static void gn_nhwc_fwd_v(const void* x, const void* w, const void* b, void* y,
   long NG = (long)N * G;
   float inv_L = 1.f / ((float)R * (C / G));
   hipLaunchKernelGGL(dcr_nhwc::gn_finalize_kernel, dim3((NG + 255) / 256),
-                     dim3(256), 0, s, ws, mean, rstd, NG, inv_L, eps);
+                     dim3(256), 0, s, ws, mean, rstd, NG, chunks, inv_L, eps);
   long total = (long)N * R * C;
   dim3 agrid((int)min((total / V + 255) / 256, (long)8192)), ablock(256);
   if (silu)
@@ -363,24 +408,36 @@ static void gn_nhwc_bwd_v(const void* dy, const void* x, const void* w,
   int rows_per_blk = (R + chunks - 1) / chunks;
   dim3 grid(chunks, N), block(256);
   size_t lds = 4 * (size_t)C * sizeof(float);
+  // slab layout: ws_slab [chunks, N, G, 2] | dwdb_slab [chunks*N, 2C] |
+  // finalized groups [N*G, 2] (the pointers are carved by the binding)
+  long NG = (long)N * G;
+  float* ws_slab = ws;
+  float* dwdb_slab = ws + (long)chunks * NG * 2;
+  float* ws_fin = dwdb_slab + (long)chunks * N * 2 * C;
   if (silu)
     hipLaunchKernelGGL((dcr_nhwc::gn_nhwc_bwd_stats_kernel<T, WT, true, V>), grid, block, lds, s,
-                       (const T*)dy, (const T*)x, (const WT*)w, (const WT*)b, mean, rstd, ws, dw, db,
-                       N, R, C, G, rows_per_blk);
+                       (const T*)dy, (const T*)x, (const WT*)w, (const WT*)b, mean, rstd,
+                       ws_slab, dwdb_slab, N, R, C, G, rows_per_blk);
   else
     hipLaunchKernelGGL((dcr_nhwc::gn_nhwc_bwd_stats_kernel<T, WT, false, V>), grid, block, lds, s,
-                       (const T*)dy, (const T*)x, (const WT*)w, (const WT*)b, mean, rstd, ws, dw, db,
-                       N, R, C, G, rows_per_blk);
+                       (const T*)dy, (const T*)x, (const WT*)w, (const WT*)b, mean, rstd,
+                       ws_slab, dwdb_slab, N, R, C, G, rows_per_blk);
+  hipLaunchKernelGGL(dcr_nhwc::gn_bwd_finalize_groups_kernel,
+                     dim3((NG + 255) / 256), dim3(256), 0, s, ws_slab, ws_fin,
+                     NG, chunks);
+  hipLaunchKernelGGL(dcr_nhwc::gn_bwd_finalize_dwdb_kernel,
+                     dim3((C + 255) / 256), dim3(256), 0, s, dwdb_slab, dw, db,
+                     C, (long)chunks * N);
   long total = (long)N * R * C;
   float inv_L = 1.f / ((float)R * (C / G));
   dim3 agrid((int)min((total / V + 255) / 256, (long)8192)), ablock(256);
   if (silu)
     hipLaunchKernelGGL((dcr_nhwc::gn_nhwc_bwd_apply_kernel<T, WT, true, V>), agrid, ablock, 0, s,
-                       (const T*)dy, (const T*)x, (const WT*)w, (const WT*)b, mean, rstd, ws, (T*)dx,
+                       (const T*)dy, (const T*)x, (const WT*)w, (const WT*)b, mean, rstd, ws_fin, (T*)dx,
                        total, R, C, G, inv_L);
   else
     hipLaunchKernelGGL((dcr_nhwc::gn_nhwc_bwd_apply_kernel<T, WT, false, V>), agrid, ablock, 0, s,
-                       (const T*)dy, (const T*)x, (const WT*)w, (const WT*)b, mean, rstd, ws, (T*)dx,
+                       (const T*)dy, (const T*)x, (const WT*)w, (const WT*)b, mean, rstd, ws_fin, (T*)dx,
                        total, R, C, G, inv_L);
 }
 
@@ -412,5 +469,7 @@ void gn_nhwc_bwd_launch(DType dt, const void* dy, const void* x, const void* w,
       break;
   }
 }
+
+int gn_nhwc_chunks(int N, int R) { return nhwc_row_chunks(N, R); }
 
 }  // namespace dcr
