@@ -553,85 +553,6 @@ at::Tensor conv2d_nhwc_fwd_v2(at::Tensor x, at::Tensor w,
   return conv2d_nhwc_fwd_v2_impl(x, w, bias, stride, pad, res, temb);
 }
 
-// v4 draft (DCR_NATIVE_CONV_V4): glds-staged tiles; host pre-pads the
-// input (3x3) and the weight K dim so every DMA address is in bounds.
-at::Tensor conv2d_nhwc_fwd_v4(at::Tensor x, at::Tensor w,
-                              c10::optional<at::Tensor> bias, int64_t stride,
-                              int64_t pad,
-                              c10::optional<at::Tensor> res = c10::nullopt,
-                              c10::optional<at::Tensor> temb = c10::nullopt) {
-  TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16);
-  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast) &&
-              w.is_contiguous(at::MemoryFormat::ChannelsLast));
-  const int64_t Nb = x.size(0), C = x.size(1), Hin = x.size(2), Win = x.size(3);
-  const int64_t K = w.size(0), R = w.size(2), S = w.size(3);
-  TORCH_CHECK(w.size(1) == C && C % 64 == 0 && K % 64 == 0);
-  TORCH_CHECK((R == 3 && S == 3 && pad == 1 && (stride == 1 || stride == 2)) ||
-              (R == 1 && S == 1 && pad == 0 && stride == 1),
-              "conv v4: 3x3 p1 s1|2 or 1x1 p0 s1");
-  const int64_t P = (Hin + 2 * pad - R) / stride + 1;
-  const int64_t Q = (Win + 2 * pad - S) / stride + 1;
-  const int64_t NPQ = Nb * P * Q;
-  TORCH_CHECK(NPQ % 128 == 0, "conv v4: NPQ % 128 != 0");
-
-  at::Tensor xp = x;
-  int64_t Hp = Hin, Wp = Win;
-  if (R == 3) {
-    xp = at::constant_pad_nd(x, {0, 0, 1, 1, 1, 1}, 0)
-             .contiguous(at::MemoryFormat::ChannelsLast);
-    Hp += 2;
-    Wp += 2;
-  }
-  at::Tensor wp = w;
-  int64_t Kp = K;
-  if (K % 128 != 0) {
-    Kp = ((K + 127) / 128) * 128;
-    wp = at::constant_pad_nd(w, {0, 0, 0, 0, 0, 0, 0, Kp - K}, 0)
-             .contiguous(at::MemoryFormat::ChannelsLast);
-  }
-
-  auto y = at::empty({Nb, K, P, Q},
-                     x.options().memory_format(at::MemoryFormat::ChannelsLast));
-  at::Tensor bf;
-  const float* bp = nullptr;
-  if (bias.has_value()) {
-    bf = bias->to(at::kFloat).contiguous();
-    bp = bf.data_ptr<float>();
-  }
-  int64_t blocks = (NPQ / 128) * (Kp / 128);
-  int64_t nsteps = C * R * S / 64;
-  int splitz = 1;
-  if (blocks < 384) {
-    splitz = (int)std::min<int64_t>({(384 + blocks - 1) / blocks, nsteps, 16});
-    if (splitz < 1) splitz = 1;
-  }
-  at::Tensor wsb;
-  float* wsp = nullptr;
-  if (splitz > 1) {
-    wsb = at::zeros({NPQ * K}, x.options().dtype(at::kFloat));
-    wsp = wsb.data_ptr<float>();
-  }
-  const void* rp = nullptr;
-  const void* tp = nullptr;
-  if (res.has_value()) {
-    TORCH_CHECK(res->scalar_type() == at::kBFloat16 &&
-                res->is_contiguous(at::MemoryFormat::ChannelsLast) &&
-                res->sizes() == y.sizes());
-    rp = res->data_ptr();
-  }
-  if (temb.has_value()) {
-    TORCH_CHECK(temb->scalar_type() == at::kBFloat16 && temb->is_contiguous()
-                && temb->dim() == 2 && temb->size(0) == Nb &&
-                temb->size(1) == K);
-    tp = temb->data_ptr();
-  }
-  conv_nhwc_fwd_v4_launch(xp.data_ptr(), wp.data_ptr(), bp, y.data_ptr(), wsp,
-                          splitz, rp, tp, (int)Nb, (int)Hp, (int)Wp, (int)C,
-                          (int)K, (int)Kp, (int)P, (int)Q, (int)R, (int)S,
-                          (int)stride, cur_stream());
-  return y;
-}
-
 // conv backward drafts (round-2; validated before any dispatch)
 std::vector<at::Tensor> conv2d_nhwc_bwd(at::Tensor dy, at::Tensor x,
                                         at::Tensor w, int64_t stride,
@@ -717,11 +638,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("gemm_bf16", &gemm_bf16);
   mod.def("conv2d_nhwc_bwd", &conv2d_nhwc_bwd);
   mod.def("conv2d_nhwc_fwd", &conv2d_nhwc_fwd);
-  mod.def("conv2d_nhwc_fwd_v4", &conv2d_nhwc_fwd_v4,
-          pybind11::arg("x"), pybind11::arg("w"), pybind11::arg("bias"),
-          pybind11::arg("stride"), pybind11::arg("pad"),
-          pybind11::arg("res") = c10::nullopt,
-          pybind11::arg("temb") = c10::nullopt);
   mod.def("conv2d_nhwc_fwd_v2", &conv2d_nhwc_fwd_v2,
           pybind11::arg("x"), pybind11::arg("w"), pybind11::arg("bias"),
           pybind11::arg("stride"), pybind11::arg("pad"),

@@ -295,180 +295,7 @@ __global__ void conv_splitk_finalize_kernel(const float* __restrict__ ws,
 
 }  // namespace dcr_conv
 
-// ===========================================================================
-// v4 (round-2 draft, DCR_NATIVE_CONV_V4): v2's tile/MFMA discipline with
-// the staging replaced by direct global->LDS DMA (global_load_lds width
-// 16 — the guide's step-3 lever, +69% on the GEMM ladder). Requirements
-// glds imposes and how they are met:
-//   * no conditional zero-fill: the host pre-pads the input (3x3) and
-//     the weight K dim (to a 128 multiple), so every staging address is
-//     in bounds — no guards anywhere in the staging path;
-//   * lane-linear LDS destination: tiles are stored UNPADDED [128][64]
-//     (128 B rows) with a chunk-XOR swizzle ((row&7) on the 16 B chunk
-//     index) applied to the per-lane SOURCE address and to the fragment
-//     reads (guide rule 21 / T2) — <=2-way ds_read_b128 conflicts;
-//   * NPQ % 128 == 0 and C % 64 == 0 (every SD/VAE shape at these batch
-//     sizes; the dispatcher falls back to v2 otherwise).
-// ===========================================================================
-namespace dcr_conv {
-
-__global__ __launch_bounds__(256)
-void conv_nhwc_fwd_v4_kernel(const bf16_t* __restrict__ x, const bf16_t* __restrict__ w,
-                             const float* __restrict__ bias, bf16_t* __restrict__ y,
-                             float* __restrict__ ws, int splitz,
-                             const bf16_t* __restrict__ res,
-                             const bf16_t* __restrict__ temb,
-                             int Nb, int Hp, int Wp, int C, int K, int Kp,
-                             int P, int Q, int R, int S, int stride) {
-  constexpr int BK = 64;
-  __shared__ short sA[128 * BK];
-  __shared__ short sB[128 * BK];
-
-  const long m0 = (long)blockIdx.x * 128;
-  const int k0 = blockIdx.y * 128;
-  const long NPQ = (long)Nb * P * Q;
-  const int rsc_total = R * S * C;
-
-  const int lane = threadIdx.x & 63;
-  const int wid = threadIdx.x >> 6;
-  const int l16 = lane & 15;
-  const int kgrp = lane >> 4;
-  const int wr = (wid >> 1) * 64;
-  const int wc = (wid & 1) * 64;
-
-  f32x4_t acc[4][4];
-#pragma unroll
-  for (int i = 0; i < 4; ++i)
-#pragma unroll
-    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
-
-  // per-lane staging geometry (fixed across k-steps): 8 glds per thread
-  // -> wave i covers chunk ids (wid*4+g)*64 + lane of the 1024 16-byte
-  // chunks of each [128][64] tile
-  int arow[4], achk[4];
-  long apix[4];  // (n*Hp + p*stride)*Wp + q*stride  of this chunk's pixel
-  int bchk_off[4];
-  long brow_off[4];
-#pragma unroll
-  for (int g = 0; g < 4; ++g) {
-    const int cid = (wid * 4 + g) * 64 + lane;
-    const int row = cid >> 3;
-    const int chk = cid & 7;
-    arow[g] = row;
-    achk[g] = chk;
-    const long m = m0 + row;             // NPQ % 128 == 0: always valid
-    const int n = (int)(m / (P * Q));
-    const int pq = (int)(m % (P * Q));
-    apix[g] = ((long)n * Hp + (pq / Q) * stride) * Wp + (pq % Q) * stride;
-    brow_off[g] = (long)(k0 + row) * rsc_total;  // K padded: in bounds
-    bchk_off[g] = (chk ^ (row & 7)) * 8;
-  }
-
-  const int nsteps = rsc_total / BK;
-  const int spz = (nsteps + splitz - 1) / splitz;
-  const int step0 = blockIdx.z * spz;
-  const int step1 = min(nsteps, step0 + spz);
-
-  for (int step = step0; step < step1; ++step) {
-    const int rsc0 = step * BK;
-    const int tap = rsc0 / C;             // C % 64 == 0: tap slice-uniform
-    const int r = tap / S;
-    const int s = tap % S;
-    const int c0 = rsc0 - tap * C;
-
-    __syncthreads();
-#pragma unroll
-    for (int g = 0; g < 4; ++g) {
-      const bf16_t* asrc = x + (apix[g] + (long)r * Wp + s) * C + c0 +
-                           bchk_off[g];  // same swizzle map as B
-      __builtin_amdgcn_global_load_lds(
-          (const __attribute__((address_space(1))) void*)asrc,
-          (__attribute__((address_space(3))) void*)(sA + (long)(wid * 4 + g) * 512),
-          16, 0, 0);
-      const bf16_t* bsrc = w + brow_off[g] + rsc0 + bchk_off[g];
-      __builtin_amdgcn_global_load_lds(
-          (const __attribute__((address_space(1))) void*)bsrc,
-          (__attribute__((address_space(3))) void*)(sB + (long)(wid * 4 + g) * 512),
-          16, 0, 0);
-    }
-    __syncthreads();  // drains the DMA (vmcnt(0) folded into the barrier)
-
-#pragma unroll
-    for (int kk = 0; kk < BK / 32; ++kk) {
-      bf16x8 af[4], bf[4];
-#pragma unroll
-      for (int i = 0; i < 4; ++i) {
-        const int rw = wr + i * 16 + l16;
-        af[i] = *reinterpret_cast<const bf16x8*>(
-            sA + rw * BK + (((kk * 4 + kgrp) ^ (rw & 7)) * 8));
-      }
-#pragma unroll
-      for (int j = 0; j < 4; ++j) {
-        const int rw = wc + j * 16 + l16;
-        bf[j] = *reinterpret_cast<const bf16x8*>(
-            sB + rw * BK + (((kk * 4 + kgrp) ^ (rw & 7)) * 8));
-      }
-#pragma unroll
-      for (int i = 0; i < 4; ++i)
-#pragma unroll
-        for (int j = 0; j < 4; ++j)
-          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[i], bf[j],
-                                                              acc[i][j], 0, 0, 0);
-    }
-  }
-
-  const int PQ = P * Q;
-#pragma unroll
-  for (int i = 0; i < 4; ++i) {
-#pragma unroll
-    for (int rr = 0; rr < 4; ++rr) {
-      const long m = m0 + wr + i * 16 + kgrp * 4 + rr;
-#pragma unroll
-      for (int j = 0; j < 4; ++j) {
-        const int k = k0 + wc + j * 16 + l16;
-        if (k >= K) continue;            // K was padded to Kp
-        if (splitz > 1) {
-          atomicAdd(&ws[m * K + k], acc[i][j][rr]);
-        } else {
-          float v = acc[i][j][rr] + (bias ? bias[k] : 0.f);
-          if (res) v += __bfloat162float(res[m * K + k]);
-          if (temb) v += __bfloat162float(temb[(m / PQ) * K + k]);
-          y[m * K + k] = __float2bfloat16(v);
-        }
-      }
-    }
-  }
-}
-
-}  // namespace dcr_conv
-
 namespace dcr {
-
-void conv_nhwc_fwd_v4_launch(const void* x, const void* w, const float* bias,
-                             void* y, float* ws, int splitz, const void* res,
-                             const void* temb, int Nb, int Hp, int Wp, int C,
-                             int K, int Kp, int P, int Q, int R, int S,
-                             int stride, hipStream_t st) {
-  long NPQ = (long)Nb * P * Q;
-  dim3 grid((unsigned)(NPQ / 128), (unsigned)(Kp / 128), (unsigned)splitz),
-      block(256);
-  hipLaunchKernelGGL(dcr_conv::conv_nhwc_fwd_v4_kernel, grid, block, 0, st,
-                     (const dcr_conv::bf16_t*)x, (const dcr_conv::bf16_t*)w,
-                     bias, (dcr_conv::bf16_t*)y, ws, splitz,
-                     (const dcr_conv::bf16_t*)res,
-                     (const dcr_conv::bf16_t*)temb, Nb, Hp, Wp, C, K, Kp, P,
-                     Q, R, S, stride);
-  if (splitz > 1) {
-    long total = NPQ * K;
-    long b = (total / 4 + 255) / 256;
-    if (b > 8192) b = 8192;
-    hipLaunchKernelGGL(dcr_conv::conv_splitk_finalize_kernel,
-                       dim3((unsigned)b), dim3(256), 0, st, ws, bias,
-                       (const dcr_conv::bf16_t*)res,
-                       (const dcr_conv::bf16_t*)temb, (dcr_conv::bf16_t*)y,
-                       total, K, P * Q);
-  }
-}
 
 void conv_nhwc_fwd_v2_launch(const void* x, const void* w, const float* bias,
                              void* y, float* ws, int splitz, const void* res,

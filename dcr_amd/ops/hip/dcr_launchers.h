@@ -101,12 +101,6 @@ void conv_nhwc_fwd_v2_launch(const void* x, const void* w, const float* bias,
                              const void* temb, int Nb, int Hin, int Win, int C,
                              int K, int P, int Q, int R, int S, int stride,
                              int pad, hipStream_t st);
-// v4: glds-staged, swizzled, pre-padded input (DCR_NATIVE_CONV_V4 draft)
-void conv_nhwc_fwd_v4_launch(const void* x, const void* w, const float* bias,
-                             void* y, float* ws, int splitz, const void* res,
-                             const void* temb, int Nb, int Hp, int Wp, int C,
-                             int K, int Kp, int P, int Q, int R, int S,
-                             int stride, hipStream_t st);
 // conv_nhwc_bwd.hip (dgrad + wgrad + fused bias-grad; DCR_NATIVE_CONV_BWD)
 void conv_bwd_weight_launch(const void* dy, const void* x, float* dw_ws,
                             int Nb, int Hin, int Win, int C, int K, int P,

@@ -45,16 +45,14 @@ def main():
         bb = b.to(torch.bfloat16)
         t_v1 = timeit(lambda: m.conv2d_nhwc_fwd(x, w, b, stride, pad))
         t_v2 = timeit(lambda: m.conv2d_nhwc_fwd_v2(x, w, b, stride, pad))
-        t_v4 = timeit(lambda: m.conv2d_nhwc_fwd_v4(x, w, b, stride, pad))
         t_miopen = timeit(lambda: F.conv2d(x, w, bb, stride=stride, padding=pad))
         P = (H + 2 * pad - R) // stride + 1
         flops = 2 * N * P * P * K * C * R * R
         print(f"N{N} C{C} H{H} K{K} R{R}s{stride}: "
               f"v1 {t_v1:.3f} ms ({flops / t_v1 / 1e9:.0f} TF) "
               f"v2 {t_v2:.3f} ms ({flops / t_v2 / 1e9:.0f} TF) "
-              f"v4 {t_v4:.3f} ms ({flops / t_v4 / 1e9:.0f} TF) "
               f"MIOpen {t_miopen:.3f} ms ({flops / t_miopen / 1e9:.0f} TF) "
-              f"-> v2/MIOpen {t_miopen / t_v2:.2f}x v4/v2 {t_v2 / t_v4:.2f}x")
+              f"-> v2/MIOpen {t_miopen / t_v2:.2f}x")
 
         # backward: native (dcr conv_nhwc_bwd: dgrad + wgrad + fused
         # bias-grad) vs MIOpen via aten::convolution_backward (+ the

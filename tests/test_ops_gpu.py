@@ -745,33 +745,3 @@ def test_attn_fwd_v4_matches_ref(ext, Lq, Lk, causal):
     _close(o4, ref, 2e-2)
     o1, l1 = ext.attn_fwd(q, k, v, scale, causal)
     assert (l4 - l1).abs().max().item() < 1e-4
-
-
-@pytest.mark.parametrize("shape", [
-    (16, 320, 32, 320, 3, 1),   # res32 3x3 (ragged K pad to 384)
-    (16, 640, 16, 640, 3, 1),
-    (16, 1280, 8, 1280, 3, 1),
-    (16, 320, 32, 640, 1, 1),   # 1x1
-    (16, 640, 32, 640, 3, 2),   # downsample
-    (16, 128, 64, 128, 3, 1),
-])
-def test_conv_nhwc_fwd_v4(ext, shape):
-    """v4 glds-staged conv (padded input, swizzled tiles) vs fp32 torch,
-    incl. the fused res/temb epilogue."""
-    (N, C, H, K, R, stride) = shape
-    pad = 1 if R == 3 else 0
-    torch.manual_seed(C + K)
-    x = torch.randn(N, C, H, H, device="cuda").bfloat16() \
-        .to(memory_format=torch.channels_last)
-    w = (torch.randn(K, C, R, R, device="cuda") * 0.05).bfloat16() \
-        .to(memory_format=torch.channels_last)
-    b = torch.randn(K, device="cuda")
-    y = ext.conv2d_nhwc_fwd_v4(x, w, b, stride, pad)
-    ref = F.conv2d(x.float(), w.float(), b, stride=stride, padding=pad)
-    _close(y, ref, 2e-2)
-    P = ref.shape[2]
-    res = torch.randn(N, K, P, P, device="cuda").bfloat16() \
-        .to(memory_format=torch.channels_last)
-    temb = torch.randn(N, K, device="cuda").bfloat16()
-    y2 = ext.conv2d_nhwc_fwd_v4(x, w, b, stride, pad, res, temb)
-    _close(y2, ref + res.float() + temb.float()[:, :, None, None], 2e-2)
