@@ -52,15 +52,15 @@ def main():
         vp = v.permute(0, 2, 1, 3).contiguous()
 
         t_ours = timeit(lambda: m.attn_fwd(q, k, v, scale, causal))
-        t_v3 = timeit(lambda: m.attn_fwd_v4(q, k, v, scale, causal))
+        t_v4 = timeit(lambda: m.attn_fwd_v4(q, k, v, scale, causal))
         t_sdpa = timeit(lambda: F.scaled_dot_product_attention(
             qp, kp, vp, is_causal=causal, scale=scale))
         flops = 4 * B * H * Lq * Lk * D  # fwd QK^T + PV
         print(f"B{B} H{H} Lq{Lq} Lk{Lk} causal={int(causal)}: "
               f"ours {t_ours:.3f} ms ({flops / t_ours / 1e9:.0f} TF) "
-              f"v4 {t_v3:.3f} ms ({flops / t_v3 / 1e9:.0f} TF) "
+              f"v4 {t_v4:.3f} ms ({flops / t_v4 / 1e9:.0f} TF) "
               f"sdpa {t_sdpa:.3f} ms ({flops / t_sdpa / 1e9:.0f} TF) "
-              f"-> {t_sdpa / t_ours:.2f}x, v4/v1 {t_ours / t_v3:.2f}x")
+              f"-> {t_sdpa / t_ours:.2f}x, v4/v1 {t_ours / t_v4:.2f}x")
 
         # backward A/B
         def ours_bwd():
