@@ -279,3 +279,25 @@ def test_clip_bpe_matches_transformers(tmp_path):
     re_tok = load_tokenizer(tmp_path / "rt")
     assert type(re_tok).__name__ == "CLIPBPETokenizer"
     assert re_tok.encode_words(samples[0]) == ours.encode_words(samples[0])
+
+
+def test_clip_bpe_decode_roundtrip(tmp_path):
+    """decode() inverts encode for in-vocab text (strips specials and
+    end-of-word markers)."""
+    import json
+    from dcr_amd.data.tokenizer import CLIPBPETokenizer, _bytes_to_unicode
+
+    b2u = _bytes_to_unicode()
+    vocab = {}
+    for c in b2u.values():
+        vocab[c] = len(vocab)
+    for c in b2u.values():
+        vocab[c + "</w>"] = len(vocab)
+    vocab["<|startoftext|>"] = len(vocab)
+    vocab["<|endoftext|>"] = len(vocab)
+    (tmp_path / "vocab.json").write_text(json.dumps(vocab))
+    (tmp_path / "merges.txt").write_text("#version: 0.2\n")
+    tok = CLIPBPETokenizer(tmp_path / "vocab.json", tmp_path / "merges.txt")
+    text = "a photo of a cat"
+    ids = tok(text, max_length=32).input_ids[0].tolist()
+    assert tok.decode(ids) == text
