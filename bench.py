@@ -165,3 +165,9 @@ def main():
 
 if __name__ == "__main__":
     main()
+    # explicit teardown: a rank exiting with the group alive can race the
+    # store shutdown under torchrun and turn a finished bench into a
+    # non-zero exit
+    import torch.distributed as _dist
+    if _dist.is_available() and _dist.is_initialized():
+        _dist.destroy_process_group()
