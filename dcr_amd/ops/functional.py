@@ -197,8 +197,11 @@ class _FlashAttention(torch.autograd.Function):
     def backward(ctx, dO):
         q, k, v, o, lse = ctx.saved_tensors
         m = require_hip("attn")
-        dQ, dK, dV = m.attn_bwd(q, k, v, o, dO.contiguous(), lse,
-                                ctx.scale, ctx.causal)
+        # DCR_ATTN_BWD_V4=1: swapped-operand backward draft (A/B only)
+        bwd = m.attn_bwd_v4 if os.environ.get("DCR_ATTN_BWD_V4") == "1" \
+            else m.attn_bwd
+        dQ, dK, dV = bwd(q, k, v, o, dO.contiguous(), lse,
+                         ctx.scale, ctx.causal)
         return dQ, dK, dV, None, None
 
 

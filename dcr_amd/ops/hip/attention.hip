@@ -1037,6 +1037,350 @@ void attn_fwd_v4_kernel(const bf16_t* __restrict__ q, const bf16_t* __restrict__
 }
 
 // ==========================================================================
+// Backward v4 (round-2 draft, DCR_ATTN_BWD_V4): the fwd-v4 schedule
+// applied to the FlashAttention-2 backward. Backward needs no online
+// softmax (P recomputes from the stored LSE), so both kernels keep the
+// whole P / dS state in registers with the swapped-operand layout and
+// the same pack+permlane fragment algebra fwd v4 verified:
+//   * dq_v4:   8 waves x 32 q-rows; lane owns ONE q (lse/delta are
+//     lane-local scalars); S^T/dP^T/dS^T all lane-local; dQ^T
+//     accumulates like fwd's O^T. No LDS round-trip for P or dS.
+//   * dkdv_v4: 8 waves x 32 keys; lane owns ONE key; per-q lse/delta
+//     come from an LDS stage; dV/dK accumulate over q tiles.
+// ==========================================================================
+
+// pack a lane-local 64-wide fp32 axis (two 32-subtile reg files, crow
+// layout) into the bf16 A/B fragment for contraction chunk c (16 wide):
+// words = (s0[0], s1[0], s0[1], s1[1]) — fwd-v4-verified algebra.
+__device__ __forceinline__ bf16x8 pack_frag64(const f32x16_t& t0,
+                                              const f32x16_t& t1, int c) {
+  const f32x16_t& t = (c >> 1) ? t1 : t0;
+  const int r8 = 8 * (c & 1);
+  unsigned u[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    __hip_bfloat16 lo = __float2bfloat16(t[r8 + 2 * i]);
+    __hip_bfloat16 hi_ = __float2bfloat16(t[r8 + 2 * i + 1]);
+    u[i] = (unsigned)*reinterpret_cast<unsigned short*>(&lo) |
+           ((unsigned)*reinterpret_cast<unsigned short*>(&hi_) << 16);
+  }
+  auto s0 = __builtin_amdgcn_permlane32_swap(u[0], u[2], false, false);
+  auto s1 = __builtin_amdgcn_permlane32_swap(u[1], u[3], false, false);
+  unsigned w[4] = {(unsigned)s0[0], (unsigned)s1[0],
+                   (unsigned)s0[1], (unsigned)s1[1]};
+  return *reinterpret_cast<const bf16x8*>(w);
+}
+
+__global__ __launch_bounds__(512)
+void attn_bwd_dq_v4_kernel(const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
+                           const bf16_t* __restrict__ v, const bf16_t* __restrict__ dO,
+                           const float* __restrict__ lse,
+                           const float* __restrict__ delta,
+                           bf16_t* __restrict__ dQ, int Lq, int Lk, int H,
+                           float scale, int causal) {
+  __shared__ short sK[TILE * PITCH];     // [key][dim]
+  __shared__ short sKT[TILE * PITCH];    // [dim][key]
+  __shared__ short sV[TILE * PITCH];     // [key][dim]
+
+  const int bh = blockIdx.y;
+  const int b_ = bh / H, h = bh % H;
+  const int q0 = blockIdx.x * 256;
+  const long rs = (long)H * DHEAD;
+  const bf16_t* kp = k + ((long)b_ * Lk * H + h) * DHEAD;
+  const bf16_t* vp = v + ((long)b_ * Lk * H + h) * DHEAD;
+
+  const int t = threadIdx.x;
+  const int wid = t >> 6;
+  const int lane = t & 63;
+  const int l32 = lane & 31;
+  const int hi = lane >> 5;
+  const int qrow = q0 + wid * 32 + l32;
+  const bool valid = qrow < Lq;
+
+  bf16x8 qf[4], of[4];
+  float lse_q = 1e30f, del_q = 0.f;
+  if (valid) {
+    const bf16_t* qp = q + (((long)b_ * Lq + qrow) * H + h) * DHEAD;
+    const bf16_t* dop = dO + (((long)b_ * Lq + qrow) * H + h) * DHEAD;
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      qf[c] = *reinterpret_cast<const bf16x8*>(qp + c * 16 + hi * 8);
+      of[c] = *reinterpret_cast<const bf16x8*>(dop + c * 16 + hi * 8);
+    }
+    lse_q = lse[(long)bh * Lq + qrow];
+    del_q = delta[(long)bh * Lq + qrow];
+  } else {
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      qf[c] = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+      of[c] = qf[c];
+    }
+  }
+
+  // staging: 512 threads, 8/row, one uint4 + 8-scalar transposed write
+  const int st_row = t >> 3;
+  const int st_col = (t & 7) * 8;
+
+  f32x16_t dqacc[2];
+#pragma unroll
+  for (int dt = 0; dt < 2; ++dt)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) dqacc[dt][r] = 0.f;
+
+  const int kv_end = causal ? min(Lk, q0 + 256) : Lk;
+
+  for (int kv0 = 0; kv0 < kv_end; kv0 += TILE) {
+    __syncthreads();
+    {
+      const int vld = Lk - kv0;
+      uint4 kv4 = make_uint4(0, 0, 0, 0), vv4 = kv4;
+      if (st_row < vld) {
+        kv4 = *reinterpret_cast<const uint4*>(
+            kp + (long)(kv0 + st_row) * rs + st_col);
+        vv4 = *reinterpret_cast<const uint4*>(
+            vp + (long)(kv0 + st_row) * rs + st_col);
+      }
+      *reinterpret_cast<uint4*>(sK + st_row * PITCH + st_col) = kv4;
+      *reinterpret_cast<uint4*>(sV + st_row * PITCH + st_col) = vv4;
+      const short* kk_ = reinterpret_cast<const short*>(&kv4);
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        sKT[(st_col + j) * PITCH + st_row] = kk_[j];
+    }
+    __syncthreads();
+
+    f32x16_t pt[2], dpt[2];
+#pragma unroll
+    for (int st = 0; st < 2; ++st) {
+      f32x16_t sacc, dacc;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) { sacc[r] = 0.f; dacc[r] = 0.f; }
+#pragma unroll
+      for (int c = 0; c < 4; ++c) {
+        bf16x8 kf = *reinterpret_cast<const bf16x8*>(
+            sK + (st * 32 + l32) * PITCH + c * 16 + hi * 8);
+        bf16x8 vf = *reinterpret_cast<const bf16x8*>(
+            sV + (st * 32 + l32) * PITCH + c * 16 + hi * 8);
+        sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[c], sacc, 0, 0, 0);
+        dacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, of[c], dacc, 0, 0, 0);
+      }
+      pt[st] = sacc;
+      dpt[st] = dacc;
+    }
+
+    // P^T from lse; dS^T = P^T (dP^T - delta) * scale — all lane-local
+#pragma unroll
+    for (int st = 0; st < 2; ++st)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int key = kv0 + st * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+        const bool masked = !valid || key >= Lk || (causal && key > qrow);
+        const float p = masked ? 0.f : __expf(pt[st][r] * scale - lse_q);
+        pt[st][r] = p * (dpt[st][r] - del_q) * scale;   // now dS^T
+      }
+
+    // dQ^T[d][q] += K^T dS^T
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      bf16x8 dsf = pack_frag64(pt[0], pt[1], c);
+#pragma unroll
+      for (int dt = 0; dt < 2; ++dt) {
+        bf16x8 ktf = *reinterpret_cast<const bf16x8*>(
+            sKT + (dt * 32 + l32) * PITCH + c * 16 + hi * 8);
+        dqacc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ktf, dsf,
+                                                            dqacc[dt], 0, 0, 0);
+      }
+    }
+  }
+
+  if (!valid) return;
+  bf16_t* dqp = dQ + (((long)b_ * Lq + qrow) * H + h) * DHEAD;
+#pragma unroll
+  for (int dt = 0; dt < 2; ++dt)
+#pragma unroll
+    for (int rq = 0; rq < 4; ++rq) {
+      const int d0 = dt * 32 + 8 * rq + 4 * hi;
+      ushort4 pk;
+#pragma unroll
+      for (int e = 0; e < 4; ++e) {
+        __hip_bfloat16 hv = __float2bfloat16(dqacc[dt][rq * 4 + e]);
+        (&pk.x)[e] = *reinterpret_cast<unsigned short*>(&hv);
+      }
+      *reinterpret_cast<uint2*>(dqp + d0) = *reinterpret_cast<uint2*>(&pk);
+    }
+}
+
+__global__ __launch_bounds__(512)
+void attn_bwd_dkdv_v4_kernel(const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
+                             const bf16_t* __restrict__ v, const bf16_t* __restrict__ dO,
+                             const float* __restrict__ lse,
+                             const float* __restrict__ delta,
+                             bf16_t* __restrict__ dK, bf16_t* __restrict__ dV,
+                             int Lq, int Lk, int H, float scale, int causal) {
+  __shared__ short sQ[TILE * PITCH];     // [qrow][dim]
+  __shared__ short sQT[TILE * PITCH];    // [dim][qrow]
+  __shared__ short sdO[TILE * PITCH];    // [qrow][dim]
+  __shared__ short sdOT[TILE * PITCH];   // [dim][qrow]
+  __shared__ float sLse[TILE];
+  __shared__ float sDelta[TILE];
+
+  const int bh = blockIdx.y;
+  const int b_ = bh / H, h = bh % H;
+  const int k0 = blockIdx.x * 256;
+  const long rs = (long)H * DHEAD;
+  const bf16_t* qp = q + ((long)b_ * Lq * H + h) * DHEAD;
+  const bf16_t* dop = dO + ((long)b_ * Lq * H + h) * DHEAD;
+  const float* lsep = lse + (long)bh * Lq;
+  const float* delp = delta + (long)bh * Lq;
+
+  const int t = threadIdx.x;
+  const int wid = t >> 6;
+  const int lane = t & 63;
+  const int l32 = lane & 31;
+  const int hi = lane >> 5;
+  const int key = k0 + wid * 32 + l32;
+  const bool kvalid = key < Lk;
+
+  bf16x8 kf[4], vf[4];
+  if (kvalid) {
+    const bf16_t* kp = k + (((long)b_ * Lk + key) * H + h) * DHEAD;
+    const bf16_t* vp = v + (((long)b_ * Lk + key) * H + h) * DHEAD;
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      kf[c] = *reinterpret_cast<const bf16x8*>(kp + c * 16 + hi * 8);
+      vf[c] = *reinterpret_cast<const bf16x8*>(vp + c * 16 + hi * 8);
+    }
+  } else {
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      kf[c] = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+      vf[c] = kf[c];
+    }
+  }
+
+  const int st_row = t >> 3;
+  const int st_col = (t & 7) * 8;
+
+  f32x16_t dkacc[2], dvacc[2];
+#pragma unroll
+  for (int dt = 0; dt < 2; ++dt)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) { dkacc[dt][r] = 0.f; dvacc[dt][r] = 0.f; }
+
+  const int q_start = causal ? (k0 & ~(TILE - 1)) : 0;
+
+  for (int q0 = q_start; q0 < Lq; q0 += TILE) {
+    __syncthreads();
+    {
+      const int vld = Lq - q0;
+      uint4 q4 = make_uint4(0, 0, 0, 0), o4 = q4;
+      if (st_row < vld) {
+        q4 = *reinterpret_cast<const uint4*>(
+            qp + (long)(q0 + st_row) * rs + st_col);
+        o4 = *reinterpret_cast<const uint4*>(
+            dop + (long)(q0 + st_row) * rs + st_col);
+      }
+      *reinterpret_cast<uint4*>(sQ + st_row * PITCH + st_col) = q4;
+      *reinterpret_cast<uint4*>(sdO + st_row * PITCH + st_col) = o4;
+      const short* qq_ = reinterpret_cast<const short*>(&q4);
+      const short* oo_ = reinterpret_cast<const short*>(&o4);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        sQT[(st_col + j) * PITCH + st_row] = qq_[j];
+        sdOT[(st_col + j) * PITCH + st_row] = oo_[j];
+      }
+      if (t < TILE) {
+        const int qr = q0 + t;
+        sLse[t] = (qr < Lq) ? lsep[qr] : 1e30f;
+        sDelta[t] = (qr < Lq) ? delp[qr] : 0.f;
+      }
+    }
+    __syncthreads();
+
+    // S'^T[key][q] and dP'^T[key][q] (lane = one key, q in regs)
+    f32x16_t pt[2], dpt[2];
+#pragma unroll
+    for (int st = 0; st < 2; ++st) {
+      f32x16_t sacc, dacc;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) { sacc[r] = 0.f; dacc[r] = 0.f; }
+#pragma unroll
+      for (int c = 0; c < 4; ++c) {
+        bf16x8 qfr = *reinterpret_cast<const bf16x8*>(
+            sQ + (st * 32 + l32) * PITCH + c * 16 + hi * 8);
+        bf16x8 ofr = *reinterpret_cast<const bf16x8*>(
+            sdO + (st * 32 + l32) * PITCH + c * 16 + hi * 8);
+        // A = Q rows (q), B = K lane frags -> C[q][key]: lane key-col?
+        // No: A rows become C rows; we want C[key][q], so A = K? K is
+        // in registers (kf) which are this lane's key only — use the
+        // swapped form: mfma(A=Q_row_frag? ) — instead compute
+        // C[q-subtile][key]^T by A=Q, B=kf: C[qrow][keycol]: lane holds
+        // key-col l32?? kf holds THIS lane's key rows... B-frag rows
+        // must be the output-col axis across lanes: kf[c] per lane IS
+        // row l&31 of the B operand when B = K tile — but kf was loaded
+        // for key = wid*32+l32, matching B rows = this wave's 32 keys.
+        sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qfr, kf[c], sacc, 0, 0, 0);
+        dacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ofr, vf[c], dacc, 0, 0, 0);
+      }
+      // C[q][key]: lane holds key-col l32 (ours), q rows in regs
+      pt[st] = sacc;
+      dpt[st] = dacc;
+    }
+
+#pragma unroll
+    for (int st = 0; st < 2; ++st)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int qr = q0 + st * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+        const int qloc = qr - q0;
+        const float lse_q = sLse[qloc];
+        const float del_q = sDelta[qloc];
+        const bool masked = !kvalid || qr >= Lq || (causal && key > qr);
+        const float p = masked ? 0.f : __expf(pt[st][r] * scale - lse_q);
+        pt[st][r] = p;                         // P'^T
+        dpt[st][r] = p * (dpt[st][r] - del_q) * scale;  // dS'^T
+      }
+
+    // dV[key][d] += P'^T dO ; dK[key][d] += dS'^T Q   (contraction q)
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      bf16x8 pf = pack_frag64(pt[0], pt[1], c);
+      bf16x8 dsf = pack_frag64(dpt[0], dpt[1], c);
+#pragma unroll
+      for (int dt = 0; dt < 2; ++dt) {
+        bf16x8 dotf = *reinterpret_cast<const bf16x8*>(
+            sdOT + (dt * 32 + l32) * PITCH + c * 16 + hi * 8);
+        bf16x8 qtf = *reinterpret_cast<const bf16x8*>(
+            sQT + (dt * 32 + l32) * PITCH + c * 16 + hi * 8);
+        dvacc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pf, dotf,
+                                                            dvacc[dt], 0, 0, 0);
+        dkacc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsf, qtf,
+                                                            dkacc[dt], 0, 0, 0);
+      }
+    }
+  }
+
+  // C[key][d]: lane holds d-col?? dvacc came from mfma(A=P'^T[key][q],
+  // B=dO^T[d][q]) -> C rows = key (A rows across lanes: l&31 = THIS
+  // wave's key block rows), cols = d: lane holds col d = l32, rows
+  // key = wkey + crow(r, hi). Store column-wise (scalar stores).
+  if (true) {
+    bf16_t* dkp = dK + ((long)b_ * Lk * H + h) * DHEAD;
+    bf16_t* dvp = dV + ((long)b_ * Lk * H + h) * DHEAD;
+#pragma unroll
+    for (int dt = 0; dt < 2; ++dt)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int krow = k0 + wid * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+        if (krow >= Lk) continue;
+        const int d = dt * 32 + l32;
+        dkp[(long)krow * rs + d] = __float2bfloat16(dkacc[dt][r]);
+        dvp[(long)krow * rs + d] = __float2bfloat16(dvacc[dt][r]);
+      }
+  }
+}
+
+// ==========================================================================
 // Generalized-head-dim forward (SD-1.4 / sd_mitigation parity: head_dim
 // 40/80/160 — /root/reference/sd_mitigation.py:46; inference-only, so no
 // backward). Same tile discipline as attn_fwd_kernel with the head dim
@@ -1307,6 +1651,31 @@ void attn_fwd_gen_launch(const void* q, const void* k, const void* v, void* o,
     default:
       break;  // binding guards D
   }
+}
+
+// backward v4 (swapped-operand schedule; DCR_ATTN_BWD_V4 draft)
+void attn_bwd_v4_launch(const void* q, const void* k, const void* v,
+                        const void* o, const void* dO, const float* lse,
+                        float* delta, void* dQ, void* dK, void* dV, int BH,
+                        int Lq, int Lk, int H, float scale, bool causal,
+                        hipStream_t s) {
+  long rows = (long)BH * Lq;
+  dim3 gd((rows + 255) / 256), bd(256);
+  hipLaunchKernelGGL(dcr_attn::attn_bwd_delta_kernel, gd, bd, 0, s,
+                     (const dcr_attn::bf16_t*)dO, (const dcr_attn::bf16_t*)o,
+                     delta, rows, Lq, H);
+  dim3 g1((Lk + 255) / 256, BH), b1(512);
+  hipLaunchKernelGGL(dcr_attn::attn_bwd_dkdv_v4_kernel, g1, b1, 0, s,
+                     (const dcr_attn::bf16_t*)q, (const dcr_attn::bf16_t*)k,
+                     (const dcr_attn::bf16_t*)v, (const dcr_attn::bf16_t*)dO,
+                     lse, delta, (dcr_attn::bf16_t*)dK, (dcr_attn::bf16_t*)dV,
+                     Lq, Lk, H, scale, causal ? 1 : 0);
+  dim3 g2((Lq + 255) / 256, BH), b2(512);
+  hipLaunchKernelGGL(dcr_attn::attn_bwd_dq_v4_kernel, g2, b2, 0, s,
+                     (const dcr_attn::bf16_t*)q, (const dcr_attn::bf16_t*)k,
+                     (const dcr_attn::bf16_t*)v, (const dcr_attn::bf16_t*)dO,
+                     lse, delta, (dcr_attn::bf16_t*)dQ, Lq, Lk, H, scale,
+                     causal ? 1 : 0);
 }
 
 void mfma_probe_launch(const void* A, const void* B, float* C, hipStream_t s) {

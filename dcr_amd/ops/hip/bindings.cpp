@@ -448,6 +448,25 @@ std::vector<at::Tensor> attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
   return {dQ, dK, dV};
 }
 
+// backward v4 draft (DCR_ATTN_BWD_V4): swapped-operand schedule
+std::vector<at::Tensor> attn_bwd_v4(at::Tensor q, at::Tensor k, at::Tensor v,
+                                 at::Tensor o, at::Tensor dO, at::Tensor lse,
+                                 double scale, bool causal) {
+  int64_t B, H, Lq, Lk;
+  attn_dims(q, k, B, H, Lq, Lk);
+  auto dQ = at::empty_like(q);
+  auto dK = at::empty_like(k);
+  auto dV = at::empty_like(v);
+  auto delta = at::empty({B * H, Lq}, q.options().dtype(at::kFloat));
+  attn_bwd_v4_launch(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
+                  dO.contiguous().data_ptr(), lse.data_ptr<float>(),
+                  delta.data_ptr<float>(), dQ.data_ptr(), dK.data_ptr(),
+                  dV.data_ptr(), (int)(B * H), (int)Lq, (int)Lk, (int)H,
+                  (float)scale, causal, cur_stream());
+  return {dQ, dK, dV};
+}
+
+
 at::Tensor mfma_probe(at::Tensor A, at::Tensor B) {
   TORCH_CHECK(A.is_cuda() && A.scalar_type() == at::kBFloat16);
   TORCH_CHECK(A.sizes() == at::IntArrayRef({16, 32}) &&
@@ -649,6 +668,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("attn_fwd_v3", &attn_fwd_v3);
   mod.def("attn_fwd_v4", &attn_fwd_v4);
   mod.def("attn_bwd", &attn_bwd);
+  mod.def("attn_bwd_v4", &attn_bwd_v4);
   mod.def("mfma_probe", &mfma_probe);
   mod.def("groupnorm_silu_fwd", &groupnorm_silu_fwd);
   mod.def("groupnorm_silu_bwd", &groupnorm_silu_bwd);

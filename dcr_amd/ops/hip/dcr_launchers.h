@@ -89,6 +89,12 @@ void attn_bwd_launch(const void* q, const void* k, const void* v,
                      float* delta, void* dQ, void* dK, void* dV, int BH,
                      int Lq, int Lk, int H, float scale, bool causal,
                      hipStream_t s);
+// backward v4 (swapped-operand schedule; DCR_ATTN_BWD_V4 draft)
+void attn_bwd_v4_launch(const void* q, const void* k, const void* v,
+                        const void* o, const void* dO, const float* lse,
+                        float* delta, void* dQ, void* dK, void* dV, int BH,
+                        int Lq, int Lk, int H, float scale, bool causal,
+                        hipStream_t s);
 void mfma_probe_launch(const void* A, const void* B, float* C, hipStream_t s);
 
 // conv_nhwc.hip (implicit-GEMM conv fwd, opt-in)

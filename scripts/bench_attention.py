@@ -67,6 +67,10 @@ def main():
             o, lse = m.attn_fwd(q, k, v, scale, causal)
             m.attn_bwd(q, k, v, o, o, lse, scale, causal)
 
+        def ours_bwd_v4():
+            o, lse = m.attn_fwd_v4(q, k, v, scale, causal)
+            m.attn_bwd_v4(q, k, v, o, o, lse, scale, causal)
+
         def sdpa_bwd():
             q2 = qp.detach().requires_grad_(True)
             k2 = kp.detach().requires_grad_(True)
@@ -76,9 +80,11 @@ def main():
             out.backward(out.detach())
 
         t_ob = timeit(ours_bwd, n=10)
+        t_ob4 = timeit(ours_bwd_v4, n=10)
         t_sb = timeit(sdpa_bwd, n=10)
-        print(f"    fwd+bwd: ours {t_ob:.3f} ms sdpa {t_sb:.3f} ms "
-              f"-> {t_sb / t_ob:.2f}x")
+        print(f"    fwd+bwd: ours {t_ob:.3f} ms v4 {t_ob4:.3f} ms "
+              f"sdpa {t_sb:.3f} ms -> {t_sb / t_ob:.2f}x, "
+              f"bwd v4/v1 {t_ob / t_ob4:.2f}x")
 
 
 if __name__ == "__main__":

@@ -745,3 +745,36 @@ def test_attn_fwd_v4_matches_ref(ext, Lq, Lk, causal):
     _close(o4, ref, 2e-2)
     o1, l1 = ext.attn_fwd(q, k, v, scale, causal)
     assert (l4 - l1).abs().max().item() < 1e-4
+
+
+@pytest.mark.parametrize("Lq,Lk,causal", [(256, 256, False), (1024, 1024, False),
+                                          (256, 77, False), (300, 100, False),
+                                          (512, 512, True)])
+def test_attn_bwd_v4_matches_ref(ext, Lq, Lk, causal):
+    """v4 backward (swapped-operand, register-resident P/dS) vs torch
+    fp32 autograd on random data."""
+    torch.manual_seed(Lq * 3 + Lk)
+    B, H, D = 2, 3, 64
+    q = torch.randn(B, Lq, H, D, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(B, Lk, H, D, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(B, Lk, H, D, device="cuda", dtype=torch.bfloat16)
+    g = torch.randn(B, Lq, H, D, device="cuda", dtype=torch.bfloat16)
+    scale = 1.0 / D ** 0.5
+    o, lse = ext.attn_fwd(q, k, v, scale, causal)
+    dQ, dK, dV = ext.attn_bwd_v4(q, k, v, o, g, lse, scale, causal)
+
+    qf = q.permute(0, 2, 1, 3).float().requires_grad_(True)
+    kf = k.permute(0, 2, 1, 3).float().requires_grad_(True)
+    vf = v.permute(0, 2, 1, 3).float().requires_grad_(True)
+    of = F.scaled_dot_product_attention(qf, kf, vf, is_causal=causal,
+                                        scale=scale)
+    of.backward(g.permute(0, 2, 1, 3).float())
+    _close(dQ, qf.grad.permute(0, 2, 1, 3), 3e-2)
+    _close(dK, kf.grad.permute(0, 2, 1, 3), 3e-2)
+    _close(dV, vf.grad.permute(0, 2, 1, 3), 3e-2)
+
+    # and against the production v1 backward (same inputs/lse)
+    dQ1, dK1, dV1 = ext.attn_bwd(q, k, v, o, g, lse, scale, causal)
+    _close(dQ, dQ1.float(), 2e-2)
+    _close(dK, dK1.float(), 2e-2)
+    _close(dV, dV1.float(), 2e-2)
