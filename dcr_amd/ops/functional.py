@@ -197,9 +197,14 @@ class _FlashAttention(torch.autograd.Function):
     def backward(ctx, dO):
         q, k, v, o, lse = ctx.saved_tensors
         m = require_hip("attn")
-        # DCR_ATTN_BWD_V4=1: swapped-operand backward draft (A/B only)
-        bwd = m.attn_bwd_v4 if os.environ.get("DCR_ATTN_BWD_V4") == "1" \
-            else m.attn_bwd
+        # measured dispatch (gpurun_out/r02c16): the swapped-operand v4
+        # backward wins for self-attention (1.27x @256, 1.13x @1024,
+        # 1.56x @4096 vs v1) and loses at tiny L / 77-token cross-attn
+        # where its 256-row blocks starve. DCR_ATTN_BWD_V4=0/1 overrides.
+        env = os.environ.get("DCR_ATTN_BWD_V4", "")
+        use_v4 = (env == "1") if env in ("0", "1") \
+            else (q.shape[1] >= 256 and k.shape[1] >= 256)
+        bwd = m.attn_bwd_v4 if use_v4 else m.attn_bwd
         dQ, dK, dV = bwd(q, k, v, o, dO.contiguous(), lse,
                          ctx.scale, ctx.causal)
         return dQ, dK, dV, None, None
