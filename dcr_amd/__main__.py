@@ -27,7 +27,7 @@ def main():
         print(f"dispatch counts: {dict(ops.dispatch_counts)}")
     import os
     gates = ["DCR_NATIVE_CONV", "DCR_NATIVE_CONV_BWD", "DCR_NATIVE_GEMM",
-             "DCR_ATTN_V4",
+             "DCR_ATTN_V4", "DCR_ATTN_BWD_V4", "DCR_HIPGRAPH",
              "DCR_ATTN_V2", "DCR_DEV_ADAMW", "DCR_PROFILE",
              "DCR_AMD_ALLOW_FALLBACK"]
     active = {g: os.environ[g] for g in gates if g in os.environ}
