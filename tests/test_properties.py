@@ -89,3 +89,59 @@ def test_add_noise_velocity_identity(n, t):
     ac = s.alphas_cumprod[tt].view(-1, 1, 1, 1)
     rec = ac.sqrt() * xt - (1 - ac).sqrt() * v
     assert torch.allclose(rec, x0, atol=1e-5)
+
+
+@settings(**SET)
+@given(text=st.text(alphabet=st.characters(min_codepoint=32,
+                                           max_codepoint=126), max_size=60),
+       max_length=st.integers(8, 64))
+def test_clip_bpe_tokenizer_properties(tmp_path_factory, text, max_length):
+    """CLIPBPETokenizer: deterministic, BOS/EOS framing, fixed padded
+    length, ids in-vocab — for arbitrary printable input."""
+    import json
+    from dcr_amd.data.tokenizer import CLIPBPETokenizer, _bytes_to_unicode
+    global _BPE_TOK
+    try:
+        tok = _BPE_TOK
+    except NameError:
+        d = tmp_path_factory.mktemp("bpe")
+        b2u = _bytes_to_unicode()
+        vocab = {}
+        for c in b2u.values():
+            vocab[c] = len(vocab)
+        for c in b2u.values():
+            vocab[c + "</w>"] = len(vocab)
+        vocab["<|startoftext|>"] = len(vocab)
+        vocab["<|endoftext|>"] = len(vocab)
+        (d / "vocab.json").write_text(json.dumps(vocab))
+        (d / "merges.txt").write_text("#version: 0.2\n")
+        tok = _BPE_TOK = CLIPBPETokenizer(d / "vocab.json", d / "merges.txt")
+    ids = tok([text], max_length=max_length).input_ids[0].tolist()
+    assert len(ids) == max_length
+    assert ids[0] == tok.bos_token_id
+    assert tok.eos_token_id in ids[1:]
+    assert all(0 <= i < tok.vocab_size for i in ids)
+    assert ids == tok([text], max_length=max_length).input_ids[0].tolist()
+
+
+@settings(**SET)
+@given(sizes=st.lists(st.integers(1, 40), min_size=1, max_size=6),
+       accum=st.integers(1, 3))
+def test_gather_grads_equals_autograd_sum(sizes, accum):
+    """gather_grads over `accum` backward passes reproduces exactly what
+    plain autograd accumulation would (copy-then-add invariant of the
+    flat-arena design)."""
+    torch.manual_seed(0)
+    params = [torch.nn.Parameter(torch.randn(n)) for n in sizes]
+    from dcr_amd.ops.adamw import FusedAdamW
+    opt = FusedAdamW(params, lr=0.0, weight_decay=0.0)
+    opt.zero_grad()
+    expect = [torch.zeros_like(p) for p in params]
+    for a in range(accum):
+        loss = sum(((i + 1) * p * (a + 1)).sum() for i, p in enumerate(params))
+        loss.backward()
+        for i, p in enumerate(params):
+            expect[i] += torch.full_like(p, float((i + 1) * (a + 1)))
+        opt.gather_grads()
+    for i, p in enumerate(params):
+        assert torch.allclose(opt._view_of[id(p)], expect[i], atol=1e-5)
